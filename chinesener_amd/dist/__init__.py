@@ -1,0 +1,144 @@
+"""Data-parallel runtime: RCCL over xGMI, bucketed all-reduce overlap.
+
+New capability vs the single-process reference:
+one process per GPU (torchrun), gradients all-reduced in ~25 MB buckets
+launched asynchronously as backward produces them, so communication
+overlaps the remaining backward compute. Buckets default to bf16 on the
+RCCL backend (half the xGMI bytes; the 7 p2p links x ~153 GB/s make
+all-reduce per-link bound) and fp32 on gloo (CPU tests).
+"""
+from __future__ import annotations
+
+import os
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+
+def init_process_group(backend: Optional[str] = None):
+    """Initialize from torchrun env vars; returns (rank, world_size)."""
+    if dist.is_initialized():
+        return dist.get_rank(), dist.get_world_size()
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29500")
+    dist.init_process_group(backend=backend, rank=rank, world_size=world)
+    return rank, world
+
+
+class _Bucket:
+    __slots__ = ("params", "flat", "ready", "work", "offsets")
+
+    def __init__(self, params: List[torch.nn.Parameter], dtype: torch.dtype,
+                 device: torch.device):
+        self.params = params
+        self.offsets: List[int] = []
+        total = 0
+        for p in params:
+            self.offsets.append(total)
+            total += p.numel()
+        self.flat = torch.zeros(total, dtype=dtype, device=device)
+        self.ready = 0
+        self.work = None
+
+
+class BucketedDataParallel:
+    """Bucketed gradient all-reduce with backward overlap.
+
+    Usage: construct after moving the model to its device; run backward;
+    call ``finalize_backward()`` before clipping/stepping; call
+    ``zero_grad()`` instead of optimizer.zero_grad(set_to_none=True).
+    """
+
+    def __init__(self, model: torch.nn.Module, bucket_cap_mb: float = 25.0,
+                 grad_dtype: Optional[torch.dtype] = None,
+                 process_group=None):
+        assert dist.is_initialized(), "init_process_group first"
+        self.model = model
+        self.group = process_group
+        self.world_size = dist.get_world_size(process_group)
+        backend = dist.get_backend(process_group)
+        if grad_dtype is None:
+            grad_dtype = torch.bfloat16 if backend == "nccl" else torch.float32
+        self.grad_dtype = grad_dtype
+
+        params = [p for p in model.parameters() if p.requires_grad]
+        # broadcast initial params from rank 0 (DP init)
+        with torch.no_grad():
+            for p in params:
+                dist.broadcast(p.data, src=0, group=self.group)
+
+        # bucket in REVERSE registration order ~= backward readiness order
+        cap = int(bucket_cap_mb * 1e6 / grad_dtype.itemsize)
+        self.buckets: List[_Bucket] = []
+        self.param2bucket: Dict[torch.nn.Parameter, tuple] = {}
+        cur: List[torch.nn.Parameter] = []
+        size = 0
+        device = params[0].device
+        for p in reversed(params):
+            cur.append(p)
+            size += p.numel()
+            if size >= cap:
+                self._seal(cur, device)
+                cur, size = [], 0
+        if cur:
+            self._seal(cur, device)
+
+        for p in params:
+            p.register_post_accumulate_grad_hook(self._hook)
+
+    def _seal(self, params: List[torch.nn.Parameter], device):
+        b = _Bucket(params, self.grad_dtype, device)
+        for i, p in enumerate(params):
+            self.param2bucket[p] = (b, i)
+        self.buckets.append(b)
+
+    def _hook(self, p: torch.nn.Parameter):
+        b, i = self.param2bucket[p]
+        off = b.offsets[i]
+        b.flat[off:off + p.numel()].copy_(p.grad.detach().reshape(-1))
+        b.ready += 1
+        if b.ready == len(b.params):
+            b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                     group=self.group, async_op=True)
+
+    def finalize_backward(self):
+        """Wait outstanding all-reduces, write averaged grads back."""
+        inv = 1.0 / self.world_size
+        for b in self.buckets:
+            if b.ready != len(b.params):
+                # params unused this step (e.g. frozen paths): reduce what we have
+                b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                         group=self.group, async_op=True)
+            if b.work is not None:
+                b.work.wait()
+            for i, p in enumerate(b.params):
+                if p.grad is None:
+                    continue
+                off = b.offsets[i]
+                p.grad.detach().reshape(-1).copy_(b.flat[off:off + p.numel()])
+                p.grad.detach().mul_(inv)
+            b.ready = 0
+            b.work = None
+
+    def zero_grad(self):
+        for b in self.buckets:
+            b.flat.zero_()
+            b.ready = 0
+            b.work = None
+        for p in self.param2bucket:
+            if p.grad is not None:
+                p.grad = None
+
+
+def all_reduce_scalar(value: float, device="cpu") -> float:
+    t = torch.tensor([value], device=device)
+    dist.all_reduce(t)
+    return float(t.item()) / dist.get_world_size()

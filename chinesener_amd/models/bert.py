@@ -1,0 +1,131 @@
+"""BERT encoder, MI355X-first.
+
+Capability parity with the reference's use of bert_base.bert.modeling
+BertModel (tools/layer.py:63-81; config pretrain_model/ch_google/
+bert_config.json: 12L-768H-12A, gelu, vocab 21128). Re-designed for the
+hardware: the projection/FFN GEMMs go through rocBLAS/hipBLASLt
+(torch.matmul); attention core, residual+LayerNorm and bias+GELU are the
+hand-written gfx950 kernels behind chinesener_amd.ops.
+Compute dtype is bf16 on GPU; LayerNorm statistics are fp32 inside the
+kernel.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+
+
+@dataclass
+class BertConfig:
+    vocab_size: int = 21128
+    hidden_size: int = 768
+    num_hidden_layers: int = 12
+    num_attention_heads: int = 12
+    intermediate_size: int = 3072
+    max_position_embeddings: int = 512
+    type_vocab_size: int = 2
+    hidden_dropout_prob: float = 0.1
+    attention_probs_dropout_prob: float = 0.1
+    layer_norm_eps: float = 1e-12
+    initializer_range: float = 0.02
+
+
+class BertEmbeddings(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.word = nn.Embedding(cfg.vocab_size, cfg.hidden_size, padding_idx=0)
+        self.position = nn.Embedding(cfg.max_position_embeddings, cfg.hidden_size)
+        self.token_type = nn.Embedding(cfg.type_vocab_size, cfg.hidden_size)
+        self.ln_weight = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.ln_bias = nn.Parameter(torch.zeros(cfg.hidden_size))
+        self.dropout = nn.Dropout(cfg.hidden_dropout_prob)
+        self.eps = cfg.layer_norm_eps
+
+    def forward(self, token_ids, token_type_ids=None):
+        B, L = token_ids.shape
+        pos = torch.arange(L, device=token_ids.device).unsqueeze(0)
+        x = self.word(token_ids) + self.position(pos)
+        if token_type_ids is not None:
+            x = x + self.token_type(token_type_ids)
+        else:
+            x = x + self.token_type.weight[0]
+        x = ops.layernorm(x, self.ln_weight, self.ln_bias, self.eps)
+        return self.dropout(x)
+
+
+class BertLayer(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        H = cfg.hidden_size
+        self.n_heads = cfg.num_attention_heads
+        self.head_dim = H // cfg.num_attention_heads
+        self.qkv = nn.Linear(H, 3 * H)       # fused QKV projection (one GEMM)
+        self.attn_out = nn.Linear(H, H)
+        self.ln1_w = nn.Parameter(torch.ones(H))
+        self.ln1_b = nn.Parameter(torch.zeros(H))
+        self.ffn_in = nn.Linear(H, cfg.intermediate_size, bias=False)
+        self.ffn_in_bias = nn.Parameter(torch.zeros(cfg.intermediate_size))
+        self.ffn_out = nn.Linear(cfg.intermediate_size, H)
+        self.ln2_w = nn.Parameter(torch.ones(H))
+        self.ln2_b = nn.Parameter(torch.zeros(H))
+        self.dropout = nn.Dropout(cfg.hidden_dropout_prob)
+        self.eps = cfg.layer_norm_eps
+
+    def forward(self, x, mask, lens):
+        B, L, H = x.shape
+        qkv = self.qkv(x).reshape(B, L, 3, self.n_heads, self.head_dim)
+        q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # [B,h,L,d]
+        ctx = ops.attention(q, k, v, mask=mask, lens=lens)
+        ctx = ctx.transpose(1, 2).reshape(B, L, H)
+        a = self.dropout(self.attn_out(ctx))
+        x = ops.add_layernorm(a, x, self.ln1_w, self.ln1_b, self.eps)
+        f = ops.bias_gelu(self.ffn_in(x), self.ffn_in_bias)
+        f = self.dropout(self.ffn_out(f))
+        return ops.add_layernorm(f, x, self.ln2_w, self.ln2_b, self.eps)
+
+
+class BertModel(nn.Module):
+    """Returns the last-layer sequence output [B,L,H] (the reference's
+    pretrain_bert_embedding returns sequence output + dropout,
+    tools/layer.py:63-81)."""
+
+    def __init__(self, cfg: BertConfig | None = None):
+        super().__init__()
+        self.cfg = cfg or BertConfig()
+        self.embeddings = BertEmbeddings(self.cfg)
+        self.layers = nn.ModuleList(
+            BertLayer(self.cfg) for _ in range(self.cfg.num_hidden_layers))
+        self.apply(self._init)
+
+    def _init(self, m):
+        if isinstance(m, (nn.Linear, nn.Embedding)):
+            nn.init.normal_(m.weight, std=self.cfg.initializer_range)
+            if isinstance(m, nn.Linear) and m.bias is not None:
+                nn.init.zeros_(m.bias)
+            if isinstance(m, nn.Embedding) and m.padding_idx is not None:
+                with torch.no_grad():
+                    m.weight[m.padding_idx].zero_()
+
+    def forward(self, token_ids, mask=None, token_type_ids=None):
+        lens = mask.long().sum(1) if mask is not None else None
+        x = self.embeddings(token_ids, token_type_ids)
+        for layer in self.layers:
+            x = layer(x, mask, lens)
+        return x
+
+
+class BertEmbeddingOnly(nn.Module):
+    """Wordpiece embedding + position postprocessor only, no encoder —
+    the reference's bert_token_embedding (tools/layer.py:84-109), used by
+    bert_bilstm_crf_bigram with use_bert=False."""
+
+    def __init__(self, cfg: BertConfig | None = None):
+        super().__init__()
+        self.embeddings = BertEmbeddings(cfg or BertConfig())
+
+    def forward(self, token_ids):
+        return self.embeddings(token_ids)

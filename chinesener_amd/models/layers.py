@@ -1,0 +1,100 @@
+"""Shared layer library (parity with reference tools/layer.py).
+
+BiLSTM (:10-41), multi-kernel CNN (:44-60), CRF layer (:112-149) live
+here as nn.Modules composed from the ops layer; the BERT encoder is in
+bert.py, transformer encoders in transformer.py.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+
+
+class BiLSTM(nn.Module):
+    """Bidirectional LSTM with the reference's DropoutWrapper semantics
+    (output keep-prob; activation tanh/relu selectable —
+    tools/layer.py:10-41, model/bilstm_crf.py:48-52)."""
+
+    def __init__(self, input_size: int, hidden_size: int, activation: str = "tanh",
+                 keep_prob: float = 0.8):
+        super().__init__()
+        self.hidden_size = hidden_size
+        self.activation = activation
+        self.dropout = nn.Dropout(1.0 - keep_prob)
+        k = 1.0 / (hidden_size ** 0.5)
+
+        def p(*shape):
+            return nn.Parameter(torch.empty(*shape).uniform_(-k, k))
+        self.w_ih_f, self.w_hh_f, self.b_f = p(input_size, 4 * hidden_size), \
+            p(hidden_size, 4 * hidden_size), p(4 * hidden_size)
+        self.w_ih_b, self.w_hh_b, self.b_b = p(input_size, 4 * hidden_size), \
+            p(hidden_size, 4 * hidden_size), p(4 * hidden_size)
+
+    def forward(self, x: torch.Tensor, lens: torch.Tensor) -> torch.Tensor:
+        out = ops.bilstm(x, self.w_ih_f, self.w_hh_f, self.b_f,
+                         self.w_ih_b, self.w_hh_b, self.b_b, lens,
+                         self.activation)
+        return self.dropout(out)
+
+
+class MultiKernelCNN(nn.Module):
+    """Per-kernel conv1d SAME + dropout, concat filters
+    (reference cnn_layer, tools/layer.py:44-60)."""
+
+    def __init__(self, input_size: int, filters: int = 128,
+                 kernel_sizes: List[int] = (2, 3, 4), keep_prob: float = 0.8):
+        super().__init__()
+        self.convs = nn.ModuleList([
+            nn.Conv1d(input_size, filters, k, padding="same")
+            for k in kernel_sizes])
+        self.dropout = nn.Dropout(1.0 - keep_prob)
+        self.output_size = filters * len(kernel_sizes)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:   # [B,L,E]
+        xt = x.transpose(1, 2)
+        outs = [self.dropout(torch.relu(c(xt))) for c in self.convs]
+        return torch.cat(outs, dim=1).transpose(1, 2)
+
+
+class CRF(nn.Module):
+    """Linear-chain CRF head (reference crf_layer/crf_decode,
+    tools/layer.py:112-149): loss over the full padded length including
+    CLS/SEP (seq_len counts them, comment :121)."""
+
+    def __init__(self, num_tags: int):
+        super().__init__()
+        self.num_tags = num_tags
+        self.transitions = nn.Parameter(torch.empty(num_tags, num_tags))
+        nn.init.xavier_uniform_(self.transitions)
+
+    def neg_log_likelihood(self, emissions, tags, mask) -> torch.Tensor:
+        """Sum of -log p(tags | emissions) over the batch."""
+        ll = ops.crf_nll(emissions.float(), tags, mask, self.transitions)
+        return -ll.sum()
+
+    def decode(self, emissions, mask) -> torch.Tensor:
+        return ops.crf_viterbi(emissions.float(), mask, self.transitions)
+
+
+class TokenEmbedding(nn.Module):
+    """Char/word embedding, optionally initialized from a pretrained matrix
+    carried in data params (reference base_preprocess.py:223-226) and
+    optionally frozen (bichar pretrained constant, bilstm_crf_bichar.py)."""
+
+    def __init__(self, vocab_size: int, dim: int,
+                 pretrained: Optional[torch.Tensor] = None, freeze: bool = False,
+                 padding_idx: int = 0):
+        super().__init__()
+        self.emb = nn.Embedding(vocab_size, dim, padding_idx=padding_idx)
+        if pretrained is not None:
+            with torch.no_grad():
+                self.emb.weight.copy_(torch.as_tensor(pretrained))
+        if freeze:
+            self.emb.weight.requires_grad_(False)
+
+    def forward(self, ids):
+        return self.emb(ids)

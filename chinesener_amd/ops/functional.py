@@ -1,0 +1,278 @@
+"""Dispatching op API: HIP autograd.Functions on GPU, reference on CPU.
+
+Each public function mirrors one row of SURVEY.md §2.6's kernel table
+(K2 softlexicon, K3/K8 attention, K4 BiLSTM, K5/K6 CRF, K9 TENER,
+K10 LayerNorm, K11 bias-GELU, K13 masked CE, K14 dice).
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+
+from . import reference as ref
+from . import hip_enabled, get_ext
+
+
+# ----------------------------------------------------------- layer norm
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        y, mean, rstd = get_ext().layernorm_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, rstd = ctx.saved_tensors
+        dx, dw, db = get_ext().layernorm_bwd(dy.contiguous(), x, weight, mean, rstd)
+        return dx, dw, db, None
+
+
+def layernorm(x, weight, bias, eps: float = 1e-12):
+    if hip_enabled(x):
+        shape = x.shape
+        y = _LayerNormFn.apply(x.reshape(-1, shape[-1]).contiguous(), weight, bias, eps)
+        return y.reshape(shape)
+    return ref.layernorm(x, weight, bias, eps)
+
+
+class _AddLayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, residual, weight, bias, eps):
+        y, s, mean, rstd = get_ext().add_layernorm_fwd(x, residual, weight, bias, eps)
+        ctx.save_for_backward(s, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        s, weight, mean, rstd = ctx.saved_tensors
+        dx, dw, db = get_ext().layernorm_bwd(dy.contiguous(), s, weight, mean, rstd)
+        return dx, dx, dw, db, None
+
+
+def add_layernorm(x, residual, weight, bias, eps: float = 1e-12):
+    """LayerNorm(x + residual) — the BERT post-LN residual pattern."""
+    if hip_enabled(x):
+        shape = x.shape
+        y = _AddLayerNormFn.apply(x.reshape(-1, shape[-1]).contiguous(),
+                                  residual.reshape(-1, shape[-1]).contiguous(),
+                                  weight, bias, eps)
+        return y.reshape(shape)
+    return ref.add_layernorm(x, residual, weight, bias, eps)
+
+
+# ------------------------------------------------------------ bias gelu
+class _BiasGeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, bias):
+        y = get_ext().bias_gelu_fwd(x, bias)
+        ctx.save_for_backward(x, bias)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, bias = ctx.saved_tensors
+        dx, dbias = get_ext().bias_gelu_bwd(dy.contiguous(), x, bias)
+        return dx, dbias
+
+
+def bias_gelu(x, bias):
+    if bias is not None and hip_enabled(x):
+        shape = x.shape
+        y = _BiasGeluFn.apply(x.reshape(-1, shape[-1]).contiguous(), bias)
+        return y.reshape(shape)
+    return ref.bias_gelu(x, bias)
+
+
+# ------------------------------------------------------------ attention
+class _AttentionFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, lens, scale):
+        out, lse = get_ext().attn_fwd(q, k, v, lens, scale)
+        ctx.save_for_backward(q, k, v, out, lse, lens)
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse, lens = ctx.saved_tensors
+        dq, dk, dv = get_ext().attn_bwd(dout.contiguous(), q, k, v, out, lse,
+                                        lens, ctx.scale)
+        return dq, dk, dv, None, None
+
+
+def attention(q, k, v, mask: Optional[torch.Tensor] = None,
+              scale: Optional[float] = None, lens: Optional[torch.Tensor] = None):
+    """Fused attention. q,k,v [B,H,L,D]; mask [B,L] prefix mask or lens [B]."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if hip_enabled(q):
+        if lens is None:
+            lens = (mask.long().sum(1) if mask is not None
+                    else torch.full((q.shape[0],), q.shape[2],
+                                    dtype=torch.long, device=q.device))
+        return _AttentionFn.apply(q.contiguous(), k.contiguous(), v.contiguous(),
+                                  lens.to(torch.int32), float(scale))
+    return ref.attention(q, k, v, mask, scale)
+
+
+class _TenerAttentionFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, u, vb, rel, lens):
+        out, lse = get_ext().tener_attn_fwd(q, k, v, u, vb, rel, lens)
+        ctx.save_for_backward(q, k, v, u, vb, rel, out, lse, lens)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, u, vb, rel, out, lse, lens = ctx.saved_tensors
+        dq, dk, dv, du, dvb = get_ext().tener_attn_bwd(
+            dout.contiguous(), q, k, v, u, vb, rel, out, lse, lens)
+        return dq, dk, dv, du, dvb, None, None
+
+
+def tener_attention(q, k, v, u, vb, rel, mask: Optional[torch.Tensor] = None):
+    """TENER relative attention (unscaled, no key projection upstream)."""
+    if hip_enabled(q):
+        lens = (mask.long().sum(1) if mask is not None
+                else torch.full((q.shape[0],), q.shape[2], dtype=torch.long,
+                                device=q.device))
+        return _TenerAttentionFn.apply(
+            q.contiguous(), k.contiguous(), v.contiguous(), u.contiguous(),
+            vb.contiguous(), rel.contiguous(), lens.to(torch.int32))
+    return ref.tener_attention(q, k, v, u, vb, rel, mask)
+
+
+# ------------------------------------------------------------------ crf
+class _CrfNllFn(torch.autograd.Function):
+    """log-likelihood [B]; kernel computes grads (marginals) in the same
+    forward-backward pass."""
+
+    @staticmethod
+    def forward(ctx, emissions, tags, lens, transitions):
+        ll, demis, dtrans = get_ext().crf_fwd(emissions, tags, lens, transitions)
+        ctx.save_for_backward(demis, dtrans)
+        return ll
+
+    @staticmethod
+    def backward(ctx, dll):
+        demis, dtrans = ctx.saved_tensors
+        # d(ll_b)/d(emissions) stored as +(gold - expected); chain rule with dll.
+        return (demis * dll[:, None, None],
+                None, None,
+                (dtrans * dll[:, None, None]).sum(0))
+
+
+def crf_nll(emissions, tags, mask, transitions):
+    """Per-sequence log-likelihood [B] (caller negates/averages)."""
+    if hip_enabled(emissions):
+        lens = mask.long().sum(1).to(torch.int32)
+        return _CrfNllFn.apply(emissions.float().contiguous(),
+                               tags.to(torch.int32).contiguous(), lens,
+                               transitions.float().contiguous())
+    return ref.crf_log_likelihood(emissions, tags, mask, transitions)
+
+
+def crf_viterbi(emissions, mask, transitions):
+    if hip_enabled(emissions):
+        lens = mask.long().sum(1).to(torch.int32)
+        return get_ext().crf_viterbi(emissions.float().contiguous(), lens,
+                                     transitions.float().contiguous()).long()
+    with torch.no_grad():
+        return ref.crf_decode(emissions, mask, transitions)
+
+
+# ----------------------------------------------------------------- lstm
+class _LstmDirFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gates_x, w_hh, lens, reverse, act_relu):
+        hs, cs, gates = get_ext().lstm_fwd(gates_x, w_hh, lens, reverse, act_relu)
+        ctx.save_for_backward(hs, cs, gates, w_hh, lens)
+        ctx.flags = (reverse, act_relu)
+        return hs
+
+    @staticmethod
+    def backward(ctx, dhs):
+        hs, cs, gates, w_hh, lens = ctx.saved_tensors
+        reverse, act_relu = ctx.flags
+        dgates_x, dw_hh = get_ext().lstm_bwd(dhs.contiguous(), hs, cs, gates,
+                                             w_hh, lens, reverse, act_relu)
+        return dgates_x, dw_hh, None, None, None
+
+
+def _lstm_dir(x, w_ih, w_hh, b, lens, reverse, activation):
+    # x-projection as one library GEMM; the HIP kernel owns the recurrence.
+    gates_x = (x @ w_ih.to(x.dtype) + b.to(x.dtype))
+    return _LstmDirFn.apply(gates_x.contiguous(),
+                            w_hh.to(gates_x.dtype).contiguous(),
+                            lens.to(torch.int32), bool(reverse),
+                            activation == "relu")
+
+
+def bilstm(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b, lens,
+           activation: str = "tanh", state_dropout=None):
+    """BiLSTM over padded [B,L,E] -> [B,L,2h]."""
+    if hip_enabled(x):
+        fw = _lstm_dir(x, w_ih_f, w_hh_f, b_f, lens, False, activation)
+        bw = _lstm_dir(x, w_ih_b, w_hh_b, b_b, lens, True, activation)
+        return torch.cat([fw, bw], dim=-1)
+    return ref.bilstm_forward(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b,
+                              lens, activation, state_dropout)
+
+
+# ----------------------------------------------------------- softlexicon
+class _SoftLexiconFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, table, ids, weights):
+        out = get_ext().softlexicon_fwd(table, ids, weights)
+        ctx.save_for_backward(table, ids, weights)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        table, ids, weights = ctx.saved_tensors
+        dtable, dweights = get_ext().softlexicon_bwd(dout.contiguous(), table,
+                                                     ids, weights)
+        return dtable, None, dweights
+
+
+def softlexicon_fuse(table, ids, weights):
+    """Fused gather-scale-reduce: [V,E] x [B,L,40] -> [B,L,4E]."""
+    if hip_enabled(table):
+        return _SoftLexiconFn.apply(table.contiguous(),
+                                    ids.to(torch.int32).contiguous(),
+                                    weights.contiguous())
+    return ref.softlexicon_fuse(table, ids, weights)
+
+
+# ---------------------------------------------------------------- losses
+class _MaskedCeFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels, mask):
+        loss, probs, nvalid = get_ext().masked_ce_fwd(logits, labels, mask)
+        ctx.save_for_backward(probs, labels, mask, nvalid)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        probs, labels, mask, nvalid = ctx.saved_tensors
+        dlogits = get_ext().masked_ce_bwd(dloss, probs, labels, mask, nvalid)
+        return dlogits, None, None
+
+
+def masked_cross_entropy(logits, labels, mask):
+    if hip_enabled(logits):
+        T = logits.shape[-1]
+        return _MaskedCeFn.apply(logits.reshape(-1, T).float().contiguous(),
+                                 labels.reshape(-1).to(torch.int32).contiguous(),
+                                 mask.reshape(-1).to(torch.int32).contiguous())
+    return ref.masked_cross_entropy(logits, labels, mask)
+
+
+def dice_loss(logits, labels, mask, idx_skip: Tuple[int, ...],
+              alpha: float = 0.1, gamma: float = 1.0):
+    # reduction over tags, not perf-critical (eval-style loss) — reference
+    # implementation runs on both devices.
+    return ref.dice_loss(logits, labels, mask, idx_skip, alpha, gamma)

@@ -1,0 +1,174 @@
+"""Optimizers + LR schedules (reference tools/train_utils.py:246-390).
+
+Three families, chosen by model-name substring like the reference
+(:156-164):
+
+* ``bert``: AdamWeightDecay (no bias correction, decoupled wd excluding
+  LayerNorm/bias — :276-284) with linear warmup + polynomial decay, plus
+  per-layer-group differential LR (``diff_lr_times`` name-substring
+  groups sharing globally-clipped grads — :287-337).
+* ``transformer``: Adam with the Noam scheme (modules.py:209-217).
+* ``custom``: Adam with exponential decay (:340-376) and value clipping.
+
+On GPU the step runs the multi-tensor fused Adam HIP kernel (SURVEY.md
+K16); CPU uses torch foreach ops with identical math.
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, Iterable, List, Optional, Tuple
+
+import torch
+
+from .. import ops
+
+
+def _no_decay(name: str) -> bool:
+    # LayerNorm + bias excluded from weight decay (reference :282)
+    n = name.lower()
+    return n.endswith("bias") or "ln_" in n or "layernorm" in n or ".ln" in n
+
+
+def build_param_groups(model: torch.nn.Module, base_lr: float,
+                       weight_decay: float,
+                       diff_lr_times: Optional[Dict[str, float]] = None
+                       ) -> List[Dict]:
+    """Split params into (diff-lr substring x wd/no-wd) groups.
+
+    ``diff_lr_times`` maps a name substring to an LR multiplier
+    (reference bert_train_op :287-337: one optimizer per group, shared
+    clipped grads — here one optimizer with per-group lr_scale).
+    """
+    diff = diff_lr_times or {}
+    groups: Dict[Tuple[str, bool], List] = {}
+    for name, p in model.named_parameters():
+        if not p.requires_grad:
+            continue
+        scale_key = ""
+        for sub in diff:
+            if sub in name:
+                scale_key = sub
+                break
+        groups.setdefault((scale_key, _no_decay(name)), []).append(p)
+    out = []
+    for (key, nodecay), params in groups.items():
+        out.append({
+            "params": params,
+            "lr_scale": float(diff.get(key, 1.0)),
+            "weight_decay": 0.0 if nodecay else weight_decay,
+        })
+    return out
+
+
+class AdamWeightDecay(torch.optim.Optimizer):
+    """BERT AdamWeightDecay: m,v EMA without bias correction; decoupled wd;
+    per-group lr = schedule_lr * lr_scale. GPU steps use the fused
+    multi-tensor HIP kernel when available."""
+
+    def __init__(self, params: Iterable, lr: float = 5e-5, betas=(0.9, 0.999),
+                 eps: float = 1e-6, weight_decay: float = 0.01):
+        defaults = dict(lr=lr, betas=betas, eps=eps,
+                        weight_decay=weight_decay, lr_scale=1.0)
+        super().__init__(params, defaults)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = closure() if closure is not None else None
+        for group in self.param_groups:
+            params = [p for p in group["params"] if p.grad is not None]
+            if not params:
+                continue
+            grads = [p.grad for p in params]
+            states = [self.state[p] for p in params]
+            for p, s in zip(params, states):
+                if "m" not in s:
+                    s["m"] = torch.zeros_like(p)
+                    s["v"] = torch.zeros_like(p)
+            ms = [s["m"] for s in states]
+            vs = [s["v"] for s in states]
+            lr = group["lr"] * group.get("lr_scale", 1.0)
+            b1, b2 = group["betas"]
+            eps, wd = group["eps"], group["weight_decay"]
+            if params[0].is_cuda and ops.ext_available():
+                ops.get_ext().multi_tensor_adamw(
+                    params, grads, ms, vs, lr, b1, b2, eps, wd)
+            else:
+                torch._foreach_mul_(ms, b1)
+                torch._foreach_add_(ms, grads, alpha=1 - b1)
+                torch._foreach_mul_(vs, b2)
+                torch._foreach_addcmul_(vs, grads, grads, value=1 - b2)
+                denom = torch._foreach_sqrt(vs)
+                torch._foreach_add_(denom, eps)
+                update = torch._foreach_div(ms, denom)
+                if wd:
+                    torch._foreach_add_(update, params, alpha=wd)
+                torch._foreach_add_(params, update, alpha=-lr)
+        return loss
+
+
+class LrSchedule:
+    """Per-step LR multiplier playing the reference's three schemes."""
+
+    def __init__(self, family: str, base_lr: float, num_train_steps: int,
+                 warmup_ratio: float = 0.1, step_per_epoch: int = 1000,
+                 decay_rate: float = 0.95, d_model: int = 160,
+                 warmup_steps: Optional[int] = None):
+        self.family = family
+        self.base_lr = base_lr
+        self.total = max(1, num_train_steps)
+        self.warmup = warmup_steps if warmup_steps is not None else max(
+            1, int(self.total * warmup_ratio))
+        self.step_per_epoch = max(1, step_per_epoch)
+        self.decay_rate = decay_rate
+        self.d_model = d_model
+
+    def lr_at(self, step: int) -> float:
+        s = max(1, step)
+        if self.family == "bert":
+            # linear warmup + polynomial (power 1) decay to 0 (:246-284)
+            if s < self.warmup:
+                return self.base_lr * s / self.warmup
+            frac = min(1.0, (s - self.warmup) / max(1, self.total - self.warmup))
+            return self.base_lr * (1.0 - frac)
+        if self.family == "transformer":
+            # Noam (modules.py:209-217)
+            return (self.base_lr * (self.d_model ** -0.5)
+                    * min(s ** -0.5, s * self.warmup ** -1.5) * self.warmup ** 0.5)
+        # exponential decay per epoch, staircase (:365-376)
+        return self.base_lr * (self.decay_rate ** (s // self.step_per_epoch))
+
+    def apply(self, optimizer: torch.optim.Optimizer, step: int) -> float:
+        lr = self.lr_at(step)
+        for g in optimizer.param_groups:
+            g["lr"] = lr
+        return lr
+
+
+def clip_gradients(model: torch.nn.Module, family: str) -> Optional[float]:
+    """bert/transformer: clip_by_global_norm 1.0 (:315,334); custom:
+    clip_by_value +-5 (:379-390). Returns grad-norm if computed."""
+    if family in ("bert", "transformer"):
+        return float(torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0))
+    torch.nn.utils.clip_grad_value_(model.parameters(), 5.0)
+    return None
+
+
+def build_optimizer(model: torch.nn.Module, family: str, params: Dict
+                    ) -> Tuple[torch.optim.Optimizer, LrSchedule]:
+    base_lr = params.get("lr", 1e-3)
+    schedule = LrSchedule(
+        family, base_lr,
+        num_train_steps=params.get("num_train_steps", 10000),
+        warmup_ratio=params.get("warmup_ratio", 0.1),
+        step_per_epoch=params.get("step_per_epoch", 1000),
+        decay_rate=params.get("lr_decay_rate", 0.95),
+        d_model=params.get("transformer_params", {}).get("d_model", 160))
+    if family == "bert":
+        groups = build_param_groups(model, base_lr,
+                                    params.get("weight_decay", 0.01),
+                                    params.get("diff_lr_times"))
+        opt = AdamWeightDecay(groups, lr=base_lr,
+                              weight_decay=params.get("weight_decay", 0.01))
+    else:
+        opt = torch.optim.Adam(model.parameters(), lr=base_lr)
+    return opt, schedule

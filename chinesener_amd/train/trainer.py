@@ -1,0 +1,157 @@
+"""Training loop (the reference's Estimator train_and_evaluate analog,
+main.py:47-49): step loop with bf16 autocast on GPU, gradient clipping,
+LR schedule, periodic logging/checkpointing, early stopping on eval loss
+(stop_if_no_decrease_hook, main.py:43-46), and predict-dump parity
+(main.py:52-55)."""
+from __future__ import annotations
+
+import logging
+import os
+import pickle
+import time
+from typing import Dict, Iterable, Optional
+
+import torch
+
+from ..models import ModelOutput, NerModel, optimizer_family
+from .checkpoints import CheckpointManager
+from .metrics import TagMetrics
+from .optimizers import build_optimizer, clip_gradients
+
+log = logging.getLogger("chinesener_amd")
+
+
+class Trainer:
+    def __init__(self, model: NerModel, model_name: str, params: Dict,
+                 ckpt_dir: str, device: Optional[str] = None,
+                 dp_engine=None, rank: int = 0):
+        self.model = model
+        self.model_name = model_name
+        self.params = params
+        self.rank = rank
+        self.device = device or ("cuda" if torch.cuda.is_available() else "cpu")
+        self.model.to(self.device)
+        self.family = optimizer_family(model_name)
+        self.optimizer, self.schedule = build_optimizer(model, self.family, params)
+        self.ckpt = CheckpointManager(ckpt_dir,
+                                      params.get("keep_checkpoint_max", 3))
+        self.dp = dp_engine
+        self.use_bf16 = (self.device.startswith("cuda")
+                         and params.get("dtype", "bf16") == "bf16")
+        self.step = self.ckpt.restore(model, self.optimizer,
+                                      map_location=self.device)
+
+    # ------------------------------------------------------------- train
+    def _forward(self, batch: Dict[str, torch.Tensor]) -> ModelOutput:
+        if self.use_bf16:
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                return self.model(batch)
+        return self.model(batch)
+
+    def train_step(self, batch: Dict[str, torch.Tensor]) -> float:
+        batch = {k: v.to(self.device, non_blocking=True) for k, v in batch.items()}
+        if self.dp is not None:
+            self.dp.zero_grad()
+        else:
+            self.optimizer.zero_grad(set_to_none=True)
+        out = self._forward(batch)
+        out.loss.backward()
+        if self.dp is not None:
+            self.dp.finalize_backward()      # wait bucketed all-reduces
+        clip_gradients(self.model, self.family)
+        self.step += 1
+        self.schedule.apply(self.optimizer, self.step)
+        self.optimizer.step()
+        return float(out.loss.detach())
+
+    def train(self, batches: Iterable[Dict[str, torch.Tensor]],
+              eval_fn=None, log_steps: int = 100, save_steps: int = 500,
+              early_stop_patience: Optional[int] = None,
+              max_steps: Optional[int] = None) -> Dict:
+        best_eval, since_best = float("inf"), 0
+        t0, last_log_step = time.time(), self.step
+        losses = []
+        for batch in batches:
+            loss = self.train_step(batch)
+            losses.append(loss)
+            if self.step % log_steps == 0 and self.rank == 0:
+                dt = time.time() - t0
+                steps = self.step - last_log_step
+                sps = steps * batch["token_ids"].shape[0] / max(dt, 1e-9)
+                log.info("step %d loss %.4f lr %.2e %.1f samples/s",
+                         self.step, sum(losses) / len(losses),
+                         self.schedule.lr_at(self.step), sps)
+                losses, t0, last_log_step = [], time.time(), self.step
+            if save_steps and self.step % save_steps == 0 and self.rank == 0:
+                self.ckpt.save(self.step, self.model, self.optimizer)
+                if eval_fn is not None:
+                    ev = eval_fn()
+                    if ev < best_eval:
+                        best_eval, since_best = ev, 0
+                    else:
+                        since_best += 1
+                        if (early_stop_patience is not None
+                                and since_best >= early_stop_patience):
+                            log.info("early stop at step %d", self.step)
+                            break
+            if max_steps is not None and self.step >= max_steps:
+                break
+        if self.rank == 0:
+            self.ckpt.save(self.step, self.model, self.optimizer)
+        return {"step": self.step, "best_eval": best_eval}
+
+    # -------------------------------------------------------------- eval
+    @torch.no_grad()
+    def evaluate(self, batches: Iterable[Dict[str, torch.Tensor]],
+                 idx2tag: Optional[Dict[int, str]] = None,
+                 label_size: Optional[int] = None) -> Dict[str, float]:
+        self.model.eval()
+        metrics = TagMetrics(label_size or self.params.get("label_size", 10),
+                             idx2tag or self.params.get("idx2tag"))
+        total_loss, n = 0.0, 0
+        for batch in batches:
+            batch = {k: v.to(self.device) for k, v in batch.items()}
+            if self.use_bf16:
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    out = self.model(batch, compute_pred=True)
+            else:
+                out = self.model(batch, compute_pred=True)
+            if out.loss is not None:
+                total_loss += float(out.loss)
+                n += 1
+            metrics.update(out.pred_ids, batch["label_ids"], batch["mask"])
+        self.model.train()
+        result = metrics.compute()
+        result["loss"] = total_loss / max(n, 1)
+        return result
+
+    # ----------------------------------------------------------- predict
+    @torch.no_grad()
+    def predict(self, batches: Iterable[Dict[str, torch.Tensor]]):
+        """Returns list of per-sample dicts (pred_ids, label_ids, mask) —
+        the reference pickles the Estimator.predict list (main.py:52-55)."""
+        self.model.eval()
+        out_rows = []
+        for batch in batches:
+            dev = {k: v.to(self.device) for k, v in batch.items()}
+            if self.use_bf16:
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    out = self.model(dev, compute_pred=True)
+            else:
+                out = self.model(dev, compute_pred=True)
+            pred = out.pred_ids.cpu()
+            for b in range(pred.shape[0]):
+                out_rows.append({
+                    "pred_ids": pred[b].numpy(),
+                    "label_ids": batch["label_ids"][b].numpy(),
+                    "mask": batch["mask"][b].numpy(),
+                })
+        self.model.train()
+        return out_rows
+
+    def dump_predictions(self, rows, data_dir: str, file_prefix: str) -> str:
+        path = os.path.join(data_dir, f"{file_prefix}_predict.pkl")
+        os.makedirs(data_dir, exist_ok=True)
+        with open(path, "wb") as f:
+            pickle.dump(rows, f)
+        return path

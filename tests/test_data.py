@@ -1,0 +1,140 @@
+"""Data-layer tests: tokenizers, trie, word-enhance features, preprocess
+shapes, loaders (reference test strategy upgraded per SURVEY.md §4)."""
+import numpy as np
+import pytest
+import torch
+
+from chinesener_amd.data import word_enhance as we
+from chinesener_amd.data.datasets import get_spec, synthetic_corpus
+from chinesener_amd.data.loader import MultiDataset, NerDataset, make_synthetic_batch
+from chinesener_amd.data.preprocess import extract_prefix_surfix, get_instance
+from chinesener_amd.data.tokenizer import (CharTokenizer, Vocab,
+                                           WordpieceTokenizer, full_to_half,
+                                           get_tokenizer)
+from chinesener_amd.data.trie import Trie
+from chinesener_amd.data.word_enhance import Lexicon
+
+
+def test_full_to_half():
+    assert full_to_half("Ａｂ１！") == "Ab1!"
+    assert full_to_half("　") == " "
+
+
+def test_trie_match():
+    t = Trie(["南京", "南京市", "市长", "长江", "长江大桥", "大桥"])
+    assert "南京" in t and "南京市" in t and "北京" not in t
+    assert t.prefixes("南京市长江大桥", 0) == ["南京", "南京市"]
+    spans = t.max_match_segment("南京市长江大桥")
+    words = ["南京市长江大桥"[s:e] for s, e in spans]
+    assert words == ["南京市", "长江大桥"]
+
+
+def test_char_tokenizer_roundtrip():
+    v = Vocab.synthetic(500)
+    tok = CharTokenizer(v)
+    ids = tok.convert_tokens_to_ids(tok.tokenize(v.itos[10] + v.itos[20]))
+    assert ids == [10, 20]
+
+
+def test_wordpiece_unknown():
+    v = Vocab.synthetic(200)
+    tok = WordpieceTokenizer(v)
+    toks = tok.tokenize("X" + v.itos[9])   # X not in CJK synthetic vocab
+    assert toks[0] == "[UNK]" and toks[1] == v.itos[9]
+
+
+def _lexicon():
+    chars = [chr(0x4E00 + i) for i in range(50)]
+    return Lexicon.synthetic(chars, n_words=100, dim=8, seed=7)
+
+
+def test_softword_bmes():
+    lex = _lexicon()
+    w = next(w for w in lex.words[3:] if len(w) == 3)
+    ids = we.build_softword(w, lex)
+    assert ids == [we.SOFT2IDX["B"], we.SOFT2IDX["M"], we.SOFT2IDX["E"]]
+
+
+def test_ex_softword_multihot():
+    lex = _lexicon()
+    w = next(w for w in lex.words[3:] if len(w) == 2)
+    hot = we.build_ex_softword(w, lex)
+    assert hot[0][we.SOFT2IDX["B"]] == 1
+    assert hot[1][we.SOFT2IDX["E"]] == 1
+    # char with no match gets None
+    hot2 = we.build_ex_softword("龠", lex)   # char outside lexicon range
+    assert hot2[0][we.SOFT2IDX["None"]] == 1
+
+
+def test_softlexicon_shapes_and_weights():
+    lex = _lexicon()
+    sent = lex.words[5] + lex.words[6]
+    ids, wts = we.build_soft_lexicon(sent, lex)
+    n = len(sent)
+    assert ids.shape == (n, 40) and wts.shape == (n, 40)
+    # weights normalized per char
+    np.testing.assert_allclose(wts.sum(1), np.ones(n), rtol=1e-5)
+
+
+def test_preprocess_bert_frame():
+    spec = get_spec("msra")
+    proc = get_instance("bert", 32, spec.tag2idx)
+    feat = proc.build_seq_feature("中国人民银行", ["B-ORG"] + ["I-ORG"] * 5)
+    assert feat["token_ids"].shape == (32,)
+    assert feat["seq_len"] == 8          # 6 chars + CLS/SEP
+    assert feat["label_ids"][0] == spec.tag2idx["[CLS]"]
+    assert feat["label_ids"][7] == spec.tag2idx["[SEP]"]
+    assert feat["mask"][:8].sum() == 8 and feat["mask"][8:].sum() == 0
+
+
+def test_preprocess_softlexicon_alignment():
+    spec = get_spec("msra")
+    proc = get_instance("bert", 32, spec.tag2idx, "softlexicon")
+    sent = "".join(proc.lexicon.words[5])
+    feat = proc.build_seq_feature(sent, ["O"] * len(sent))
+    # row 0 is CLS -> all-zero enhance row; row 1 aligns with sentence[0]
+    assert feat["softlexicon_ids"].shape == (32, 40)
+    assert feat["softlexicon_ids"][0].sum() == 0
+    assert feat["softlexicon_ids"][1].sum() > 0
+
+
+def test_extract_prefix_surfix():
+    assert extract_prefix_surfix("bert_bilstm_crf") == (None, "bert")
+    assert extract_prefix_surfix("bilstm_crf_softlexicon") == ("softlexicon", "char")
+    assert extract_prefix_surfix("bert_bilstm_crf_softlexicon") == ("softlexicon", "bert")
+    assert extract_prefix_surfix("transformer_crf_bichar") == ("bichar", "char")
+
+
+def test_synthetic_corpus_tags_valid():
+    spec = get_spec("msra")
+    sents, tags = synthetic_corpus(spec, "valid", n=20)
+    for s, t in zip(sents, tags):
+        assert len(s) == len(t)
+        assert all(tag in spec.tag2idx for tag in t)
+
+
+def test_ner_dataset_batches(tmp_path):
+    pipe = NerDataset(str(tmp_path), "people_daily", 4, 1, "bilstm_crf")
+    p = pipe.params
+    assert p["label_size"] == get_spec("people_daily").label_size
+    assert p["step_per_epoch"] > 0
+    batch = next(iter(pipe.iter_batches("train")))
+    assert batch["token_ids"].shape == (4, 150)
+    assert batch["label_ids"].dtype == torch.int64
+
+
+def test_multidataset_mix(tmp_path):
+    md = MultiDataset(str(tmp_path), ["msra", "msr"], 8, 1, "bert_bilstm_crf_mtl")
+    p = md.params
+    assert p["msra"]["label_size"] == 10 and p["msr"]["label_size"] == 7
+    batch = next(iter(md.iter_batches("train")))
+    task = batch["task_ids"][:, 0]
+    # strict alternation -> exactly half each
+    assert int((task == 0).sum()) == 4 and int((task == 1).sum()) == 4
+
+
+def test_make_synthetic_batch_softlexicon():
+    b = make_synthetic_batch(2, 16, word_enhance="softlexicon")
+    assert b["softlexicon_ids"].shape == (2, 16, 40)
+    s = b["softlexicon_weights"].sum(-1)
+    assert torch.allclose(s, torch.ones_like(s), atol=1e-5)

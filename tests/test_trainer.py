@@ -1,0 +1,131 @@
+"""Trainer loop, schedules, checkpoint/resume, CLI e2e (CPU)."""
+import itertools
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+
+from chinesener_amd.config import RUN_CONFIG, resolve_params
+from chinesener_amd.data.loader import NerDataset
+from chinesener_amd.models import build_model, model_params, optimizer_family
+from chinesener_amd.train.optimizers import (AdamWeightDecay, LrSchedule,
+                                             build_optimizer, build_param_groups)
+from chinesener_amd.train.trainer import Trainer
+
+from conftest import make_tiny_batch, make_tiny_params
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_optimizer_family_dispatch():
+    assert optimizer_family("bert_bilstm_crf") == "bert"
+    assert optimizer_family("transformer_tener_crf_bichar") == "transformer"
+    assert optimizer_family("bilstm_crf") == "custom"
+
+
+def test_lr_schedule_shapes():
+    s = LrSchedule("bert", 1e-4, num_train_steps=100, warmup_ratio=0.1)
+    assert s.lr_at(5) < s.lr_at(10) - 1e-12          # warming up
+    assert s.lr_at(10) == pytest.approx(1e-4)        # peak at warmup end
+    assert s.lr_at(100) < 1e-5                        # decayed
+    noam = LrSchedule("transformer", 1.0, num_train_steps=1000, warmup_steps=40)
+    assert noam.lr_at(10) < noam.lr_at(40) and noam.lr_at(400) < noam.lr_at(40)
+    exp = LrSchedule("custom", 1e-3, num_train_steps=100, step_per_epoch=10,
+                     decay_rate=0.5)
+    assert exp.lr_at(5) == pytest.approx(1e-3)
+    assert exp.lr_at(25) == pytest.approx(2.5e-4)
+
+
+def test_diff_lr_groups():
+    params = make_tiny_params("bert_bilstm_crf")
+    model = build_model("bert_bilstm_crf", params)
+    groups = build_param_groups(model, 5e-5, 0.01,
+                                {"crf": 500, "logit": 500, "bilstm": 100})
+    scales = {g["lr_scale"] for g in groups}
+    assert {1.0, 500.0, 100.0} <= scales
+    # no-decay groups exist (LayerNorm/bias)
+    assert any(g["weight_decay"] == 0.0 for g in groups)
+    assert any(g["weight_decay"] > 0.0 for g in groups)
+
+
+def test_adamw_decreases_loss():
+    torch.manual_seed(0)
+    w = torch.nn.Parameter(torch.randn(10))
+    opt = AdamWeightDecay([{"params": [w], "lr_scale": 1.0,
+                            "weight_decay": 0.0}], lr=0.1)
+    target = torch.randn(10)
+    first = None
+    for _ in range(50):
+        opt.zero_grad()
+        loss = ((w - target) ** 2).sum()
+        if first is None:
+            first = float(loss)
+        loss.backward()
+        opt.step()
+    assert float(loss) < first * 0.1
+
+
+def test_trainer_steps_and_resume(tmp_path):
+    torch.manual_seed(0)
+    name = "bilstm_crf"
+    params = make_tiny_params(name)
+    params.update(num_train_steps=50, step_per_epoch=10, lr=1e-2)
+    model = build_model(name, params)
+    tr = Trainer(model, name, params, str(tmp_path / "ckpt"), device="cpu")
+    batches = [make_tiny_batch(name, seed=i) for i in range(12)]
+    tr.train(iter(batches), log_steps=1000, save_steps=5)
+    assert tr.step == 12
+    # resume picks up the saved step
+    model2 = build_model(name, params)
+    tr2 = Trainer(model2, name, params, str(tmp_path / "ckpt"), device="cpu")
+    assert tr2.step >= 10
+    for p1, p2 in zip(model.parameters(), model2.parameters()):
+        if tr2.step == tr.step:
+            torch.testing.assert_close(p1, p2)
+
+
+def test_trainer_loss_decreases_overfit():
+    torch.manual_seed(0)
+    name = "bilstm_crf"
+    params = make_tiny_params(name)
+    params.update(num_train_steps=200, step_per_epoch=50, lr=5e-2)
+    model = build_model(name, params)
+    import tempfile
+    with tempfile.TemporaryDirectory() as d:
+        tr = Trainer(model, name, params, d, device="cpu")
+        batch = make_tiny_batch(name, batch_size=4, seed=0)
+        losses = [tr.train_step(batch) for _ in range(60)]
+    assert losses[-1] < losses[0] * 0.5, losses[::10]
+
+
+def test_predict_dump_roundtrip(tmp_path):
+    name = "bilstm_crf"
+    params = make_tiny_params(name)
+    model = build_model(name, params)
+    tr = Trainer(model, name, params, str(tmp_path / "c"), device="cpu")
+    rows = tr.predict([make_tiny_batch(name, seed=s) for s in range(2)])
+    path = tr.dump_predictions(rows, str(tmp_path), "bilstm_crf")
+    import pickle
+    with open(path, "rb") as f:
+        loaded = pickle.load(f)
+    assert len(loaded) == 6 and "pred_ids" in loaded[0]
+
+
+def test_main_cli_e2e(tmp_path):
+    """Full CLI slice: train 4 steps of bilstm_crf on synthetic people_daily
+    (BASELINE config 1 plumbing)."""
+    env = dict(os.environ, PYTHONPATH=REPO)
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "main.py"),
+         "--model_name", "bilstm_crf", "--data", "people_daily",
+         "--epochs", "1", "--batch_size", "8", "--max_steps", "4",
+         "--data_dir", str(tmp_path / "data"),
+         "--ckpt_root", str(tmp_path / "ckpt")],
+        capture_output=True, text=True, env=env, timeout=600,
+        cwd=str(tmp_path))
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert os.path.exists(tmp_path / "ckpt" / "ner_people_daily_bilstm_crf")
+    assert os.path.exists(tmp_path / "data" / "people_daily" /
+                          "bilstm_crf_predict.pkl")
