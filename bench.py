@@ -1,0 +1,147 @@
+#!/usr/bin/env python3
+"""Flagship training benchmark (driver contract).
+
+Measures training samples/sec of bert_bilstm_crf (BERT-base encoder +
+BiLSTM + CRF, the BASELINE.json headline config) on MSRA-shaped
+synthetic data, bf16, random-init weights. Weak scaling: per-GPU batch
+fixed; `value` is the whole-job aggregate samples/sec.
+
+  python bench.py --gpus N --steps K --warmup W
+  (N>1 is launched by the driver via torch.distributed.run, one rank/GPU)
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from chinesener_amd.config import resolve_params
+from chinesener_amd.data.loader import make_synthetic_batch
+from chinesener_amd.models import build_model, model_params, optimizer_family
+from chinesener_amd.train.optimizers import build_optimizer, clip_gradients
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--batch_size", type=int, default=64,
+                    help="per-GPU batch (weak scaling; global=batch*gpus)")
+    ap.add_argument("--seq_len", type=int, default=128)
+    ap.add_argument("--model", default="bert_bilstm_crf")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    use_gpu = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if use_gpu else "cpu"
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        if use_gpu:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl" if use_gpu else "gloo")
+
+    torch.manual_seed(1234 + rank)
+    # MSRA-shaped config: vocab 21128, 10 BIO labels, BERT-base encoder
+    params = resolve_params(model_params(args.model), {
+        "vocab_size": 21128, "label_size": 10,
+        "num_train_steps": 100000, "step_per_epoch": 1000,
+    })
+    model = build_model(args.model, params).to(device)
+    dp = None
+    if world > 1:
+        from chinesener_amd.dist import BucketedDataParallel
+        dp = BucketedDataParallel(model)
+    family = optimizer_family(args.model)
+    opt, schedule = build_optimizer(model, family, params)
+
+    from chinesener_amd.data.preprocess import extract_prefix_surfix
+    enhance, _ = extract_prefix_surfix(args.model)
+    batches = [make_synthetic_batch(args.batch_size, args.seq_len, 10,
+                                    word_enhance=enhance, seed=rank * 100 + i,
+                                    device=device)
+               for i in range(8)]
+
+    use_bf16 = use_gpu and params.get("dtype", "bf16") == "bf16"
+    step_num = 0
+
+    def train_step(batch):
+        nonlocal step_num
+        if dp is not None:
+            dp.zero_grad()
+        else:
+            opt.zero_grad(set_to_none=True)
+        if use_bf16:
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                out = model(batch)
+        else:
+            out = model(batch)
+        out.loss.backward()
+        if dp is not None:
+            dp.finalize_backward()
+        clip_gradients(model, family)
+        step_num += 1
+        schedule.apply(opt, step_num)
+        opt.step()
+        return out.loss
+
+    for i in range(args.warmup):
+        train_step(batches[i % len(batches)])
+    if dist is not None:
+        dist.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        train_step(batches[i % len(batches)])
+    if dist is not None:
+        dist.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    # max over ranks
+    if dist is not None:
+        t = torch.tensor([elapsed], device=device if use_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        total_samples = args.steps * args.batch_size * world
+        result = {
+            "metric": f"training samples/sec, {args.model} (BERT-base + BiLSTM + CRF), msra-shaped"
+                      if args.model == "bert_bilstm_crf"
+                      else f"training samples/sec, {args.model}, msra-shaped",
+            "value": round(total_samples / elapsed, 2),
+            "unit": "samples/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_bf16 else "fp32",
+            "data": "synthetic",
+            "config": {"model": args.model,
+                       "global_batch": args.batch_size * world,
+                       "seq_len": args.seq_len,
+                       "parallelism": f"dp{world}"},
+        }
+        print(json.dumps(result))
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

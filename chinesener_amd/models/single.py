@@ -96,10 +96,12 @@ class BilstmCrfSoftlexicon(BilstmCrf):
         return 4 * params.get("word_dim", 50)
 
     def _enhance(self, features, emb):
-        fused = ops.softlexicon_fuse(self.word_emb.emb.weight.to(emb.dtype),
+        # table stays fp32 (HIP kernel contract); fuse output cast to the
+        # compute dtype afterwards
+        fused = ops.softlexicon_fuse(self.word_emb.emb.weight,
                                      features["softlexicon_ids"],
                                      features["softlexicon_weights"])
-        return torch.cat([emb, fused], -1)
+        return torch.cat([emb, fused.to(emb.dtype)], -1)
 
 
 class BilstmCrfBichar(BilstmCrf):
@@ -274,7 +276,8 @@ class BertBilstmCrfSoftlexicon(BertCrf):
 
     def encode_hidden(self, features):
         seq = self.encode(features)
-        fused = ops.softlexicon_fuse(self.word_emb.emb.weight.to(seq.dtype),
+        fused = ops.softlexicon_fuse(self.word_emb.emb.weight,
                                      features["softlexicon_ids"],
                                      features["softlexicon_weights"])
-        return self.bilstm(torch.cat([seq, fused], -1), _lens(features))
+        return self.bilstm(torch.cat([seq, fused.to(seq.dtype)], -1),
+                           _lens(features))

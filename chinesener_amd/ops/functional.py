@@ -10,9 +10,15 @@ import math
 from typing import Optional, Tuple
 
 import torch
+import torch.nn.functional as F
 
 from . import reference as ref
 from . import hip_enabled, get_ext
+
+
+def _pad_last(t: torch.Tensor, target: int) -> torch.Tensor:
+    d = t.shape[-1]
+    return t if d == target else F.pad(t, (0, target - d))
 
 
 # ----------------------------------------------------------- layer norm
@@ -105,43 +111,65 @@ class _AttentionFn(torch.autograd.Function):
 
 def attention(q, k, v, mask: Optional[torch.Tensor] = None,
               scale: Optional[float] = None, lens: Optional[torch.Tensor] = None):
-    """Fused attention. q,k,v [B,H,L,D]; mask [B,L] prefix mask or lens [B]."""
+    """Fused attention. q,k,v [B,H,L,D]; mask [B,L] prefix mask or lens [B].
+
+    HIP path: bf16, head dim padded to 32/64, L <= 176 (covers every
+    reference config; falls back to the torch path otherwise)."""
+    D = q.shape[-1]
     if scale is None:
-        scale = 1.0 / math.sqrt(q.shape[-1])
-    if hip_enabled(q):
+        scale = 1.0 / math.sqrt(D)
+    if (hip_enabled(q) and q.dtype == torch.bfloat16 and D <= 64
+            and q.shape[2] <= 176):
+        Dp = 32 if D <= 32 else 64
         if lens is None:
             lens = (mask.long().sum(1) if mask is not None
                     else torch.full((q.shape[0],), q.shape[2],
                                     dtype=torch.long, device=q.device))
-        return _AttentionFn.apply(q.contiguous(), k.contiguous(), v.contiguous(),
-                                  lens.to(torch.int32), float(scale))
+        out = _AttentionFn.apply(_pad_last(q, Dp).contiguous(),
+                                 _pad_last(k, Dp).contiguous(),
+                                 _pad_last(v, Dp).contiguous(),
+                                 lens.to(torch.int32), float(scale))
+        return out[..., :D] if Dp != D else out
     return ref.attention(q, k, v, mask, scale)
 
 
 class _TenerAttentionFn(torch.autograd.Function):
+    """Kernel contract: qu = q+u, qv = q+v precomputed (u/v grads are
+    reductions of dqu/dqv handled by autograd outside this Function)."""
+
     @staticmethod
-    def forward(ctx, q, k, v, u, vb, rel, lens):
-        out, lse = get_ext().tener_attn_fwd(q, k, v, u, vb, rel, lens)
-        ctx.save_for_backward(q, k, v, u, vb, rel, out, lse, lens)
+    def forward(ctx, qu, qv, k, v, rel, lens):
+        out, lse = get_ext().tener_attn_fwd(qu, qv, k, v, rel, lens)
+        ctx.save_for_backward(qu, qv, k, v, rel, out, lse, lens)
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        q, k, v, u, vb, rel, out, lse, lens = ctx.saved_tensors
-        dq, dk, dv, du, dvb = get_ext().tener_attn_bwd(
-            dout.contiguous(), q, k, v, u, vb, rel, out, lse, lens)
-        return dq, dk, dv, du, dvb, None, None
+        qu, qv, k, v, rel, out, lse, lens = ctx.saved_tensors
+        dqu, dqv, dk, dv = get_ext().tener_attn_bwd(
+            dout.contiguous(), qu, qv, k, v, rel, out, lse, lens)
+        return dqu, dqv, dk, dv, None, None
 
 
 def tener_attention(q, k, v, u, vb, rel, mask: Optional[torch.Tensor] = None):
-    """TENER relative attention (unscaled, no key projection upstream)."""
-    if hip_enabled(q):
+    """TENER relative attention (unscaled, no key projection upstream).
+
+    HIP path: bf16, head dim padded to 32, L <= 160."""
+    D = q.shape[-1]
+    if (hip_enabled(q) and q.dtype == torch.bfloat16 and D <= 32
+            and q.shape[2] <= 160):
         lens = (mask.long().sum(1) if mask is not None
                 else torch.full((q.shape[0],), q.shape[2], dtype=torch.long,
                                 device=q.device))
-        return _TenerAttentionFn.apply(
-            q.contiguous(), k.contiguous(), v.contiguous(), u.contiguous(),
-            vb.contiguous(), rel.contiguous(), lens.to(torch.int32))
+        qu = _pad_last(q + u[None, :, None, :].to(q.dtype), 32)
+        qv = _pad_last(q + vb[None, :, None, :].to(q.dtype), 32)
+        out = _TenerAttentionFn.apply(
+            qu.contiguous(), qv.contiguous(),
+            _pad_last(k, 32).to(torch.bfloat16).contiguous(),
+            _pad_last(v, 32).to(torch.bfloat16).contiguous(),
+            _pad_last(rel, 32).to(torch.bfloat16).contiguous(),
+            lens.to(torch.int32))
+        return out[..., :D] if D != 32 else out
     return ref.tener_attention(q, k, v, u, vb, rel, mask)
 
 
@@ -197,9 +225,18 @@ class _LstmDirFn(torch.autograd.Function):
     def backward(ctx, dhs):
         hs, cs, gates, w_hh, lens = ctx.saved_tensors
         reverse, act_relu = ctx.flags
-        dgates_x, dw_hh = get_ext().lstm_bwd(dhs.contiguous(), hs, cs, gates,
-                                             w_hh, lens, reverse, act_relu)
-        return dgates_x, dw_hh, None, None, None
+        (dgates_x,) = get_ext().lstm_bwd(dhs.contiguous(), hs, cs, gates,
+                                         w_hh, lens, reverse, act_relu)
+        # dW_hh = sum_t h_{t-1}^T dgates_t — one library GEMM. hs holds the
+        # carried state at valid steps and 0 elsewhere; dgates is 0 at
+        # invalid steps, so the shifted product is exact.
+        B, L, h = hs.shape
+        if reverse:
+            h_prev = torch.cat([hs[:, 1:], torch.zeros_like(hs[:, :1])], dim=1)
+        else:
+            h_prev = torch.cat([torch.zeros_like(hs[:, :1]), hs[:, :-1]], dim=1)
+        dw_hh = h_prev.reshape(-1, h).T @ dgates_x.reshape(-1, 4 * h)
+        return dgates_x, dw_hh.to(w_hh.dtype), None, None, None
 
 
 def _lstm_dir(x, w_ih, w_hh, b, lens, reverse, activation):
@@ -213,13 +250,28 @@ def _lstm_dir(x, w_ih, w_hh, b, lens, reverse, activation):
 
 def bilstm(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b, lens,
            activation: str = "tanh", state_dropout=None):
-    """BiLSTM over padded [B,L,E] -> [B,L,2h]."""
-    if hip_enabled(x):
+    """BiLSTM over padded [B,L,E] -> [B,L,2h].
+
+    HIP path: hidden % 32 == 0 and hidden <= 128 (LDS-resident W_hh);
+    other sizes run the torch recurrence (slow path, logged once)."""
+    h = w_hh_f.shape[0]
+    if hip_enabled(x) and h % 32 == 0 and h <= 128:
         fw = _lstm_dir(x, w_ih_f, w_hh_f, b_f, lens, False, activation)
         bw = _lstm_dir(x, w_ih_b, w_hh_b, b_b, lens, True, activation)
         return torch.cat([fw, bw], dim=-1)
+    if x.is_cuda:
+        global _LSTM_FALLBACK_WARNED
+        if not _LSTM_FALLBACK_WARNED:
+            import logging
+            logging.getLogger("chinesener_amd").warning(
+                "BiLSTM hidden=%d not kernel-eligible (needs %%32==0, <=128);"
+                " using torch recurrence", h)
+            _LSTM_FALLBACK_WARNED = True
     return ref.bilstm_forward(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b,
                               lens, activation, state_dropout)
+
+
+_LSTM_FALLBACK_WARNED = False
 
 
 # ----------------------------------------------------------- softlexicon

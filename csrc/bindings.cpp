@@ -1,0 +1,81 @@
+// Python bindings for the gfx950 kernel library (_hip_ops).
+#include <torch/extension.h>
+
+#include <vector>
+
+// elementwise.hip
+std::vector<at::Tensor> layernorm_fwd(const at::Tensor&, const at::Tensor&,
+                                      const at::Tensor&, double);
+std::vector<at::Tensor> add_layernorm_fwd(const at::Tensor&, const at::Tensor&,
+                                          const at::Tensor&, const at::Tensor&,
+                                          double);
+std::vector<at::Tensor> layernorm_bwd(const at::Tensor&, const at::Tensor&,
+                                      const at::Tensor&, const at::Tensor&,
+                                      const at::Tensor&);
+at::Tensor bias_gelu_fwd(const at::Tensor&, const at::Tensor&);
+std::vector<at::Tensor> bias_gelu_bwd(const at::Tensor&, const at::Tensor&,
+                                      const at::Tensor&);
+std::vector<at::Tensor> masked_ce_fwd(const at::Tensor&, const at::Tensor&,
+                                      const at::Tensor&);
+at::Tensor masked_ce_bwd(const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&);
+// crf.hip
+std::vector<at::Tensor> crf_fwd(const at::Tensor&, const at::Tensor&,
+                                const at::Tensor&, const at::Tensor&);
+at::Tensor crf_viterbi(const at::Tensor&, const at::Tensor&,
+                       const at::Tensor&);
+// softlexicon.hip
+at::Tensor softlexicon_fwd(const at::Tensor&, const at::Tensor&,
+                           const at::Tensor&);
+std::vector<at::Tensor> softlexicon_bwd(const at::Tensor&, const at::Tensor&,
+                                        const at::Tensor&, const at::Tensor&);
+// adam.hip
+void multi_tensor_adamw(std::vector<at::Tensor>, std::vector<at::Tensor>,
+                        std::vector<at::Tensor>, std::vector<at::Tensor>,
+                        double, double, double, double, double);
+// attention.hip
+std::vector<at::Tensor> attn_fwd(const at::Tensor&, const at::Tensor&,
+                                 const at::Tensor&, const at::Tensor&, double);
+std::vector<at::Tensor> attn_bwd(const at::Tensor&, const at::Tensor&,
+                                 const at::Tensor&, const at::Tensor&,
+                                 const at::Tensor&, const at::Tensor&,
+                                 const at::Tensor&, double);
+// tener.hip
+std::vector<at::Tensor> tener_attn_fwd(const at::Tensor&, const at::Tensor&,
+                                       const at::Tensor&, const at::Tensor&,
+                                       const at::Tensor&, const at::Tensor&);
+std::vector<at::Tensor> tener_attn_bwd(const at::Tensor&, const at::Tensor&,
+                                       const at::Tensor&, const at::Tensor&,
+                                       const at::Tensor&, const at::Tensor&,
+                                       const at::Tensor&, const at::Tensor&,
+                                       const at::Tensor&);
+// lstm.hip
+std::vector<at::Tensor> lstm_fwd(const at::Tensor&, const at::Tensor&,
+                                 const at::Tensor&, bool, bool);
+std::vector<at::Tensor> lstm_bwd(const at::Tensor&, const at::Tensor&,
+                                 const at::Tensor&, const at::Tensor&,
+                                 const at::Tensor&, const at::Tensor&, bool,
+                                 bool);
+
+PYBIND11_MODULE(_hip_ops, m) {
+  m.doc() = "chinesener_amd gfx950 HIP kernels";
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("add_layernorm_fwd", &add_layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("bias_gelu_fwd", &bias_gelu_fwd);
+  m.def("bias_gelu_bwd", &bias_gelu_bwd);
+  m.def("masked_ce_fwd", &masked_ce_fwd);
+  m.def("masked_ce_bwd", &masked_ce_bwd);
+  m.def("crf_fwd", &crf_fwd);
+  m.def("crf_viterbi", &crf_viterbi);
+  m.def("softlexicon_fwd", &softlexicon_fwd);
+  m.def("softlexicon_bwd", &softlexicon_bwd);
+  m.def("multi_tensor_adamw", &multi_tensor_adamw);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
+  m.def("tener_attn_fwd", &tener_attn_fwd);
+  m.def("tener_attn_bwd", &tener_attn_bwd);
+  m.def("lstm_fwd", &lstm_fwd);
+  m.def("lstm_bwd", &lstm_bwd);
+}

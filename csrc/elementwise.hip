@@ -1,0 +1,443 @@
+// Fused LayerNorm (fwd/bwd), residual-add+LayerNorm, bias+GELU, masked CE.
+// SURVEY.md kernels K10/K11/K13: memory-bound rowwise ops, vectorized
+// 8-wide bf16 loads (guide G13), fp32 statistics, one wave per row.
+#include "common.h"
+
+// ---------------------------------------------------------------------
+// LayerNorm forward: y = (x - mean) * rstd * w + b  (stats in fp32)
+// one wave per row; H % 8 == 0
+// ---------------------------------------------------------------------
+template <typename T>
+__global__ void layernorm_fwd_kernel(const T* __restrict__ x,
+                                     const T* __restrict__ res,  // optional
+                                     const float* __restrict__ w,
+                                     const float* __restrict__ b,
+                                     T* __restrict__ y,
+                                     T* __restrict__ sum_out,  // optional x+res
+                                     float* __restrict__ mean,
+                                     float* __restrict__ rstd,
+                                     int N, int H, float eps) {
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long row = (long)blockIdx.x * (blockDim.x / WAVE) + wid;
+  if (row >= N) return;
+  const T* xr = x + row * H;
+  const T* rr = res ? res + row * H : nullptr;
+  float acc = 0.f, acc2 = 0.f;
+  const int HV = H / 8;
+  // static trip count keeps vals in registers (guide §5.4 rule 20)
+  constexpr int MAXIT = 6;  // H <= 3072
+  float vals[MAXIT][8];
+#pragma unroll
+  for (int it = 0; it < MAXIT; ++it) {
+    const int i = lane + it * WAVE;
+    if (i >= HV) break;
+    const s16x8 raw = reinterpret_cast<const s16x8*>(xr)[i];
+    s16x8 raw2{};
+    if (rr) raw2 = reinterpret_cast<const s16x8*>(rr)[i];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float v = to_f32(reinterpret_cast<const T*>(&raw)[e]);
+      if (rr) v += to_f32(reinterpret_cast<const T*>(&raw2)[e]);
+      vals[it][e] = v;
+      acc += v;
+      acc2 += v * v;
+    }
+  }
+  acc = wave_reduce_sum(acc);
+  acc2 = wave_reduce_sum(acc2);
+  const float m = acc / H;
+  const float rs = rsqrtf(fmaxf(acc2 / H - m * m, 0.f) + eps);
+  if (lane == 0) {
+    mean[row] = m;
+    rstd[row] = rs;
+  }
+  T* yr = y + row * H;
+  T* sr = sum_out ? sum_out + row * H : nullptr;
+#pragma unroll
+  for (int it = 0; it < MAXIT; ++it) {
+    const int i = lane + it * WAVE;
+    if (i >= HV) break;
+    T out[8], sout[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int col = i * 8 + e;
+      const float v = vals[it][e];
+      from_f32((v - m) * rs * w[col] + b[col], &out[e]);
+      if (sr) from_f32(v, &sout[e]);
+    }
+    reinterpret_cast<s16x8*>(yr)[i] = *reinterpret_cast<s16x8*>(out);
+    if (sr) reinterpret_cast<s16x8*>(sr)[i] = *reinterpret_cast<s16x8*>(sout);
+  }
+}
+
+// fp32 specialization uses wider registers; keep one code path by
+// reinterpreting through s16x8 only for 16-bit T. Provide a scalar body
+// for float.
+__global__ void layernorm_fwd_kernel_f32(const float* __restrict__ x,
+                                         const float* __restrict__ res,
+                                         const float* __restrict__ w,
+                                         const float* __restrict__ b,
+                                         float* __restrict__ y,
+                                         float* __restrict__ sum_out,
+                                         float* __restrict__ mean,
+                                         float* __restrict__ rstd,
+                                         int N, int H, float eps) {
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long row = (long)blockIdx.x * (blockDim.x / WAVE) + wid;
+  if (row >= N) return;
+  const float* xr = x + row * H;
+  const float* rr = res ? res + row * H : nullptr;
+  float acc = 0.f, acc2 = 0.f;
+  for (int i = lane; i < H; i += WAVE) {
+    float v = xr[i] + (rr ? rr[i] : 0.f);
+    acc += v;
+    acc2 += v * v;
+  }
+  acc = wave_reduce_sum(acc);
+  acc2 = wave_reduce_sum(acc2);
+  const float m = acc / H;
+  const float rs = rsqrtf(fmaxf(acc2 / H - m * m, 0.f) + eps);
+  if (lane == 0) {
+    mean[row] = m;
+    rstd[row] = rs;
+  }
+  for (int i = lane; i < H; i += WAVE) {
+    float v = xr[i] + (rr ? rr[i] : 0.f);
+    if (sum_out) sum_out[row * H + i] = v;
+    y[row * H + i] = (v - m) * rs * w[i] + b[i];
+  }
+}
+
+// ---------------------------------------------------------------------
+// LayerNorm backward.
+// dx = rs * (dy*w - mean(dy*w) - xhat * mean(dy*w*xhat))
+// dw += dy * xhat ; db += dy   (block-local LDS, then global atomics)
+// ---------------------------------------------------------------------
+template <typename T>
+__global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
+                                     const T* __restrict__ x,
+                                     const float* __restrict__ w,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ rstd,
+                                     T* __restrict__ dx,
+                                     float* __restrict__ dw,
+                                     float* __restrict__ db,
+                                     int N, int H) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* dw_s = reinterpret_cast<float*>(smem_raw);
+  float* db_s = dw_s + H;
+  for (int i = threadIdx.x; i < 2 * H; i += blockDim.x) dw_s[i] = 0.f;
+  __syncthreads();
+
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int rows_per_blk = blockDim.x / WAVE;
+  const long row = (long)blockIdx.x * rows_per_blk + wid;
+  if (row < N) {
+    const T* dyr = dy + row * H;
+    const T* xr = x + row * H;
+    const float m = mean[row], rs = rstd[row];
+    float c1 = 0.f, c2 = 0.f;
+    for (int i = lane; i < H; i += WAVE) {
+      const float g = to_f32(dyr[i]) * w[i];
+      const float xhat = (to_f32(xr[i]) - m) * rs;
+      c1 += g * xhat;
+      c2 += g;
+    }
+    c1 = wave_reduce_sum(c1) / H;
+    c2 = wave_reduce_sum(c2) / H;
+    T* dxr = dx + row * H;
+    for (int i = lane; i < H; i += WAVE) {
+      const float g = to_f32(dyr[i]) * w[i];
+      const float d = to_f32(dyr[i]);
+      const float xhat = (to_f32(xr[i]) - m) * rs;
+      from_f32((g - c2 - xhat * c1) * rs, &dxr[i]);
+      atomicAdd(&dw_s[i], d * xhat);
+      atomicAdd(&db_s[i], d);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < H; i += blockDim.x) {
+    atomicAdd(&dw[i], dw_s[i]);
+    atomicAdd(&db[i], db_s[i]);
+  }
+}
+
+// ---------------------------------------------------------------------
+// bias + GELU (erf form, BERT): y = 0.5*(x+b)*(1+erf((x+b)/sqrt(2)))
+// ---------------------------------------------------------------------
+__device__ __forceinline__ float gelu_f(float u) {
+  return 0.5f * u * (1.f + erff(u * 0.70710678118654752f));
+}
+__device__ __forceinline__ float dgelu_f(float u) {
+  const float cdf = 0.5f * (1.f + erff(u * 0.70710678118654752f));
+  const float pdf = 0.3989422804014327f * __expf(-0.5f * u * u);
+  return cdf + u * pdf;
+}
+
+template <typename T, bool BWD>
+__global__ void bias_gelu_kernel(const T* __restrict__ x,
+                                 const float* __restrict__ bias,
+                                 const T* __restrict__ dy,  // BWD only
+                                 T* __restrict__ out, long n, int F) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long nv = n / 8;
+  for (long v = i; v < nv; v += (long)gridDim.x * blockDim.x) {
+    const s16x8 raw = reinterpret_cast<const s16x8*>(x)[v];
+    s16x8 draw{};
+    if (BWD) draw = reinterpret_cast<const s16x8*>(dy)[v];
+    const int col0 = (int)((v * 8) % F);
+    T o[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float u = to_f32(reinterpret_cast<const T*>(&raw)[e]) + bias[col0 + e];
+      if (BWD) {
+        const float g = to_f32(reinterpret_cast<const T*>(&draw)[e]);
+        from_f32(g * dgelu_f(u), &o[e]);
+      } else {
+        from_f32(gelu_f(u), &o[e]);
+      }
+    }
+    reinterpret_cast<s16x8*>(out)[v] = *reinterpret_cast<s16x8*>(o);
+  }
+}
+
+__global__ void bias_gelu_kernel_f32(const float* __restrict__ x,
+                                     const float* __restrict__ bias,
+                                     const float* __restrict__ dy, bool bwd,
+                                     float* __restrict__ out, long n, int F) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (long v = i; v < n; v += (long)gridDim.x * blockDim.x) {
+    const float u = x[v] + bias[v % F];
+    out[v] = bwd ? dy[v] * dgelu_f(u) : gelu_f(u);
+  }
+}
+
+// ---------------------------------------------------------------------
+// masked softmax-CE over [N, T] logits (fp32), one thread per row.
+// fwd also writes probs for bwd. loss = sum(per-row) / nvalid.
+// ---------------------------------------------------------------------
+__global__ void masked_ce_fwd_kernel(const float* __restrict__ logits,
+                                     const int* __restrict__ labels,
+                                     const int* __restrict__ mask,
+                                     float* __restrict__ probs,
+                                     float* __restrict__ loss_sum,
+                                     int* __restrict__ nvalid, long N, int T) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* red = reinterpret_cast<float*>(smem_raw);
+  const long r = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  float my_loss = 0.f;
+  int my_valid = 0;
+  if (r < N) {
+    const float* lr = logits + r * T;
+    float mx = -1e30f;
+    for (int t = 0; t < T; ++t) mx = fmaxf(mx, lr[t]);
+    float s = 0.f;
+    for (int t = 0; t < T; ++t) s += __expf(lr[t] - mx);
+    const float lse = mx + __logf(s);
+    for (int t = 0; t < T; ++t) probs[r * T + t] = __expf(lr[t] - lse);
+    if (mask[r]) {
+      my_loss = lse - lr[labels[r]];
+      my_valid = 1;
+    }
+  }
+  const float tot = block_reduce_sum(my_loss, red);
+  __syncthreads();
+  const float nv = block_reduce_sum((float)my_valid, red);
+  if (threadIdx.x == 0) {
+    atomicAdd(loss_sum, tot);
+    atomicAdd(nvalid, (int)nv);
+  }
+}
+
+__global__ void masked_ce_bwd_kernel(const float* __restrict__ probs,
+                                     const int* __restrict__ labels,
+                                     const int* __restrict__ mask,
+                                     const float* __restrict__ dloss,
+                                     const int* __restrict__ nvalid,
+                                     float* __restrict__ dlogits, long N, int T) {
+  const long r = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (r >= N) return;
+  const float scale = mask[r] ? dloss[0] / (float)max(*nvalid, 1) : 0.f;
+  for (int t = 0; t < T; ++t) {
+    const float oh = (t == labels[r]) ? 1.f : 0.f;
+    dlogits[r * T + t] = (probs[r * T + t] - oh) * scale;
+  }
+}
+
+// =====================================================================
+// host wrappers
+// =====================================================================
+static std::vector<at::Tensor> layernorm_fwd_impl(const at::Tensor& x,
+                                                  const c10::optional<at::Tensor>& res,
+                                                  const at::Tensor& w,
+                                                  const at::Tensor& b,
+                                                  double eps, bool want_sum) {
+  CHECK_CUDA_CONTIG(x);
+  const long N = x.size(0);
+  const int H = x.size(1);
+  TORCH_CHECK(H % 8 == 0, "layernorm: H % 8 != 0");
+  auto y = at::empty_like(x);
+  auto sum = want_sum ? at::empty_like(x) : at::Tensor();
+  auto mean = at::empty({N}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({N}, x.options().dtype(at::kFloat));
+  auto wf = w.to(at::kFloat).contiguous();
+  auto bf = b.to(at::kFloat).contiguous();
+  const int rows_per_blk = 4;
+  const dim3 grid((N + rows_per_blk - 1) / rows_per_blk);
+  const dim3 block(rows_per_blk * WAVE);
+  auto stream = cur_stream(x);
+  if (x.scalar_type() == at::kBFloat16) {
+    TORCH_CHECK(H <= 3072, "layernorm bf16 kernel: H too large");
+    hipLaunchKernelGGL(layernorm_fwd_kernel<bf16>, grid, block, 0, stream,
+                       (const bf16*)x.data_ptr(),
+                       res ? (const bf16*)res->data_ptr() : nullptr,
+                       wf.data_ptr<float>(), bf.data_ptr<float>(),
+                       (bf16*)y.data_ptr(),
+                       want_sum ? (bf16*)sum.data_ptr() : nullptr,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(), N, H,
+                       (float)eps);
+  } else {
+    hipLaunchKernelGGL(layernorm_fwd_kernel_f32, grid, block, 0, stream,
+                       x.data_ptr<float>(),
+                       res ? res->data_ptr<float>() : nullptr,
+                       wf.data_ptr<float>(), bf.data_ptr<float>(),
+                       y.data_ptr<float>(),
+                       want_sum ? sum.data_ptr<float>() : nullptr,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(), N, H,
+                       (float)eps);
+  }
+  HIP_CHECK_LAST();
+  if (want_sum) return {y, sum, mean, rstd};
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x, const at::Tensor& w,
+                                      const at::Tensor& b, double eps) {
+  return layernorm_fwd_impl(x, c10::nullopt, w, b, eps, false);
+}
+
+std::vector<at::Tensor> add_layernorm_fwd(const at::Tensor& x,
+                                          const at::Tensor& res,
+                                          const at::Tensor& w,
+                                          const at::Tensor& b, double eps) {
+  CHECK_CUDA_CONTIG(res);
+  return layernorm_fwd_impl(x, res, w, b, eps, true);
+}
+
+std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                      const at::Tensor& w, const at::Tensor& mean,
+                                      const at::Tensor& rstd) {
+  CHECK_CUDA_CONTIG(dy);
+  CHECK_CUDA_CONTIG(x);
+  const long N = x.size(0);
+  const int H = x.size(1);
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto db = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto wf = w.to(at::kFloat).contiguous();
+  const int rows_per_blk = 4;
+  const dim3 grid((N + rows_per_blk - 1) / rows_per_blk);
+  const dim3 block(rows_per_blk * WAVE);
+  const size_t smem = 2 * H * sizeof(float);
+  auto stream = cur_stream(x);
+  if (x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(layernorm_bwd_kernel<bf16>, grid, block, smem, stream,
+                       (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
+                       wf.data_ptr<float>(), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), (bf16*)dx.data_ptr(),
+                       dw.data_ptr<float>(), db.data_ptr<float>(), N, H);
+  } else {
+    hipLaunchKernelGGL(layernorm_bwd_kernel<float>, grid, block, smem, stream,
+                       dy.data_ptr<float>(), x.data_ptr<float>(),
+                       wf.data_ptr<float>(), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), dx.data_ptr<float>(),
+                       dw.data_ptr<float>(), db.data_ptr<float>(), N, H);
+  }
+  HIP_CHECK_LAST();
+  return {dx, dw.to(w.scalar_type()), db.to(w.scalar_type())};
+}
+
+static at::Tensor bias_gelu_impl(const at::Tensor& x, const at::Tensor& bias,
+                                 const c10::optional<at::Tensor>& dy) {
+  CHECK_CUDA_CONTIG(x);
+  const long n = x.numel();
+  const int F = x.size(-1);
+  TORCH_CHECK(F % 8 == 0, "bias_gelu: F % 8 != 0");
+  auto out = at::empty_like(x);
+  auto bf = bias.to(at::kFloat).contiguous();
+  auto stream = cur_stream(x);
+  const int block = 256;
+  if (x.scalar_type() == at::kBFloat16) {
+    const long grid = std::min<long>((n / 8 + block - 1) / block, 4096);
+    if (dy) {
+      hipLaunchKernelGGL((bias_gelu_kernel<bf16, true>), dim3(grid), dim3(block),
+                         0, stream, (const bf16*)x.data_ptr(),
+                         bf.data_ptr<float>(), (const bf16*)dy->data_ptr(),
+                         (bf16*)out.data_ptr(), n, F);
+    } else {
+      hipLaunchKernelGGL((bias_gelu_kernel<bf16, false>), dim3(grid), dim3(block),
+                         0, stream, (const bf16*)x.data_ptr(),
+                         bf.data_ptr<float>(), nullptr,
+                         (bf16*)out.data_ptr(), n, F);
+    }
+  } else {
+    const long grid = std::min<long>((n + block - 1) / block, 4096);
+    hipLaunchKernelGGL(bias_gelu_kernel_f32, dim3(grid), dim3(block), 0, stream,
+                       x.data_ptr<float>(), bf.data_ptr<float>(),
+                       dy ? dy->data_ptr<float>() : nullptr, (bool)dy,
+                       out.data_ptr<float>(), n, F);
+  }
+  HIP_CHECK_LAST();
+  return out;
+}
+
+at::Tensor bias_gelu_fwd(const at::Tensor& x, const at::Tensor& bias) {
+  return bias_gelu_impl(x, bias, c10::nullopt);
+}
+
+std::vector<at::Tensor> bias_gelu_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                      const at::Tensor& bias) {
+  auto dx = bias_gelu_impl(x, bias, dy);
+  auto dbias = dx.to(at::kFloat).sum({0}).to(bias.scalar_type());
+  return {dx, dbias};
+}
+
+std::vector<at::Tensor> masked_ce_fwd(const at::Tensor& logits,
+                                      const at::Tensor& labels,
+                                      const at::Tensor& mask) {
+  CHECK_CUDA_CONTIG(logits);
+  const long N = logits.size(0);
+  const int T = logits.size(1);
+  auto probs = at::empty_like(logits);
+  auto loss = at::zeros({}, logits.options());
+  auto nvalid = at::zeros({}, logits.options().dtype(at::kInt));
+  const int block = 256;
+  const long grid = (N + block - 1) / block;
+  hipLaunchKernelGGL(masked_ce_fwd_kernel, dim3(grid), dim3(block),
+                     (block / WAVE) * sizeof(float) * 2, cur_stream(logits),
+                     logits.data_ptr<float>(), labels.data_ptr<int>(),
+                     mask.data_ptr<int>(), probs.data_ptr<float>(),
+                     loss.data_ptr<float>(), nvalid.data_ptr<int>(), N, T);
+  HIP_CHECK_LAST();
+  auto mean = loss / at::clamp_min(nvalid, 1).to(at::kFloat);
+  return {mean, probs, nvalid};
+}
+
+at::Tensor masked_ce_bwd(const at::Tensor& dloss, const at::Tensor& probs,
+                         const at::Tensor& labels, const at::Tensor& mask,
+                         const at::Tensor& nvalid) {
+  const long N = probs.size(0);
+  const int T = probs.size(1);
+  auto dlogits = at::empty_like(probs);
+  const int block = 256;
+  hipLaunchKernelGGL(masked_ce_bwd_kernel, dim3((N + block - 1) / block),
+                     dim3(block), 0, cur_stream(probs), probs.data_ptr<float>(),
+                     labels.data_ptr<int>(), mask.data_ptr<int>(),
+                     dloss.to(at::kFloat).contiguous().data_ptr<float>(),
+                     nvalid.data_ptr<int>(), dlogits.data_ptr<float>(), N, T);
+  HIP_CHECK_LAST();
+  return dlogits;
+}

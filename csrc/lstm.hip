@@ -1,0 +1,307 @@
+// Persistent LSTM recurrence (the "BiLSTM tagger" kernel
+// named in BASELINE.json). The x-projection (x @ W_ih + b) is a single
+// library GEMM done by the Python wrapper; these kernels own the
+// sequential part: per step  gates = gates_x[t] + h_{t-1} @ W_hh,
+// LSTM cell update, variable-length masking.
+//
+// Design: recurrent weights stay resident in LDS for the whole sequence
+// (W_hh bf16 <= 128 KiB for h <= 128); one workgroup (4 waves) owns a
+// 16-row batch tile and loops time in-kernel — no per-step launches.
+// MFMA 16x16x32 bf16 computes [16,4h] gates; wave w owns hidden slice
+// j in [w*32,(w+1)*32) so each lane locally combines its i/f/g/o gates
+// and carries the cell state in registers.
+//
+// fwd stores activated gates + cell states (fp32) for backward; bwd
+// replays in reverse, producing pre-activation gate grads (dgates_x);
+// dW_hh / dW_ih / db / dx are library GEMMs in the wrapper.
+#include "common.h"
+
+using bfrag = mfma_bf16x8;
+using cfrag = mfma_f32x4;
+
+__device__ __forceinline__ bfrag lds_frag_l(const bf16* base, int i0, int ld,
+                                            int k0) {
+  const int l = threadIdx.x & (WAVE - 1);
+  return *reinterpret_cast<const bfrag*>(base + (long)(i0 + (l & 15)) * ld +
+                                         k0 + ((l >> 4) << 3));
+}
+
+__device__ __forceinline__ float act_f(float x, bool relu) {
+  return relu ? fmaxf(x, 0.f) : tanhf(x);
+}
+__device__ __forceinline__ float dact_from_out(float y, bool relu) {
+  // derivative expressed from the activated value y = act(x)
+  return relu ? (y > 0.f ? 1.f : 0.f) : (1.f - y * y);
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void lstm_fwd_kernel(
+    const T* __restrict__ gates_x,  // [B,L,4h]
+    const bf16* __restrict__ w_hh_t,  // [4h,h] (transposed by wrapper)
+    const int* __restrict__ lens, T* __restrict__ hs,  // [B,L,h]
+    float* __restrict__ cs,                            // [B,L,h]
+    float* __restrict__ gates_out,                     // [B,L,4h] activated
+    int B, int L, int h, bool reverse, bool relu) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  bf16* wT = reinterpret_cast<bf16*>(smem_raw);  // [4h][h]
+  bf16* hb = wT + 4 * h * h;                     // [16][h] current h (bf16)
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int b0 = blockIdx.x * 16;
+
+  for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x)
+    reinterpret_cast<s16x8*>(wT)[i] = reinterpret_cast<const s16x8*>(w_hh_t)[i];
+  for (int i = threadIdx.x; i < 16 * h / 8; i += blockDim.x)
+    reinterpret_cast<s16x8*>(hb)[i] = s16x8{};
+  __syncthreads();
+
+  const int NKK = h / 32;
+  // lane-owned cell state: rows r=0..3 (row = (lane>>4)*4+r), cols jj=0..1
+  // (j = wid*32 + jj*16 + (lane&15))
+  float c_reg[4][2] = {};
+  const int lrow = ((lane >> 4) << 2);  // +r
+  int mylen[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int b = b0 + lrow + r;
+    mylen[r] = (b < B) ? lens[b] : 0;
+  }
+
+  for (int step = 0; step < L; ++step) {
+    const int t = reverse ? (L - 1 - step) : step;
+    // gates = h_prev @ W_hh  (+ gates_x added in the epilogue)
+    cfrag acc[4][2];
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+#pragma unroll
+      for (int jj = 0; jj < 2; ++jj) acc[g][jj] = cfrag{0.f, 0.f, 0.f, 0.f};
+    for (int kk = 0; kk < NKK; ++kk) {
+      const bfrag ah = lds_frag_l(hb, 0, h, kk * 32);
+#pragma unroll
+      for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int jj = 0; jj < 2; ++jj) {
+          if (wid * 32 + jj * 16 >= h) continue;  // h < 128: idle slices
+          const int col0 = g * h + wid * 32 + jj * 16;
+          acc[g][jj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ah, lds_frag_l(wT, col0, h, kk * 32), acc[g][jj], 0, 0, 0);
+        }
+    }
+    __syncthreads();  // hb reads done; safe to overwrite below
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int b = b0 + lrow + r;
+      if (b >= B) continue;
+      const bool valid = t < mylen[r];
+      const long gbase = ((long)b * L + t) * 4 * h;
+#pragma unroll
+      for (int jj = 0; jj < 2; ++jj) {
+        const int j = wid * 32 + jj * 16 + (lane & 15);
+        if (j >= h) continue;
+        float gi = acc[0][jj][r] + to_f32(gates_x[gbase + 0 * h + j]);
+        float gf = acc[1][jj][r] + to_f32(gates_x[gbase + 1 * h + j]);
+        float gg = acc[2][jj][r] + to_f32(gates_x[gbase + 2 * h + j]);
+        float go = acc[3][jj][r] + to_f32(gates_x[gbase + 3 * h + j]);
+        gi = 1.f / (1.f + __expf(-gi));
+        gf = 1.f / (1.f + __expf(-gf));
+        go = 1.f / (1.f + __expf(-go));
+        gg = act_f(gg, relu);
+        const float c_new = gf * c_reg[r][jj] + gi * gg;
+        const float h_new = go * act_f(c_new, relu);
+        if (valid) c_reg[r][jj] = c_new;
+        const long obase = ((long)b * L + t) * h + j;
+        from_f32(valid ? h_new : 0.f, &hs[obase]);
+        cs[obase] = c_reg[r][jj];
+        gates_out[gbase + 0 * h + j] = gi;
+        gates_out[gbase + 1 * h + j] = gf;
+        gates_out[gbase + 2 * h + j] = gg;
+        gates_out[gbase + 3 * h + j] = go;
+        hb[(lrow + r) * h + j] =
+            __float2bfloat16(valid ? h_new : to_f32(hb[(lrow + r) * h + j]));
+      }
+    }
+    __syncthreads();  // hb updated for next step
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void lstm_bwd_kernel(
+    const T* __restrict__ dhs,        // [B,L,h] upstream grad
+    const float* __restrict__ cs,     // [B,L,h] carried cell states
+    const float* __restrict__ gates,  // [B,L,4h] activated
+    const bf16* __restrict__ w_hh,    // [h,4h] (original layout)
+    const int* __restrict__ lens, T* __restrict__ dgates_x,  // [B,L,4h]
+    int B, int L, int h, bool reverse, bool relu) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  bf16* w_s = reinterpret_cast<bf16*>(smem_raw);  // [h][4h]
+  bf16* dg_s = w_s + 4 * h * h;                   // [16][4h]
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int b0 = blockIdx.x * 16;
+  for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x)
+    reinterpret_cast<s16x8*>(w_s)[i] = reinterpret_cast<const s16x8*>(w_hh)[i];
+  __syncthreads();
+
+  const int lrow = ((lane >> 4) << 2);
+  float dc_reg[4][2] = {};
+  float dh_reg[4][2] = {};
+  int mylen[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int b = b0 + lrow + r;
+    mylen[r] = (b < B) ? lens[b] : 0;
+  }
+
+  for (int step = L - 1; step >= 0; --step) {
+    const int t = reverse ? (L - 1 - step) : step;  // reverse of fwd order
+    const int t_prev = reverse ? (t + 1) : (t - 1);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int b = b0 + lrow + r;
+      const bool inb = b < B;
+      const bool valid = inb && (t < mylen[r]);
+      const long gbase = inb ? ((long)b * L + t) * 4 * h : 0;
+#pragma unroll
+      for (int jj = 0; jj < 2; ++jj) {
+        const int j = wid * 32 + jj * 16 + (lane & 15);
+        if (j >= h) continue;
+        float dgi = 0.f, dgf = 0.f, dgg = 0.f, dgo = 0.f;
+        if (valid) {
+          const long obase = ((long)b * L + t) * h + j;
+          const float gi = gates[gbase + 0 * h + j];
+          const float gf = gates[gbase + 1 * h + j];
+          const float gg = gates[gbase + 2 * h + j];
+          const float go = gates[gbase + 3 * h + j];
+          const float c_t = cs[obase];
+          const float c_prev =
+              (t_prev >= 0 && t_prev < L && t_prev < mylen[r])
+                  ? cs[((long)b * L + t_prev) * h + j]
+                  : 0.f;
+          const float ac = act_f(c_t, relu);
+          const float dh = dh_reg[r][jj] + to_f32(dhs[obase]);
+          float dc = dc_reg[r][jj] + dh * go * dact_from_out(ac, relu);
+          dgo = dh * ac * go * (1.f - go);
+          dgi = dc * gg * gi * (1.f - gi);
+          dgf = dc * c_prev * gf * (1.f - gf);
+          dgg = dc * gi * dact_from_out(gg, relu);
+          dc_reg[r][jj] = dc * gf;  // carry to previous step
+        }
+        // write pre-activation gate grads (global + LDS for the MFMA)
+        if (inb) {
+          from_f32(dgi, &dgates_x[gbase + 0 * h + j]);
+          from_f32(dgf, &dgates_x[gbase + 1 * h + j]);
+          from_f32(dgg, &dgates_x[gbase + 2 * h + j]);
+          from_f32(dgo, &dgates_x[gbase + 3 * h + j]);
+        }
+        const int lr = lrow + r;
+        dg_s[lr * 4 * h + 0 * h + j] = __float2bfloat16(dgi);
+        dg_s[lr * 4 * h + 1 * h + j] = __float2bfloat16(dgf);
+        dg_s[lr * 4 * h + 2 * h + j] = __float2bfloat16(dgg);
+        dg_s[lr * 4 * h + 3 * h + j] = __float2bfloat16(dgo);
+      }
+    }
+    __syncthreads();
+    // dh_prev = dgates @ W_hh^T : [16,4h] @ [4h,h]; wave w owns output
+    // cols [w*32,(w+1)*32) = N-frags {2w, 2w+1}
+    cfrag acc[2];
+#pragma unroll
+    for (int jj = 0; jj < 2; ++jj) acc[jj] = cfrag{0.f, 0.f, 0.f, 0.f};
+    for (int kk = 0; kk < 4 * h / 32; ++kk) {
+      const bfrag adg = lds_frag_l(dg_s, 0, 4 * h, kk * 32);
+#pragma unroll
+      for (int jj = 0; jj < 2; ++jj) {
+        const int col0 = wid * 32 + jj * 16;
+        if (col0 >= h) continue;
+        acc[jj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            adg, lds_frag_l(w_s, col0, 4 * h, kk * 32), acc[jj], 0, 0, 0);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const bool valid = t < mylen[r];
+#pragma unroll
+      for (int jj = 0; jj < 2; ++jj)
+        if (valid) dh_reg[r][jj] = acc[jj][r];
+      // invalid step: h passed through unchanged -> dh carries unchanged
+    }
+    __syncthreads();  // dg_s reads done before next step overwrites
+  }
+}
+
+// ===================================================================== host
+static void lstm_checks(const at::Tensor& gates_x, int& B, int& L, int& h4) {
+  CHECK_CUDA_CONTIG(gates_x);
+  B = gates_x.size(0);
+  L = gates_x.size(1);
+  h4 = gates_x.size(2);
+  const int h = h4 / 4;
+  TORCH_CHECK(h % 32 == 0 && h <= 128,
+              "lstm kernel: hidden must be a multiple of 32 and <= 128 "
+              "(wrapper falls back otherwise), got ", h);
+}
+
+std::vector<at::Tensor> lstm_fwd(const at::Tensor& gates_x,
+                                 const at::Tensor& w_hh,
+                                 const at::Tensor& lens, bool reverse,
+                                 bool relu) {
+  int B, L, h4;
+  lstm_checks(gates_x, B, L, h4);
+  const int h = h4 / 4;
+  auto hs = at::empty({B, L, h}, gates_x.options());
+  auto cs = at::empty({B, L, h}, gates_x.options().dtype(at::kFloat));
+  auto gates = at::empty({B, L, h4}, gates_x.options().dtype(at::kFloat));
+  auto w_t = w_hh.t().contiguous().to(at::kBFloat16);  // [4h,h]
+  const size_t smem = (size_t)(4 * h * h + 16 * h) * sizeof(bf16);
+  TORCH_CHECK(smem <= 150 * 1024, "lstm fwd LDS overflow");
+  const int grid = (B + 15) / 16;
+  auto stream = cur_stream(gates_x);
+  if (gates_x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(lstm_fwd_kernel<bf16>, dim3(grid), dim3(256), smem,
+                       stream, (const bf16*)gates_x.data_ptr(),
+                       (const bf16*)w_t.data_ptr(), lens.data_ptr<int>(),
+                       (bf16*)hs.data_ptr(), cs.data_ptr<float>(),
+                       gates.data_ptr<float>(), B, L, h, reverse, relu);
+  } else {
+    hipLaunchKernelGGL(lstm_fwd_kernel<float>, dim3(grid), dim3(256), smem,
+                       stream, gates_x.data_ptr<float>(),
+                       (const bf16*)w_t.data_ptr(), lens.data_ptr<int>(),
+                       hs.data_ptr<float>(), cs.data_ptr<float>(),
+                       gates.data_ptr<float>(), B, L, h, reverse, relu);
+  }
+  HIP_CHECK_LAST();
+  return {hs, cs, gates};
+}
+
+std::vector<at::Tensor> lstm_bwd(const at::Tensor& dhs, const at::Tensor& hs,
+                                 const at::Tensor& cs, const at::Tensor& gates,
+                                 const at::Tensor& w_hh, const at::Tensor& lens,
+                                 bool reverse, bool relu) {
+  int B, L, h4;
+  B = dhs.size(0);
+  L = dhs.size(1);
+  const int h = dhs.size(2);
+  h4 = 4 * h;
+  auto dgates_x = at::empty({B, L, h4}, dhs.options());
+  auto w_b = w_hh.contiguous().to(at::kBFloat16);  // [h,4h]
+  const size_t smem = (size_t)(4 * h * h + 16 * 4 * h) * sizeof(bf16);
+  TORCH_CHECK(smem <= 150 * 1024, "lstm bwd LDS overflow");
+  const int grid = (B + 15) / 16;
+  auto stream = cur_stream(dhs);
+  auto dhs_c = dhs.contiguous();
+  if (dhs.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(lstm_bwd_kernel<bf16>, dim3(grid), dim3(256), smem,
+                       stream, (const bf16*)dhs_c.data_ptr(),
+                       cs.data_ptr<float>(), gates.data_ptr<float>(),
+                       (const bf16*)w_b.data_ptr(), lens.data_ptr<int>(),
+                       (bf16*)dgates_x.data_ptr(), B, L, h, reverse, relu);
+  } else {
+    hipLaunchKernelGGL(lstm_bwd_kernel<float>, dim3(grid), dim3(256), smem,
+                       stream, dhs_c.data_ptr<float>(),
+                       cs.data_ptr<float>(), gates.data_ptr<float>(),
+                       (const bf16*)w_b.data_ptr(), lens.data_ptr<int>(),
+                       dgates_x.data_ptr<float>(), B, L, h, reverse, relu);
+  }
+  HIP_CHECK_LAST();
+  // dW_hh etc. are library GEMMs in the python wrapper (needs h_{t-1})
+  return {dgates_x};
+}

@@ -1,0 +1,308 @@
+"""GPU numerics: every HIP kernel vs the pure-torch fp32 reference
+(SURVEY.md §4 implication (a)). bf16 kernels get bf16-scale tolerances;
+fp32 kernels (CRF, softlexicon, CE) get tight ones. All inputs are
+random (transpose-detecting, guide §5.4 rule 16/25)."""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from chinesener_amd import ops
+from chinesener_amd.ops import reference as ref
+
+
+def _cuda():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    assert ops.ext_available(), "HIP extension must be built for GPU tests"
+
+
+# ------------------------------------------------------------ layernorm
+def test_layernorm_fwd_bwd():
+    _cuda()
+    torch.manual_seed(0)
+    x32 = torch.randn(64, 768, device="cuda", requires_grad=True)
+    w = torch.randn(768, device="cuda", requires_grad=True)
+    b = torch.randn(768, device="cuda", requires_grad=True)
+    x16 = x32.detach().to(torch.bfloat16).requires_grad_()
+    w2 = w.detach().clone().requires_grad_()
+    b2 = b.detach().clone().requires_grad_()
+
+    y_ref = ref.layernorm(x32, w, b)
+    y = ops.layernorm(x16, w2, b2)
+    torch.testing.assert_close(y.float(), y_ref, atol=0.05, rtol=0.05)
+
+    g = torch.randn_like(y_ref)
+    y_ref.backward(g)
+    y.backward(g.to(torch.bfloat16))
+    torch.testing.assert_close(x16.grad.float(), x32.grad, atol=0.1, rtol=0.1)
+    torch.testing.assert_close(w2.grad, w.grad, atol=0.5, rtol=0.05)
+    torch.testing.assert_close(b2.grad, b.grad, atol=0.5, rtol=0.05)
+
+
+def test_add_layernorm():
+    _cuda()
+    torch.manual_seed(1)
+    x = torch.randn(32, 768, device="cuda")
+    r = torch.randn(32, 768, device="cuda")
+    w = torch.randn(768, device="cuda")
+    b = torch.randn(768, device="cuda")
+    y = ops.add_layernorm(x.to(torch.bfloat16), r.to(torch.bfloat16), w, b)
+    y_ref = ref.add_layernorm(x, r, w, b)
+    torch.testing.assert_close(y.float(), y_ref, atol=0.06, rtol=0.05)
+
+
+# ------------------------------------------------------------ bias gelu
+def test_bias_gelu_fwd_bwd():
+    _cuda()
+    torch.manual_seed(2)
+    x32 = torch.randn(128, 3072, device="cuda", requires_grad=True)
+    bias = torch.randn(3072, device="cuda", requires_grad=True)
+    x16 = x32.detach().to(torch.bfloat16).requires_grad_()
+    b2 = bias.detach().clone().requires_grad_()
+    y_ref = ref.bias_gelu(x32, bias)
+    y = ops.bias_gelu(x16, b2)
+    torch.testing.assert_close(y.float(), y_ref, atol=0.03, rtol=0.05)
+    g = torch.randn_like(y_ref)
+    y_ref.backward(g)
+    y.backward(g.to(torch.bfloat16))
+    torch.testing.assert_close(x16.grad.float(), x32.grad, atol=0.05, rtol=0.05)
+    torch.testing.assert_close(b2.grad, bias.grad, atol=1.0, rtol=0.05)
+
+
+# ------------------------------------------------------------ attention
+@pytest.mark.parametrize("L,D", [(128, 64), (150, 64), (64, 32)])
+def test_attention_fwd_bwd(L, D):
+    _cuda()
+    torch.manual_seed(3)
+    B, H = 2, 4
+    q32 = torch.randn(B, H, L, D, device="cuda", requires_grad=True)
+    k32 = torch.randn(B, H, L, D, device="cuda", requires_grad=True)
+    v32 = torch.randn(B, H, L, D, device="cuda", requires_grad=True)
+    lens = torch.tensor([L, max(2, L - 41)], device="cuda")
+    mask = (torch.arange(L, device="cuda")[None, :] < lens[:, None]).long()
+
+    out_ref = ref.attention(q32, k32, v32, mask)
+    q16 = q32.detach().to(torch.bfloat16).requires_grad_()
+    k16 = k32.detach().to(torch.bfloat16).requires_grad_()
+    v16 = v32.detach().to(torch.bfloat16).requires_grad_()
+    out = ops.attention(q16, k16, v16, mask=mask)
+    # compare only real query rows (padded rows differ harmlessly)
+    mrow = mask[:, None, :, None].bool()
+    torch.testing.assert_close((out.float() * mrow), (out_ref * mrow),
+                               atol=0.06, rtol=0.05)
+
+    g = torch.randn_like(out_ref) * mask[:, None, :, None]
+    out_ref.backward(g)
+    out.backward(g.to(torch.bfloat16))
+    torch.testing.assert_close(q16.grad.float(), q32.grad, atol=0.15, rtol=0.1)
+    torch.testing.assert_close(k16.grad.float(), k32.grad, atol=0.15, rtol=0.1)
+    torch.testing.assert_close(v16.grad.float(), v32.grad, atol=0.15, rtol=0.1)
+
+
+def test_attention_identity_asymmetric():
+    """A=I-style check with asymmetric V: lens full, Q=K orthonormal-ish
+    rows — catches transposed fragment layouts (guide: always asymmetric)."""
+    _cuda()
+    B, H, L, D = 1, 1, 32, 64
+    q = torch.zeros(B, H, L, D, device="cuda")
+    for i in range(L):
+        q[0, 0, i, i % D] = 10.0
+    k = q.clone() * 2
+    v = torch.arange(L * D, device="cuda").float().reshape(1, 1, L, D) / (L * D)
+    mask = torch.ones(B, L, device="cuda").long()
+    out = ops.attention(q.to(torch.bfloat16), k.to(torch.bfloat16),
+                        v.to(torch.bfloat16), mask=mask)
+    out_ref = ref.attention(q, k, v, mask)
+    torch.testing.assert_close(out.float(), out_ref, atol=0.02, rtol=0.05)
+
+
+# ---------------------------------------------------------------- tener
+def test_tener_fwd_bwd():
+    _cuda()
+    torch.manual_seed(4)
+    B, H, L, D = 2, 2, 150, 20
+    q32 = torch.randn(B, H, L, D, device="cuda", requires_grad=True) * 0.5
+    k32 = torch.randn(B, H, L, D, device="cuda", requires_grad=True) * 0.5
+    v32 = torch.randn(B, H, L, D, device="cuda", requires_grad=True)
+    u32 = torch.randn(H, D, device="cuda", requires_grad=True) * 0.1
+    vb32 = torch.randn(H, D, device="cuda", requires_grad=True) * 0.1
+    rel = ref.relative_table(L, D, device="cuda")
+    lens = torch.tensor([L, 97], device="cuda")
+    mask = (torch.arange(L, device="cuda")[None, :] < lens[:, None]).long()
+
+    out_ref = ref.tener_attention(q32, k32, v32, u32, vb32, rel, mask)
+    q16 = q32.detach().to(torch.bfloat16).requires_grad_()
+    k16 = k32.detach().to(torch.bfloat16).requires_grad_()
+    v16 = v32.detach().to(torch.bfloat16).requires_grad_()
+    u16 = u32.detach().clone().requires_grad_()
+    vb16 = vb32.detach().clone().requires_grad_()
+    out = ops.tener_attention(q16, k16, v16, u16, vb16, rel, mask)
+    mrow = mask[:, None, :, None].bool()
+    torch.testing.assert_close(out.float() * mrow, out_ref * mrow,
+                               atol=0.12, rtol=0.1)
+    g = torch.randn_like(out_ref) * mask[:, None, :, None]
+    out_ref.backward(g)
+    out.backward(g.to(torch.bfloat16))
+    torch.testing.assert_close(q16.grad.float(), q32.grad, atol=0.25, rtol=0.15)
+    torch.testing.assert_close(k16.grad.float(), k32.grad, atol=0.25, rtol=0.15)
+    torch.testing.assert_close(v16.grad.float(), v32.grad, atol=0.25, rtol=0.15)
+    torch.testing.assert_close(u16.grad, u32.grad, atol=1.0, rtol=0.1)
+    torch.testing.assert_close(vb16.grad, vb32.grad, atol=1.0, rtol=0.1)
+
+
+# ------------------------------------------------------------------ crf
+def test_crf_nll_and_grads():
+    _cuda()
+    torch.manual_seed(5)
+    B, L, T = 8, 40, 10
+    em32 = torch.randn(B, L, T, device="cuda", requires_grad=True)
+    trans32 = torch.randn(T, T, device="cuda", requires_grad=True)
+    lens = torch.randint(2, L + 1, (B,), device="cuda")
+    mask = (torch.arange(L, device="cuda")[None, :] < lens[:, None]).long()
+    tags = torch.randint(0, T, (B, L), device="cuda") * mask
+
+    ll_ref = ref.crf_log_likelihood(em32, tags, mask, trans32)
+    em2 = em32.detach().clone().requires_grad_()
+    tr2 = trans32.detach().clone().requires_grad_()
+    ll = ops.crf_nll(em2, tags, mask, tr2)
+    torch.testing.assert_close(ll, ll_ref, atol=1e-3, rtol=1e-4)
+
+    w = torch.randn(B, device="cuda")
+    (ll_ref * w).sum().backward()
+    (ll * w).sum().backward()
+    torch.testing.assert_close(em2.grad, em32.grad, atol=1e-3, rtol=1e-3)
+    torch.testing.assert_close(tr2.grad, trans32.grad, atol=1e-2, rtol=1e-3)
+
+
+def test_crf_viterbi_matches_reference():
+    _cuda()
+    torch.manual_seed(6)
+    B, L, T = 16, 50, 7
+    em = torch.randn(B, L, T, device="cuda")
+    trans = torch.randn(T, T, device="cuda")
+    lens = torch.randint(1, L + 1, (B,), device="cuda")
+    mask = (torch.arange(L, device="cuda")[None, :] < lens[:, None]).long()
+    pred = ops.crf_viterbi(em, mask, trans)
+    pred_ref = ref.crf_decode(em, mask, trans)
+    assert torch.equal(pred.cpu(), pred_ref.cpu())
+
+
+# ----------------------------------------------------------- softlexicon
+def test_softlexicon_fwd_bwd():
+    _cuda()
+    torch.manual_seed(7)
+    V, E, B, L = 100, 50, 4, 20
+    table32 = torch.randn(V, E, device="cuda", requires_grad=True)
+    ids = torch.randint(0, V, (B, L, 40), device="cuda")
+    w32 = torch.rand(B, L, 40, device="cuda", requires_grad=True)
+    out_ref = ref.softlexicon_fuse(table32, ids, w32)
+    t2 = table32.detach().clone().requires_grad_()
+    w2 = w32.detach().clone().requires_grad_()
+    out = ops.softlexicon_fuse(t2, ids, w2)
+    torch.testing.assert_close(out, out_ref, atol=1e-4, rtol=1e-4)
+    g = torch.randn_like(out_ref)
+    out_ref.backward(g)
+    out.backward(g)
+    torch.testing.assert_close(t2.grad, table32.grad, atol=1e-3, rtol=1e-3)
+    torch.testing.assert_close(w2.grad, w32.grad, atol=1e-3, rtol=1e-3)
+
+
+# ---------------------------------------------------------------- losses
+def test_masked_ce_gpu():
+    _cuda()
+    torch.manual_seed(8)
+    B, L, T = 4, 30, 10
+    logits32 = torch.randn(B, L, T, device="cuda", requires_grad=True)
+    labels = torch.randint(0, T, (B, L), device="cuda")
+    mask = (torch.rand(B, L, device="cuda") > 0.3).long()
+    loss_ref = ref.masked_cross_entropy(logits32, labels, mask)
+    l2 = logits32.detach().clone().requires_grad_()
+    loss = ops.masked_cross_entropy(l2, labels, mask)
+    torch.testing.assert_close(loss, loss_ref, atol=1e-5, rtol=1e-5)
+    loss_ref.backward()
+    loss.backward()
+    torch.testing.assert_close(l2.grad, logits32.grad, atol=1e-5, rtol=1e-4)
+
+
+# ------------------------------------------------------------------ lstm
+@pytest.mark.parametrize("activation,h", [("tanh", 32), ("relu", 128)])
+def test_lstm_dir_vs_reference(activation, h):
+    _cuda()
+    torch.manual_seed(9)
+    B, L, E = 5, 24, 16
+    x = torch.randn(B, L, E, device="cuda") * 0.5
+    w_ih = torch.randn(E, 4 * h, device="cuda") * 0.2
+    w_hh = torch.randn(h, 4 * h, device="cuda") * 0.2
+    bias = torch.randn(4 * h, device="cuda") * 0.1
+    lens = torch.tensor([24, 20, 1, 24, 7], device="cuda")
+
+    out_ref = ref.lstm_forward(x, w_ih, w_hh, bias, lens, False, activation)
+    from chinesener_amd.ops.functional import _lstm_dir
+    x16 = x.to(torch.bfloat16)
+    out = _lstm_dir(x16, w_ih, w_hh, bias, lens, False, activation)
+    torch.testing.assert_close(out.float(), out_ref, atol=0.08, rtol=0.1)
+
+
+@pytest.mark.parametrize("reverse", [False, True])
+def test_bilstm_grads_vs_reference(reverse):
+    _cuda()
+    torch.manual_seed(10)
+    B, L, E, h = 4, 16, 8, 32
+    x32 = (torch.randn(B, L, E, device="cuda") * 0.5).requires_grad_()
+    w_ih32 = (torch.randn(E, 4 * h, device="cuda") * 0.2).requires_grad_()
+    w_hh32 = (torch.randn(h, 4 * h, device="cuda") * 0.2).requires_grad_()
+    b32 = (torch.randn(4 * h, device="cuda") * 0.1).requires_grad_()
+    lens = torch.tensor([16, 12, 5, 16], device="cuda")
+
+    out_ref = ref.lstm_forward(x32, w_ih32, w_hh32, b32, lens, reverse, "tanh")
+    loss_ref = (out_ref ** 2).sum()
+    loss_ref.backward()
+
+    from chinesener_amd.ops.functional import _lstm_dir
+    x2 = x32.detach().to(torch.bfloat16).requires_grad_()
+    wih2 = w_ih32.detach().clone().requires_grad_()
+    whh2 = w_hh32.detach().clone().requires_grad_()
+    b2 = b32.detach().clone().requires_grad_()
+    out = _lstm_dir(x2, wih2, whh2, b2, lens, reverse, "tanh")
+    (out.float() ** 2).sum().backward()
+    torch.testing.assert_close(x2.grad.float(), x32.grad, atol=0.3, rtol=0.15)
+    torch.testing.assert_close(whh2.grad, w_hh32.grad, atol=0.5, rtol=0.15)
+    torch.testing.assert_close(wih2.grad, w_ih32.grad, atol=0.5, rtol=0.15)
+
+
+# ------------------------------------------------------------------ adam
+def test_multi_tensor_adamw_matches_cpu_math():
+    _cuda()
+    torch.manual_seed(11)
+    shapes = [(100,), (32, 64), (7, 9)]
+    params = [torch.randn(s, device="cuda") for s in shapes]
+    grads = [torch.randn(s, device="cuda") for s in shapes]
+    ms = [torch.rand(s, device="cuda") * 0.1 for s in shapes]
+    vs = [torch.rand(s, device="cuda") * 0.01 for s in shapes]
+    ref_p = [p.clone() for p in params]
+    ref_m = [m.clone() for m in ms]
+    ref_v = [v.clone() for v in vs]
+    lr, b1, b2, eps, wd = 1e-3, 0.9, 0.999, 1e-6, 0.01
+    ops.get_ext().multi_tensor_adamw(params, grads, ms, vs, lr, b1, b2, eps, wd)
+    for p, g, m, v in zip(ref_p, grads, ref_m, ref_v):
+        m.mul_(b1).add_(g, alpha=1 - b1)
+        v.mul_(b2).addcmul_(g, g, value=1 - b2)
+        p.add_(-lr * (m / (v.sqrt() + eps) + wd * p))
+    for a, b in zip(params, ref_p):
+        torch.testing.assert_close(a, b, atol=1e-5, rtol=1e-5)
+
+
+# ----------------------------------------------------- dispatch is native
+def test_gpu_dispatch_uses_extension():
+    """The HIP path must actually run on GPU tensors (no silent eager)."""
+    _cuda()
+    import chinesener_amd.ops.functional as fn
+    x = torch.randn(8, 768, device="cuda").to(torch.bfloat16)
+    w = torch.ones(768, device="cuda")
+    b = torch.zeros(768, device="cuda")
+    assert ops.hip_enabled(x)
+    y = ops.layernorm(x, w, b)
+    assert y.dtype == torch.bfloat16 and y.is_cuda
