@@ -50,6 +50,8 @@ std::vector<at::Tensor> tener_attn_bwd(const at::Tensor&, const at::Tensor&,
                                        const at::Tensor&, const at::Tensor&,
                                        const at::Tensor&, const at::Tensor&,
                                        const at::Tensor&);
+// probe.hip
+at::Tensor mfma_probe(const at::Tensor&, const at::Tensor&);
 // lstm.hip
 std::vector<at::Tensor> lstm_fwd(const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, bool, bool);
@@ -76,6 +78,7 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("attn_bwd", &attn_bwd);
   m.def("tener_attn_fwd", &tener_attn_fwd);
   m.def("tener_attn_bwd", &tener_attn_bwd);
+  m.def("mfma_probe", &mfma_probe);
   m.def("lstm_fwd", &lstm_fwd);
   m.def("lstm_bwd", &lstm_bwd);
 }

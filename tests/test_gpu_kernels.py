@@ -306,3 +306,14 @@ def test_gpu_dispatch_uses_extension():
     assert ops.hip_enabled(x)
     y = ops.layernorm(x, w, b)
     assert y.dtype == torch.bfloat16 and y.is_cuda
+
+
+def test_mfma_fragment_layout():
+    """Single-fragment probe: assumed A/B/C layouts vs torch matmul.
+    Asymmetric random inputs (transpose-detecting)."""
+    _cuda()
+    torch.manual_seed(42)
+    a = torch.randn(16, 32, device="cuda")
+    b = torch.randn(32, 16, device="cuda")
+    c = ops.get_ext().mfma_probe(a.contiguous(), b.t().contiguous())
+    torch.testing.assert_close(c, a @ b, atol=0.2, rtol=0.05)
