@@ -62,9 +62,10 @@ def add_layernorm(x, residual, weight, bias, eps: float = 1e-12):
     """LayerNorm(x + residual) — the BERT post-LN residual pattern."""
     if hip_enabled(x):
         shape = x.shape
-        y = _AddLayerNormFn.apply(x.reshape(-1, shape[-1]).contiguous(),
-                                  residual.reshape(-1, shape[-1]).contiguous(),
-                                  weight, bias, eps)
+        y = _AddLayerNormFn.apply(
+            x.reshape(-1, shape[-1]).contiguous(),
+            residual.to(x.dtype).reshape(-1, shape[-1]).contiguous(),
+            weight, bias, eps)
         return y.reshape(shape)
     return ref.add_layernorm(x, residual, weight, bias, eps)
 

@@ -378,8 +378,17 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
                  (lane & 15)];
 #pragma unroll
       for (int nd = 0; nd < 4; ++nd)
-        if (nd < NFD)
-          acc[nd] = mfma16(ads, lds_frag(k_s, nd * 16, D, kk * 32), acc[nd]);
+        if (nd < NFD) {
+          // B = K[key, d]: row-major [key][d] needs per-element reads
+          // (the lds_frag helper would read it as K^T)
+          bfrag bk;
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            reinterpret_cast<bf16*>(&bk)[e] =
+                k_s[(long)(kk * 32 + ((lane >> 4) << 3) + e) * D + nd * 16 +
+                    (lane & 15)];
+          acc[nd] = mfma16(ads, bk, acc[nd]);
+        }
     }
 #pragma unroll
     for (int nd = 0; nd < 4; ++nd) {

@@ -324,6 +324,8 @@ std::vector<at::Tensor> add_layernorm_fwd(const at::Tensor& x,
                                           const at::Tensor& w,
                                           const at::Tensor& b, double eps) {
   CHECK_CUDA_CONTIG(res);
+  TORCH_CHECK(res.scalar_type() == x.scalar_type(),
+              "add_layernorm: x/residual dtype mismatch");
   return layernorm_fwd_impl(x, res, w, b, eps, true);
 }
 

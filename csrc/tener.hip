@@ -386,8 +386,16 @@ __global__ __launch_bounds__(256) void tener_bwd_kernel(
             pt_s[(long)(kk * 32 + ((lane >> 4) << 3) + e) * Lpad + m0 +
                  (lane & 15)];
 #pragma unroll
-      for (int nd = 0; nd < 2; ++nd)
-        acc[nd] = mfma16t(ads, lds_frag_t(k_s, nd * 16, D, kk * 32), acc[nd]);
+      for (int nd = 0; nd < 2; ++nd) {
+        // B = K[key, d] row-major -> per-element reads (not lds_frag_t)
+        bfrag bk;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          reinterpret_cast<bf16*>(&bk)[e] =
+              k_s[(long)(kk * 32 + ((lane >> 4) << 3) + e) * D + nd * 16 +
+                  (lane & 15)];
+        acc[nd] = mfma16t(ads, bk, acc[nd]);
+      }
     }
 #pragma unroll
     for (int nd = 0; nd < 2; ++nd)

@@ -124,11 +124,11 @@ def test_tener_fwd_bwd():
     _cuda()
     torch.manual_seed(4)
     B, H, L, D = 2, 2, 150, 20
-    q32 = torch.randn(B, H, L, D, device="cuda", requires_grad=True) * 0.5
-    k32 = torch.randn(B, H, L, D, device="cuda", requires_grad=True) * 0.5
+    q32 = (torch.randn(B, H, L, D, device="cuda") * 0.5).requires_grad_()
+    k32 = (torch.randn(B, H, L, D, device="cuda") * 0.5).requires_grad_()
     v32 = torch.randn(B, H, L, D, device="cuda", requires_grad=True)
-    u32 = torch.randn(H, D, device="cuda", requires_grad=True) * 0.1
-    vb32 = torch.randn(H, D, device="cuda", requires_grad=True) * 0.1
+    u32 = (torch.randn(H, D, device="cuda") * 0.1).requires_grad_()
+    vb32 = (torch.randn(H, D, device="cuda") * 0.1).requires_grad_()
     rel = ref.relative_table(L, D, device="cuda")
     lens = torch.tensor([L, 97], device="cuda")
     mask = (torch.arange(L, device="cuda")[None, :] < lens[:, None]).long()
