@@ -78,9 +78,7 @@ class BertLayer(nn.Module):
     def forward(self, x, mask, lens):
         B, L, H = x.shape
         qkv = self.qkv(x).reshape(B, L, 3, self.n_heads, self.head_dim)
-        q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # [B,h,L,d]
-        ctx = ops.attention(q, k, v, mask=mask, lens=lens)
-        ctx = ctx.transpose(1, 2).reshape(B, L, H)
+        ctx = ops.attention_qkv(qkv, mask=mask, lens=lens).reshape(B, L, H)
         a = self.dropout(self.attn_out(ctx))
         x = ops.add_layernorm(a, x, self.ln1_w, self.ln1_b, self.eps)
         f = ops.bias_gelu(self.ffn_in(x), self.ffn_in_bias)

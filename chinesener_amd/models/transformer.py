@@ -48,9 +48,8 @@ class EncoderLayer(nn.Module):
     def forward(self, x, mask, lens=None):
         B, L, D = x.shape
         qkv = self.qkv(x).reshape(B, L, 3, self.n_heads, self.head_dim)
-        q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
-        ctx = ops.attention(q, k, v, mask=mask, lens=lens)
-        a = self.dropout(self.out(ctx.transpose(1, 2).reshape(B, L, D)))
+        ctx = ops.attention_qkv(qkv, mask=mask, lens=lens)
+        a = self.dropout(self.out(ctx.reshape(B, L, D)))
         x = ops.add_layernorm(a, x, self.ln_w, self.ln_b)
         return self.ffn(x)
 

@@ -54,6 +54,7 @@ def get_ext():
 
 
 from .functional import (  # noqa: E402,F401
-    attention, bias_gelu, add_layernorm, layernorm, crf_nll, crf_viterbi,
-    bilstm, softlexicon_fuse, masked_cross_entropy, dice_loss, tener_attention,
+    attention, attention_qkv, bias_gelu, add_layernorm, layernorm, crf_nll,
+    crf_viterbi, bilstm, softlexicon_fuse, masked_cross_entropy, dice_loss,
+    tener_attention,
 )

@@ -134,6 +134,48 @@ def attention(q, k, v, mask: Optional[torch.Tensor] = None,
     return ref.attention(q, k, v, mask, scale)
 
 
+class _AttentionQkvFn(torch.autograd.Function):
+    """Packed-layout path: qkv [B,L,3,H,D] -> out [B,L,H,D]; zero
+    transpose/copies around the kernel (strided kernel I/O)."""
+
+    @staticmethod
+    def forward(ctx, qkv, lens, scale):
+        out, lse = get_ext().attn_fwd_qkv(qkv, lens, scale)
+        ctx.save_for_backward(qkv, out, lse, lens)
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        qkv, out, lse, lens = ctx.saved_tensors
+        (dqkv,) = get_ext().attn_bwd_qkv(dout.contiguous(), qkv, out, lse,
+                                         lens, ctx.scale)
+        return dqkv, None, None
+
+
+def attention_qkv(qkv, mask: Optional[torch.Tensor] = None,
+                  lens: Optional[torch.Tensor] = None,
+                  scale: Optional[float] = None):
+    """Fused attention on the packed QKV projection output.
+
+    qkv: [B, L, 3, H, D] (the natural reshape of the fused QKV GEMM) ->
+    [B, L, H, D]. HIP path avoids every layout copy; fallback unpacks."""
+    B, L, _, H, D = qkv.shape
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    if (hip_enabled(qkv) and qkv.dtype == torch.bfloat16
+            and D in (32, 64) and L <= 176):
+        if lens is None:
+            lens = (mask.long().sum(1) if mask is not None
+                    else torch.full((B,), L, dtype=torch.long,
+                                    device=qkv.device))
+        return _AttentionQkvFn.apply(qkv.contiguous(), lens.to(torch.int32),
+                                     float(scale))
+    q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # [B,H,L,D]
+    out = attention(q, k, v, mask=mask, scale=scale, lens=lens)
+    return out.transpose(1, 2)
+
+
 class _TenerAttentionFn(torch.autograd.Function):
     """Kernel contract: qu = q+u, qv = q+v precomputed (u/v grads are
     reductions of dqu/dqv handled by autograd outside this Function)."""

@@ -74,6 +74,7 @@ class AdamWeightDecay(torch.optim.Optimizer):
     @torch.no_grad()
     def step(self, closure=None):
         loss = closure() if closure is not None else None
+        fused = []   # (params, grads, ms, vs, lrs, wds) across ALL groups
         for group in self.param_groups:
             params = [p for p in group["params"] if p.grad is not None]
             if not params:
@@ -90,8 +91,8 @@ class AdamWeightDecay(torch.optim.Optimizer):
             b1, b2 = group["betas"]
             eps, wd = group["eps"], group["weight_decay"]
             if params[0].is_cuda and ops.ext_available():
-                ops.get_ext().multi_tensor_adamw(
-                    params, grads, ms, vs, lr, b1, b2, eps, wd)
+                fused.append((params, grads, ms, vs,
+                              [lr] * len(params), [wd] * len(params)))
             else:
                 torch._foreach_mul_(ms, b1)
                 torch._foreach_add_(ms, grads, alpha=1 - b1)
@@ -103,6 +104,13 @@ class AdamWeightDecay(torch.optim.Optimizer):
                 if wd:
                     torch._foreach_add_(update, params, alpha=wd)
                 torch._foreach_add_(params, update, alpha=-lr)
+        if fused:
+            # one multi-tensor kernel launch for every group (per-tensor lr)
+            b1, b2 = self.param_groups[0]["betas"]
+            eps = self.param_groups[0]["eps"]
+            cat = [sum((f[i] for f in fused), []) for i in range(6)]
+            ops.get_ext().multi_tensor_adamw(cat[0], cat[1], cat[2], cat[3],
+                                             cat[4], cat[5], b1, b2, eps)
         return loss
 
 

@@ -13,15 +13,18 @@ struct ChunkMeta {
   float* m;
   float* v;
   long n;
+  float lr;
+  float wd;
 };
 
 __global__ void multi_tensor_adamw_kernel(ChunkMeta* metas, int n_tensors,
-                                          float lr, float b1, float b2,
-                                          float eps, float wd) {
+                                          float b1, float b2, float eps) {
   // grid-stride over (tensor, element) pairs: block handles slices of one
   // tensor chosen by blockIdx.y-style flattening
   for (int ti = blockIdx.y; ti < n_tensors; ti += gridDim.y) {
     const ChunkMeta mt = metas[ti];
+    const float lr = mt.lr;
+    const float wd = mt.wd;
     const long stride = (long)gridDim.x * blockDim.x;
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < mt.n;
          i += stride) {
@@ -38,8 +41,8 @@ __global__ void multi_tensor_adamw_kernel(ChunkMeta* metas, int n_tensors,
 void multi_tensor_adamw(std::vector<at::Tensor> params,
                         std::vector<at::Tensor> grads,
                         std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
-                        double lr, double b1, double b2, double eps,
-                        double wd) {
+                        std::vector<double> lrs, std::vector<double> wds,
+                        double b1, double b2, double eps) {
   const int n = params.size();
   TORCH_CHECK(n > 0 && n <= MAX_TENSORS, "multi_tensor_adamw: bad tensor count");
   std::vector<ChunkMeta> metas(n);
@@ -50,7 +53,7 @@ void multi_tensor_adamw(std::vector<at::Tensor> params,
     TORCH_CHECK(grads[i].is_contiguous(), "adamw: non-contiguous grad");
     metas[i] = {params[i].data_ptr<float>(), grads[i].data_ptr<float>(),
                 ms[i].data_ptr<float>(), vs[i].data_ptr<float>(),
-                params[i].numel()};
+                params[i].numel(), (float)lrs[i], (float)wds[i]};
     total += metas[i].n;
   }
   auto meta_blob = at::from_blob(metas.data(), {(long)(n * sizeof(ChunkMeta))},
@@ -61,7 +64,7 @@ void multi_tensor_adamw(std::vector<at::Tensor> params,
   dim3 grid(std::max(gx, 1), std::min(n, 64));
   hipLaunchKernelGGL(multi_tensor_adamw_kernel, grid, dim3(block), 0,
                      cur_stream(params[0]),
-                     (ChunkMeta*)meta_blob.data_ptr(), n, (float)lr, (float)b1,
-                     (float)b2, (float)eps, (float)wd);
+                     (ChunkMeta*)meta_blob.data_ptr(), n, (float)b1,
+                     (float)b2, (float)eps);
   HIP_CHECK_LAST();
 }

@@ -1,19 +1,23 @@
-// Fused attention fwd/bwd for BERT + vanilla MHA and
-// the TENER relative-position variant (K9), hand-written for gfx950:
-// MFMA 16x16x32 bf16, all tiles LDS-resident (reference seq lens are
-// <= 170, so K/V/P fit in the 160 KiB LDS whole -- no online softmax
-// needed; kernels assert L_pad <= 176).
+// Fused attention fwd/bwd for BERT + vanilla MHA,
+// hand-written for gfx950: MFMA 16x16x32 bf16, all tiles LDS-resident
+// (reference seq lens are <= 170 -> K/V/P fit whole in the 160 KiB LDS;
+// no online softmax needed; L_pad <= 176).
 //
-// Layout notes (guide cdna_hip_programming.md §3):
+// Layout notes (guide cdna_hip_programming.md par.3):
 //  * A-frag  lane l -> A[(l&15)][kk*32 + (l>>4)*8 + e], e = 0..7
-//  * B-frag  lane l -> B[kk*32 + (l>>4)*8 + e][(l&15)] — read from an
+//  * B-frag  lane l -> B[kk*32 + (l>>4)*8 + e][(l&15)] -- read from an
 //    [N][K]-stored (transposed) buffer with the same indexing as A.
 //  * C/D     lane l, reg r -> D[(l>>4)*4 + r][nf*16 + (l&15)]
-// Key (K) and V^T are staged so every MFMA operand read is one 16-byte
+// K and V^T are staged so every MFMA operand read is one 16-byte
 // contiguous LDS read.
+//
+// Strided I/O: the kernels take (batch, head, row) strides so the hot
+// path consumes the fused-QKV projection output [B,L,3,H,D] directly and
+// writes O as [B,L,H,D] -- zero transpose/copy around the kernel
+// (torch-side reshape of the GEMM output is free).
 #include "common.h"
 
-#define MAXNF 12  // N-tiles of 16: L_pad <= 176 -> 11
+#define MAXNF 11  // N-tiles of 16: L_pad <= 176
 #define LPAD_MAX 176
 
 using bfrag = mfma_bf16x8;
@@ -26,31 +30,45 @@ __device__ __forceinline__ bfrag lds_frag(const bf16* base, int i0, int ld,
                                          k0 + ((l >> 4) << 3));
 }
 
+// B operand from a ROW-major [K][N] LDS buffer (per-element reads)
+__device__ __forceinline__ bfrag lds_fragB_rowmajor(const bf16* base, int k0,
+                                                    int ld, int n0) {
+  const int l = threadIdx.x & (WAVE - 1);
+  bfrag b;
+#pragma unroll
+  for (int e = 0; e < 8; ++e)
+    reinterpret_cast<bf16*>(&b)[e] =
+        base[(long)(k0 + ((l >> 4) << 3) + e) * ld + n0 + (l & 15)];
+  return b;
+}
+
 __device__ __forceinline__ cfrag mfma16(bfrag a, bfrag b, cfrag c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
-// stage a [rows x cols] bf16 global tile into LDS row-major, zero-padding
-// to [rpad x cols]; cols % 8 == 0
-__device__ __forceinline__ void stage_tile(const bf16* g, bf16* s, int rows,
-                                           int rpad, int cols) {
-  const int nv = rpad * cols / 8;
+// stage a [rows x D] global tile (row stride rs) into LDS [rpad x D]
+__device__ __forceinline__ void stage_tile(const bf16* g, long rs, bf16* s,
+                                           int rows, int rpad, int D) {
+  const int nv = rpad * D / 8;
   for (int i = threadIdx.x; i < nv; i += blockDim.x) {
-    const int r = (i * 8) / cols;
+    const int r = (i * 8) / D;
+    const int c = (i * 8) % D;
     s16x8 val{};
-    if (r < rows) val = reinterpret_cast<const s16x8*>(g)[i];
+    if (r < rows)
+      val = *reinterpret_cast<const s16x8*>(g + (long)r * rs + c);
     reinterpret_cast<s16x8*>(s)[i] = val;
   }
 }
 
-// stage transposed: global [rows x cols] -> LDS [cols x rpad]
-__device__ __forceinline__ void stage_tile_T(const bf16* g, bf16* s, int rows,
-                                             int rpad, int cols) {
-  for (int i = threadIdx.x; i < rpad * cols / 8; i += blockDim.x) {
-    const int r = (i * 8) / cols;
-    const int c0 = (i * 8) % cols;
+// stage transposed: global [rows x D] (row stride rs) -> LDS [D x rpad]
+__device__ __forceinline__ void stage_tile_T(const bf16* g, long rs, bf16* s,
+                                             int rows, int rpad, int D) {
+  for (int i = threadIdx.x; i < rpad * D / 8; i += blockDim.x) {
+    const int r = (i * 8) / D;
+    const int c0 = (i * 8) % D;
     s16x8 val{};
-    if (r < rows) val = reinterpret_cast<const s16x8*>(g)[i];
+    if (r < rows)
+      val = *reinterpret_cast<const s16x8*>(g + (long)r * rs + c0);
 #pragma unroll
     for (int e = 0; e < 8; ++e)
       s[(long)(c0 + e) * rpad + r] = reinterpret_cast<const bf16*>(&val)[e];
@@ -58,14 +76,14 @@ __device__ __forceinline__ void stage_tile_T(const bf16* g, bf16* s, int rows,
 }
 
 // ---------------------------------------------------------------------
-// forward: one block = one (b,h); 4 waves each loop M-tiles of 16 rows.
-// out[b,h] = softmax(scale * Q K^T + keymask) V ; lse saved for bwd.
+// forward
 // ---------------------------------------------------------------------
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const int* __restrict__ lens,
     bf16* __restrict__ out, float* __restrict__ lse, int B, int H, int L,
-    int D, int Lpad, float scale) {
+    int D, int Lpad, float scale, long q_bs, long q_hs, long q_rs, long o_bs,
+    long o_hs, long o_rs) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   bf16* q_s = reinterpret_cast<bf16*>(smem_raw);    // [Lpad][D]
   bf16* k_s = q_s + Lpad * D;                       // [Lpad][D]
@@ -74,22 +92,24 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
   const int bh = blockIdx.x;
   const int b = bh / H;
-  const long base = (long)bh * L * D;
+  const int h = bh - b * H;
+  const long qb = (long)b * q_bs + (long)h * q_hs;
+  const long ob = (long)b * o_bs + (long)h * o_hs;
   const int len = lens[b];
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
 
-  stage_tile(q + base, q_s, L, Lpad, D);
-  stage_tile(k + base, k_s, L, Lpad, D);
-  stage_tile_T(v + base, vt_s, L, Lpad, D);
+  stage_tile(q + qb, q_rs, q_s, L, Lpad, D);
+  stage_tile(k + qb, q_rs, k_s, L, Lpad, D);
+  stage_tile_T(v + qb, q_rs, vt_s, L, Lpad, D);
   __syncthreads();
 
   const int NF = Lpad / 16;
   const int NKK = D / 32;
+  const int NFD = D / 16;
   bf16* pw = p_s + wid * 16 * Lpad;
 
   for (int m0 = wid * 16; m0 < L; m0 += 4 * 16) {
-    // ---- S = scale * Q K^T over this wave's 16 rows ----
     cfrag acc[MAXNF];
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) acc[nf] = cfrag{0.f, 0.f, 0.f, 0.f};
@@ -99,51 +119,53 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       if (kk < NKK) aq[kk] = lds_frag(q_s, m0, D, kk * 32);
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk)
-        if (kk < NKK)
-          acc[nf] = mfma16(aq[kk], lds_frag(k_s, nf * 16, D, kk * 32), acc[nf]);
+        for (int kk = 0; kk < 2; ++kk)
+          if (kk < NKK)
+            acc[nf] = mfma16(aq[kk], lds_frag(k_s, nf * 16, D, kk * 32),
+                             acc[nf]);
+      }
     }
-    // ---- softmax rows (C layout: reg r = row, lanes&15 = col) ----
+    // softmax over rows (C layout: reg r = row, lane&15 = col)
     float row_lse[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       float mx = -1e30f;
 #pragma unroll
       for (int nf = 0; nf < MAXNF; ++nf) {
-        if (nf >= NF) break;
-        const int col = nf * 16 + (lane & 15);
-        const float s = (col < len) ? acc[nf][r] * scale : -1e30f;
-        acc[nf][r] = s;
-        mx = fmaxf(mx, s);
+        if (nf < NF) {
+          const int col = nf * 16 + (lane & 15);
+          const float s = (col < len) ? acc[nf][r] * scale : -1e30f;
+          acc[nf][r] = s;
+          mx = fmaxf(mx, s);
+        }
       }
       mx = group16_reduce_max(mx);
       float sum = 0.f;
 #pragma unroll
       for (int nf = 0; nf < MAXNF; ++nf) {
-        if (nf >= NF) break;
-        const float p = __expf(acc[nf][r] - mx);
-        acc[nf][r] = p;
-        sum += p;
+        if (nf < NF) {
+          const float p = __expf(acc[nf][r] - mx);
+          acc[nf][r] = p;
+          sum += p;
+        }
       }
       sum = group16_reduce_sum(sum);
       const float inv = __frcp_rn(sum);
 #pragma unroll
-      for (int nf = 0; nf < MAXNF; ++nf) {
-        if (nf >= NF) break;
-        acc[nf][r] *= inv;
-      }
+      for (int nf = 0; nf < MAXNF; ++nf)
+        if (nf < NF) acc[nf][r] *= inv;
       row_lse[r] = mx + __logf(sum);
     }
-    // write normalized P (bf16) into this wave's LDS tile [16][Lpad]
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int lr = ((lane >> 4) << 2) + r;
-        pw[lr * Lpad + nf * 16 + (lane & 15)] = __float2bfloat16(acc[nf][r]);
+        for (int r = 0; r < 4; ++r) {
+          const int lr = ((lane >> 4) << 2) + r;
+          pw[lr * Lpad + nf * 16 + (lane & 15)] = __float2bfloat16(acc[nf][r]);
+        }
       }
     }
     if ((lane & 15) == 0) {
@@ -153,8 +175,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         if (row < L) lse[(long)bh * L + row] = row_lse[r];
       }
     }
-    // ---- O = P V  (A from pw, B from vt_s) ----
-    const int NFD = D / 16;
+    // O = P V
     cfrag oacc[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) oacc[i] = cfrag{0.f, 0.f, 0.f, 0.f};
@@ -163,26 +184,27 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int nd = 0; nd < 4; ++nd)
         if (nd < NFD)
-          oacc[nd] = mfma16(ap, lds_frag(vt_s, nd * 16, Lpad, kk * 32), oacc[nd]);
+          oacc[nd] = mfma16(ap, lds_frag(vt_s, nd * 16, Lpad, kk * 32),
+                            oacc[nd]);
     }
 #pragma unroll
     for (int nd = 0; nd < 4; ++nd) {
-      if (nd >= NFD) break;
+      if (nd < NFD) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = m0 + ((lane >> 4) << 2) + r;
-        if (row < L)
-          out[base + (long)row * D + nd * 16 + (lane & 15)] =
-              __float2bfloat16(oacc[nd][r]);
+        for (int r = 0; r < 4; ++r) {
+          const int row = m0 + ((lane >> 4) << 2) + r;
+          if (row < L)
+            out[ob + (long)row * o_rs + nd * 16 + (lane & 15)] =
+                __float2bfloat16(oacc[nd][r]);
+        }
       }
     }
   }
 }
 
 // ---------------------------------------------------------------------
-// backward: one block = one (b,h).
-// P recomputed from lse; dV = P^T dO ; dS = P*(dP - delta); dQ = dS K ;
-// dK = dS^T Q. P^T lives in LDS and is overwritten by dS^T in place.
+// backward: P recomputed from lse; dV = P^T dO ; dS = P*(dP - delta);
+// dK = dS^T Q ; dQ = dS K. P^T is overwritten by dS^T in place.
 // ---------------------------------------------------------------------
 __global__ __launch_bounds__(256) void attn_bwd_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q,
@@ -190,19 +212,22 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     const bf16* __restrict__ o, const float* __restrict__ lse,
     const int* __restrict__ lens, bf16* __restrict__ dq,
     bf16* __restrict__ dk, bf16* __restrict__ dv, int B, int H, int L, int D,
-    int Lpad, float scale) {
+    int Lpad, float scale, long q_bs, long q_hs, long q_rs, long o_bs,
+    long o_hs, long o_rs) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   bf16* q_s = reinterpret_cast<bf16*>(smem_raw);   // [Lpad][D]
   bf16* k_s = q_s + Lpad * D;                      // [Lpad][D]
-  bf16* v_s = k_s + Lpad * D;                      // [Lpad][D] (row major!)
+  bf16* v_s = k_s + Lpad * D;                      // [Lpad][D] row major
   bf16* do_s = v_s + Lpad * D;                     // [Lpad][D]
-  bf16* pt_s = do_s + Lpad * D;                    // [Lpad][Lpad]  P^T / dS^T
-  float* delta_s = reinterpret_cast<float*>(pt_s + (long)Lpad * Lpad);  // [Lpad]
-  float* lse_s = delta_s + Lpad;                                        // [Lpad]
+  bf16* pt_s = do_s + Lpad * D;                    // [Lpad][Lpad] P^T / dS^T
+  float* delta_s = reinterpret_cast<float*>(pt_s + (long)Lpad * Lpad);
+  float* lse_s = delta_s + Lpad;
 
   const int bh = blockIdx.x;
   const int b = bh / H;
-  const long base = (long)bh * L * D;
+  const int h = bh - b * H;
+  const long qb = (long)b * q_bs + (long)h * q_hs;
+  const long ob = (long)b * o_bs + (long)h * o_hs;
   const int len = lens[b];
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -210,17 +235,16 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   const int NKK = D / 32;
   const int NFD = D / 16;
 
-  stage_tile(q + base, q_s, L, Lpad, D);
-  stage_tile(k + base, k_s, L, Lpad, D);
-  stage_tile(v + base, v_s, L, Lpad, D);
-  stage_tile(dout + base, do_s, L, Lpad, D);
-  // delta[row] = sum_d dO[row,d] * O[row,d]; lse -> LDS
+  stage_tile(q + qb, q_rs, q_s, L, Lpad, D);
+  stage_tile(k + qb, q_rs, k_s, L, Lpad, D);
+  stage_tile(v + qb, q_rs, v_s, L, Lpad, D);
+  stage_tile(dout + ob, o_rs, do_s, L, Lpad, D);
   for (int r = threadIdx.x; r < Lpad; r += blockDim.x) {
     float acc = 0.f;
     if (r < L) {
       for (int d = 0; d < D; ++d)
-        acc += to_f32(dout[base + (long)r * D + d]) *
-               to_f32(o[base + (long)r * D + d]);
+        acc += to_f32(dout[ob + (long)r * o_rs + d]) *
+               to_f32(o[ob + (long)r * o_rs + d]);
       lse_s[r] = lse[(long)bh * L + r];
     } else {
       lse_s[r] = 0.f;
@@ -229,7 +253,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   }
   __syncthreads();
 
-  // ---- phase 1: recompute P, store P^T ----
+  // phase 1: recompute P -> P^T
   for (int m0 = wid * 16; m0 < Lpad; m0 += 64) {
     cfrag acc[MAXNF];
 #pragma unroll
@@ -240,29 +264,32 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       if (kk < NKK) aq[kk] = lds_frag(q_s, m0, D, kk * 32);
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk)
-        if (kk < NKK)
-          acc[nf] = mfma16(aq[kk], lds_frag(k_s, nf * 16, D, kk * 32), acc[nf]);
+        for (int kk = 0; kk < 2; ++kk)
+          if (kk < NKK)
+            acc[nf] = mfma16(aq[kk], lds_frag(k_s, nf * 16, D, kk * 32),
+                             acc[nf]);
+      }
     }
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = m0 + ((lane >> 4) << 2) + r;
-        const int col = nf * 16 + (lane & 15);
-        float p = 0.f;
-        if (row < L && col < len)
-          p = __expf(acc[nf][r] * scale - lse_s[row]);
-        pt_s[(long)col * Lpad + row] = __float2bfloat16(p);
+        for (int r = 0; r < 4; ++r) {
+          const int row = m0 + ((lane >> 4) << 2) + r;
+          const int col = nf * 16 + (lane & 15);
+          float p = 0.f;
+          if (row < L && col < len)
+            p = __expf(acc[nf][r] * scale - lse_s[row]);
+          pt_s[(long)col * Lpad + row] = __float2bfloat16(p);
+        }
       }
     }
   }
   __syncthreads();
 
-  // ---- phase 2: dV[kt] = P^T[kt,:] dO  (key tiles round-robin) ----
+  // phase 2: dV[keys] = P^T dO
   for (int k0 = wid * 16; k0 < L; k0 += 64) {
     cfrag acc[4];
 #pragma unroll
@@ -271,34 +298,26 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       const bfrag ap = lds_frag(pt_s, k0, Lpad, kk * 32);
 #pragma unroll
       for (int nd = 0; nd < 4; ++nd)
-        if (nd < NFD) {
-          // B = dO[rows, D] : B[k=row][col=d] contiguous in do_s? No:
-          // do_s is [row][d] row-major, element e varies row -> strided.
-          // Use transposed read via per-element loads.
-          bfrag bo;
-#pragma unroll
-          for (int e = 0; e < 8; ++e)
-            reinterpret_cast<bf16*>(&bo)[e] =
-                do_s[(long)(kk * 32 + ((lane >> 4) << 3) + e) * D + nd * 16 +
-                     (lane & 15)];
-          acc[nd] = mfma16(ap, bo, acc[nd]);
-        }
+        if (nd < NFD)
+          acc[nd] = mfma16(ap, lds_fragB_rowmajor(do_s, kk * 32, D, nd * 16),
+                           acc[nd]);
     }
 #pragma unroll
     for (int nd = 0; nd < 4; ++nd) {
-      if (nd >= NFD) break;
+      if (nd < NFD) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int krow = k0 + ((lane >> 4) << 2) + r;
-        if (krow < L)
-          dv[base + (long)krow * D + nd * 16 + (lane & 15)] =
-              __float2bfloat16(acc[nd][r]);
+        for (int r = 0; r < 4; ++r) {
+          const int krow = k0 + ((lane >> 4) << 2) + r;
+          if (krow < L)
+            dv[qb + (long)krow * q_rs + nd * 16 + (lane & 15)] =
+                __float2bfloat16(acc[nd][r]);
+        }
       }
     }
   }
   __syncthreads();
 
-  // ---- phase 3: dP = dO V^T ; dS = P (dP - delta) * scale -> dS^T ----
+  // phase 3: dP = dO V^T ; dS = P (dP - delta) * scale -> overwrite P^T
   for (int m0 = wid * 16; m0 < L; m0 += 64) {
     cfrag acc[MAXNF];
 #pragma unroll
@@ -309,30 +328,31 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       if (kk < NKK) ado[kk] = lds_frag(do_s, m0, D, kk * 32);
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk)
-        if (kk < NKK)
-          acc[nf] = mfma16(ado[kk], lds_frag(v_s, nf * 16, D, kk * 32), acc[nf]);
+        for (int kk = 0; kk < 2; ++kk)
+          if (kk < NKK)
+            acc[nf] = mfma16(ado[kk], lds_frag(v_s, nf * 16, D, kk * 32),
+                             acc[nf]);
+      }
     }
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = m0 + ((lane >> 4) << 2) + r;
-        const int col = nf * 16 + (lane & 15);
-        const float p = to_f32(pt_s[(long)col * Lpad + row]);
-        const float ds = p * (acc[nf][r] - delta_s[row]) * scale;
-        // do NOT write yet — in-place overwrite of P^T is safe only
-        // element-wise within this lane's own slots, which it is:
-        pt_s[(long)col * Lpad + row] = __float2bfloat16(ds);
+        for (int r = 0; r < 4; ++r) {
+          const int row = m0 + ((lane >> 4) << 2) + r;
+          const int col = nf * 16 + (lane & 15);
+          const float p = to_f32(pt_s[(long)col * Lpad + row]);
+          const float ds = p * (acc[nf][r] - delta_s[row]) * scale;
+          pt_s[(long)col * Lpad + row] = __float2bfloat16(ds);
+        }
       }
     }
   }
   __syncthreads();
 
-  // ---- phase 4: dK[kt] = dS^T[kt,:] Q ----
+  // phase 4: dK[keys] = dS^T Q
   for (int k0 = wid * 16; k0 < L; k0 += 64) {
     cfrag acc[4];
 #pragma unroll
@@ -341,30 +361,25 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       const bfrag ads = lds_frag(pt_s, k0, Lpad, kk * 32);
 #pragma unroll
       for (int nd = 0; nd < 4; ++nd)
-        if (nd < NFD) {
-          bfrag bq;
-#pragma unroll
-          for (int e = 0; e < 8; ++e)
-            reinterpret_cast<bf16*>(&bq)[e] =
-                q_s[(long)(kk * 32 + ((lane >> 4) << 3) + e) * D + nd * 16 +
-                    (lane & 15)];
-          acc[nd] = mfma16(ads, bq, acc[nd]);
-        }
+        if (nd < NFD)
+          acc[nd] = mfma16(ads, lds_fragB_rowmajor(q_s, kk * 32, D, nd * 16),
+                           acc[nd]);
     }
 #pragma unroll
     for (int nd = 0; nd < 4; ++nd) {
-      if (nd >= NFD) break;
+      if (nd < NFD) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int krow = k0 + ((lane >> 4) << 2) + r;
-        if (krow < L)
-          dk[base + (long)krow * D + nd * 16 + (lane & 15)] =
-              __float2bfloat16(acc[nd][r]);
+        for (int r = 0; r < 4; ++r) {
+          const int krow = k0 + ((lane >> 4) << 2) + r;
+          if (krow < L)
+            dk[qb + (long)krow * q_rs + nd * 16 + (lane & 15)] =
+                __float2bfloat16(acc[nd][r]);
+        }
       }
     }
   }
 
-  // ---- phase 5: dQ = dS K  (A = dS rows — strided reads from dS^T) ----
+  // phase 5: dQ = dS K (A strided from dS^T, B row-major K)
   for (int m0 = wid * 16; m0 < L; m0 += 64) {
     cfrag acc[4];
 #pragma unroll
@@ -378,62 +393,57 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
                  (lane & 15)];
 #pragma unroll
       for (int nd = 0; nd < 4; ++nd)
-        if (nd < NFD) {
-          // B = K[key, d]: row-major [key][d] needs per-element reads
-          // (the lds_frag helper would read it as K^T)
-          bfrag bk;
-#pragma unroll
-          for (int e = 0; e < 8; ++e)
-            reinterpret_cast<bf16*>(&bk)[e] =
-                k_s[(long)(kk * 32 + ((lane >> 4) << 3) + e) * D + nd * 16 +
-                    (lane & 15)];
-          acc[nd] = mfma16(ads, bk, acc[nd]);
-        }
+        if (nd < NFD)
+          acc[nd] = mfma16(ads, lds_fragB_rowmajor(k_s, kk * 32, D, nd * 16),
+                           acc[nd]);
     }
 #pragma unroll
     for (int nd = 0; nd < 4; ++nd) {
-      if (nd >= NFD) break;
+      if (nd < NFD) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = m0 + ((lane >> 4) << 2) + r;
-        if (row < L)
-          dq[base + (long)row * D + nd * 16 + (lane & 15)] =
-              __float2bfloat16(acc[nd][r]);
+        for (int r = 0; r < 4; ++r) {
+          const int row = m0 + ((lane >> 4) << 2) + r;
+          if (row < L)
+            dq[qb + (long)row * q_rs + nd * 16 + (lane & 15)] =
+                __float2bfloat16(acc[nd][r]);
+        }
       }
     }
   }
 }
 
 // ===================================================================== host
-static void check_attn_args(const at::Tensor& q, int& B, int& H, int& L,
-                            int& D, int& Lpad) {
-  CHECK_CUDA_CONTIG(q);
-  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention: bf16 only");
-  B = q.size(0);
-  H = q.size(1);
-  L = q.size(2);
-  D = q.size(3);
+struct AttnShape {
+  int B, H, L, D, Lpad;
+};
+
+static AttnShape attn_shape(int B, int H, int L, int D) {
   TORCH_CHECK(D == 32 || D == 64, "attention: head dim must be 32 or 64 "
               "(pad in the wrapper), got ", D);
-  Lpad = ((L + 31) / 32) * 32;
+  const int Lpad = ((L + 31) / 32) * 32;
   TORCH_CHECK(Lpad <= LPAD_MAX,
               "attention: seq len > 176 unsupported by the full-LDS kernel");
+  return {B, H, L, D, Lpad};
 }
 
+// old layout API: q,k,v,out all [B,H,L,D]
 std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
                                  const at::Tensor& v, const at::Tensor& lens,
                                  double scale) {
-  int B, H, L, D, Lpad;
-  check_attn_args(q, B, H, L, D, Lpad);
+  CHECK_CUDA_CONTIG(q);
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention: bf16 only");
+  auto s = attn_shape(q.size(0), q.size(1), q.size(2), q.size(3));
   auto out = at::empty_like(q);
-  auto lse = at::empty({B, H, L}, q.options().dtype(at::kFloat));
-  const size_t smem = (size_t)(3 * Lpad * D + 4 * 16 * Lpad) * sizeof(bf16);
+  auto lse = at::empty({s.B, s.H, s.L}, q.options().dtype(at::kFloat));
+  const size_t smem = (size_t)(3 * s.Lpad * s.D + 4 * 16 * s.Lpad) * sizeof(bf16);
   TORCH_CHECK(smem <= 160 * 1024, "attn fwd LDS overflow");
-  hipLaunchKernelGGL(attn_fwd_kernel, dim3(B * H), dim3(256), smem,
+  const long bs = (long)s.H * s.L * s.D, hs = (long)s.L * s.D, rs = s.D;
+  hipLaunchKernelGGL(attn_fwd_kernel, dim3(s.B * s.H), dim3(256), smem,
                      cur_stream(q), (const bf16*)q.data_ptr(),
                      (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
                      lens.data_ptr<int>(), (bf16*)out.data_ptr(),
-                     lse.data_ptr<float>(), B, H, L, D, Lpad, (float)scale);
+                     lse.data_ptr<float>(), s.B, s.H, s.L, s.D, s.Lpad,
+                     (float)scale, bs, hs, rs, bs, hs, rs);
   HIP_CHECK_LAST();
   return {out, lse};
 }
@@ -442,21 +452,71 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
                                  const at::Tensor& k, const at::Tensor& v,
                                  const at::Tensor& o, const at::Tensor& lse,
                                  const at::Tensor& lens, double scale) {
-  int B, H, L, D, Lpad;
-  check_attn_args(q, B, H, L, D, Lpad);
+  auto s = attn_shape(q.size(0), q.size(1), q.size(2), q.size(3));
   auto dq = at::empty_like(q);
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
-  const size_t smem = (size_t)(4 * Lpad * D + (long)Lpad * Lpad) * sizeof(bf16)
-                      + 2 * Lpad * sizeof(float);
+  const size_t smem = (size_t)(4 * s.Lpad * s.D + (long)s.Lpad * s.Lpad)
+                      * sizeof(bf16) + 2 * s.Lpad * sizeof(float);
   TORCH_CHECK(smem <= 160 * 1024, "attn bwd LDS overflow");
-  hipLaunchKernelGGL(attn_bwd_kernel, dim3(B * H), dim3(256), smem,
+  const long bs = (long)s.H * s.L * s.D, hs = (long)s.L * s.D, rs = s.D;
+  hipLaunchKernelGGL(attn_bwd_kernel, dim3(s.B * s.H), dim3(256), smem,
                      cur_stream(q), (const bf16*)dout.data_ptr(),
                      (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
                      (const bf16*)v.data_ptr(), (const bf16*)o.data_ptr(),
                      lse.data_ptr<float>(), lens.data_ptr<int>(),
                      (bf16*)dq.data_ptr(), (bf16*)dk.data_ptr(),
-                     (bf16*)dv.data_ptr(), B, H, L, D, Lpad, (float)scale);
+                     (bf16*)dv.data_ptr(), s.B, s.H, s.L, s.D, s.Lpad,
+                     (float)scale, bs, hs, rs, bs, hs, rs);
   HIP_CHECK_LAST();
   return {dq, dk, dv};
+}
+
+// packed layout API: qkv [B,L,3,H,D] -> out [B,L,H,D] (no copies)
+std::vector<at::Tensor> attn_fwd_qkv(const at::Tensor& qkv,
+                                     const at::Tensor& lens, double scale) {
+  CHECK_CUDA_CONTIG(qkv);
+  TORCH_CHECK(qkv.scalar_type() == at::kBFloat16, "attention: bf16 only");
+  TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3, "qkv must be [B,L,3,H,D]");
+  auto s = attn_shape(qkv.size(0), qkv.size(3), qkv.size(1), qkv.size(4));
+  auto out = at::empty({s.B, s.L, s.H, s.D}, qkv.options());
+  auto lse = at::empty({s.B, s.H, s.L}, qkv.options().dtype(at::kFloat));
+  const size_t smem = (size_t)(3 * s.Lpad * s.D + 4 * 16 * s.Lpad) * sizeof(bf16);
+  TORCH_CHECK(smem <= 160 * 1024, "attn fwd LDS overflow");
+  const long HD = (long)s.H * s.D;
+  const long q_bs = (long)s.L * 3 * HD, q_hs = s.D, q_rs = 3 * HD;
+  const long o_bs = (long)s.L * HD, o_hs = s.D, o_rs = HD;
+  const bf16* base = (const bf16*)qkv.data_ptr();
+  hipLaunchKernelGGL(attn_fwd_kernel, dim3(s.B * s.H), dim3(256), smem,
+                     cur_stream(qkv), base, base + HD, base + 2 * HD,
+                     lens.data_ptr<int>(), (bf16*)out.data_ptr(),
+                     lse.data_ptr<float>(), s.B, s.H, s.L, s.D, s.Lpad,
+                     (float)scale, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs);
+  HIP_CHECK_LAST();
+  return {out, lse};
+}
+
+std::vector<at::Tensor> attn_bwd_qkv(const at::Tensor& dout,
+                                     const at::Tensor& qkv,
+                                     const at::Tensor& o, const at::Tensor& lse,
+                                     const at::Tensor& lens, double scale) {
+  CHECK_CUDA_CONTIG(dout);
+  auto s = attn_shape(qkv.size(0), qkv.size(3), qkv.size(1), qkv.size(4));
+  auto dqkv = at::empty_like(qkv);
+  const size_t smem = (size_t)(4 * s.Lpad * s.D + (long)s.Lpad * s.Lpad)
+                      * sizeof(bf16) + 2 * s.Lpad * sizeof(float);
+  TORCH_CHECK(smem <= 160 * 1024, "attn bwd LDS overflow");
+  const long HD = (long)s.H * s.D;
+  const long q_bs = (long)s.L * 3 * HD, q_hs = s.D, q_rs = 3 * HD;
+  const long o_bs = (long)s.L * HD, o_hs = s.D, o_rs = HD;
+  const bf16* base = (const bf16*)qkv.data_ptr();
+  bf16* dbase = (bf16*)dqkv.data_ptr();
+  hipLaunchKernelGGL(attn_bwd_kernel, dim3(s.B * s.H), dim3(256), smem,
+                     cur_stream(qkv), (const bf16*)dout.data_ptr(), base,
+                     base + HD, base + 2 * HD, (const bf16*)o.data_ptr(),
+                     lse.data_ptr<float>(), lens.data_ptr<int>(), dbase,
+                     dbase + HD, dbase + 2 * HD, s.B, s.H, s.L, s.D, s.Lpad,
+                     (float)scale, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs);
+  HIP_CHECK_LAST();
+  return {dqkv};
 }

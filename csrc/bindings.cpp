@@ -33,7 +33,8 @@ std::vector<at::Tensor> softlexicon_bwd(const at::Tensor&, const at::Tensor&,
 // adam.hip
 void multi_tensor_adamw(std::vector<at::Tensor>, std::vector<at::Tensor>,
                         std::vector<at::Tensor>, std::vector<at::Tensor>,
-                        double, double, double, double, double);
+                        std::vector<double>, std::vector<double>, double,
+                        double, double);
 // attention.hip
 std::vector<at::Tensor> attn_fwd(const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&, double);
@@ -41,6 +42,11 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, double);
+std::vector<at::Tensor> attn_fwd_qkv(const at::Tensor&, const at::Tensor&,
+                                     double);
+std::vector<at::Tensor> attn_bwd_qkv(const at::Tensor&, const at::Tensor&,
+                                     const at::Tensor&, const at::Tensor&,
+                                     const at::Tensor&, double);
 // tener.hip
 std::vector<at::Tensor> tener_attn_fwd(const at::Tensor&, const at::Tensor&,
                                        const at::Tensor&, const at::Tensor&,
@@ -76,6 +82,8 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("multi_tensor_adamw", &multi_tensor_adamw);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("attn_fwd_qkv", &attn_fwd_qkv);
+  m.def("attn_bwd_qkv", &attn_bwd_qkv);
   m.def("tener_attn_fwd", &tener_attn_fwd);
   m.def("tener_attn_bwd", &tener_attn_bwd);
   m.def("mfma_probe", &mfma_probe);

@@ -85,16 +85,18 @@ __global__ void crf_fwd_kernel(const float* __restrict__ emis,  // [B,L,T]
     float row[TMAX];  // trans[i=lane][jj] + cur[jj]
 #pragma unroll
     for (int jj = 0; jj < TMAX; ++jj) {
-      if (jj >= T) break;
+      if (jj < T) {
       const float cur = __shfl(mine, jj);
       const float v = (act ? tr_s[j * T + jj] : NEG) + cur;
       row[jj] = v;
       m = fmaxf(m, v);
+      }
     }
 #pragma unroll
     for (int jj = 0; jj < TMAX; ++jj) {
-      if (jj >= T) break;
+      if (jj < T) {
       s += __expf(row[jj] - m);
+      }
     }
     const float beta_t = act ? m + __logf(s) : NEG;  // beta[t][i=lane]
     // expected pairwise counts: P(y_t=i, y_{t+1}=jj)
@@ -102,8 +104,9 @@ __global__ void crf_fwd_kernel(const float* __restrict__ emis,  // [B,L,T]
       const float a_ti = alpha[t * T + j];  // lane = i here
 #pragma unroll
       for (int jj = 0; jj < TMAX; ++jj) {
-        if (jj >= T) break;
+        if (jj < T) {
         exp_row[jj] += __expf(a_ti + row[jj] - logZ);
+        }
       }
       // token marginal at t for tag i=lane
       const float marg = __expf(a_ti + beta_t - logZ);
@@ -120,16 +123,18 @@ __global__ void crf_fwd_kernel(const float* __restrict__ emis,  // [B,L,T]
       if (tg[t - 1] == j) {
 #pragma unroll
         for (int jj = 0; jj < TMAX; ++jj) {
-          if (jj >= T) break;
+          if (jj < T) {
           if (tg[t] == jj) gold[jj] += 1.f;
+          }
         }
       }
     }
     float* dt = dtrans + ((long)b * T + j) * T;
 #pragma unroll
     for (int jj = 0; jj < TMAX; ++jj) {
-      if (jj >= T) break;
+      if (jj < T) {
       dt[jj] = gold[jj] - exp_row[jj];
+      }
     }
   }
 }

@@ -31,7 +31,7 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
 #pragma unroll
   for (int it = 0; it < MAXIT; ++it) {
     const int i = lane + it * WAVE;
-    if (i >= HV) break;
+    if (i < HV) {
     const s16x8 raw = reinterpret_cast<const s16x8*>(xr)[i];
     s16x8 raw2{};
     if (rr) raw2 = reinterpret_cast<const s16x8*>(rr)[i];
@@ -42,6 +42,7 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
       vals[it][e] = v;
       acc += v;
       acc2 += v * v;
+    }
     }
   }
   acc = wave_reduce_sum(acc);
@@ -57,7 +58,7 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
 #pragma unroll
   for (int it = 0; it < MAXIT; ++it) {
     const int i = lane + it * WAVE;
-    if (i >= HV) break;
+    if (i < HV) {
     T out[8], sout[8];
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
@@ -68,6 +69,7 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
     }
     reinterpret_cast<s16x8*>(yr)[i] = *reinterpret_cast<s16x8*>(out);
     if (sr) reinterpret_cast<s16x8*>(sr)[i] = *reinterpret_cast<s16x8*>(sout);
+    }
   }
 }
 
@@ -156,6 +158,96 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
       from_f32((g - c2 - xhat * c1) * rs, &dxr[i]);
       atomicAdd(&dw_s[i], d * xhat);
       atomicAdd(&db_s[i], d);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < H; i += blockDim.x) {
+    atomicAdd(&dw[i], dw_s[i]);
+    atomicAdd(&db[i], db_s[i]);
+  }
+}
+
+
+// fast bf16 backward: 32 rows/block (8 per wave), vectorized loads,
+// dw/db accumulated in registers then combined via LDS; H <= 1024.
+__global__ __launch_bounds__(256) void layernorm_bwd_bf16_kernel(
+    const bf16* __restrict__ dy, const bf16* __restrict__ x,
+    const float* __restrict__ w, const float* __restrict__ mean,
+    const float* __restrict__ rstd, bf16* __restrict__ dx,
+    float* __restrict__ dw, float* __restrict__ db, long N, int H) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* dw_s = reinterpret_cast<float*>(smem_raw);
+  float* db_s = dw_s + H;
+  for (int i = threadIdx.x; i < 2 * H; i += blockDim.x) dw_s[i] = 0.f;
+  __syncthreads();
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int HV = H / 8;                 // vec8 slots per row
+  constexpr int MAXC = 2;               // H <= 1024
+  float dwp[MAXC][8] = {}, dbp[MAXC][8] = {};
+  float wv[MAXC][8];
+#pragma unroll
+  for (int c = 0; c < MAXC; ++c) {
+    const int i = lane + c * WAVE;
+    if (i < HV) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) wv[c][e] = w[i * 8 + e];
+    }
+  }
+  for (int rr = 0; rr < 8; ++rr) {
+    const long row = (long)blockIdx.x * 32 + wid * 8 + rr;
+    if (row >= N) break;
+    const bf16* dyr = dy + row * H;
+    const bf16* xr = x + row * H;
+    const float m = mean[row], rs = rstd[row];
+    float xh[MAXC][8], dyv[MAXC][8];
+    float c1 = 0.f, c2 = 0.f;
+#pragma unroll
+    for (int c = 0; c < MAXC; ++c) {
+      const int i = lane + c * WAVE;
+      if (i < HV) {
+        const s16x8 rd = reinterpret_cast<const s16x8*>(dyr)[i];
+        const s16x8 rx = reinterpret_cast<const s16x8*>(xr)[i];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float d = to_f32(reinterpret_cast<const bf16*>(&rd)[e]);
+          const float hx = (to_f32(reinterpret_cast<const bf16*>(&rx)[e]) - m) * rs;
+          dyv[c][e] = d;
+          xh[c][e] = hx;
+          const float g = d * wv[c][e];
+          c1 += g * hx;
+          c2 += g;
+        }
+      }
+    }
+    c1 = wave_reduce_sum(c1) / H;
+    c2 = wave_reduce_sum(c2) / H;
+    bf16* dxr = dx + row * H;
+#pragma unroll
+    for (int c = 0; c < MAXC; ++c) {
+      const int i = lane + c * WAVE;
+      if (i < HV) {
+        bf16 outv[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float g = dyv[c][e] * wv[c][e];
+          from_f32((g - c2 - xh[c][e] * c1) * rs, &outv[e]);
+          dwp[c][e] += dyv[c][e] * xh[c][e];
+          dbp[c][e] += dyv[c][e];
+        }
+        reinterpret_cast<s16x8*>(dxr)[i] = *reinterpret_cast<s16x8*>(outv);
+      }
+    }
+  }
+#pragma unroll
+  for (int c = 0; c < MAXC; ++c) {
+    const int i = lane + c * WAVE;
+    if (i < HV) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        atomicAdd(&dw_s[i * 8 + e], dwp[c][e]);
+        atomicAdd(&db_s[i * 8 + e], dbp[c][e]);
+      }
     }
   }
   __syncthreads();
@@ -340,18 +432,20 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
   auto db = at::zeros({H}, x.options().dtype(at::kFloat));
   auto wf = w.to(at::kFloat).contiguous();
-  const int rows_per_blk = 4;
-  const dim3 grid((N + rows_per_blk - 1) / rows_per_blk);
-  const dim3 block(rows_per_blk * WAVE);
   const size_t smem = 2 * H * sizeof(float);
   auto stream = cur_stream(x);
   if (x.scalar_type() == at::kBFloat16) {
-    hipLaunchKernelGGL(layernorm_bwd_kernel<bf16>, grid, block, smem, stream,
+    TORCH_CHECK(H % 8 == 0 && H <= 1024, "layernorm bwd bf16: bad H ", H);
+    hipLaunchKernelGGL(layernorm_bwd_bf16_kernel, dim3((N + 31) / 32),
+                       dim3(256), smem, stream,
                        (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
                        wf.data_ptr<float>(), mean.data_ptr<float>(),
                        rstd.data_ptr<float>(), (bf16*)dx.data_ptr(),
                        dw.data_ptr<float>(), db.data_ptr<float>(), N, H);
   } else {
+    const int rows_per_blk = 4;
+    const dim3 grid((N + rows_per_blk - 1) / rows_per_blk);
+    const dim3 block(rows_per_blk * WAVE);
     hipLaunchKernelGGL(layernorm_bwd_kernel<float>, grid, block, smem, stream,
                        dy.data_ptr<float>(), x.data_ptr<float>(),
                        wf.data_ptr<float>(), mean.data_ptr<float>(),
@@ -403,7 +497,7 @@ at::Tensor bias_gelu_fwd(const at::Tensor& x, const at::Tensor& bias) {
 std::vector<at::Tensor> bias_gelu_bwd(const at::Tensor& dy, const at::Tensor& x,
                                       const at::Tensor& bias) {
   auto dx = bias_gelu_impl(x, bias, dy);
-  auto dbias = dx.to(at::kFloat).sum({0}).to(bias.scalar_type());
+  auto dbias = at::sum(dx, {0}, false, at::kFloat).to(bias.scalar_type());
   return {dx, dbias};
 }
 

@@ -45,6 +45,7 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   bf16* wT = reinterpret_cast<bf16*>(smem_raw);  // [4h][h]
   bf16* hb = wT + 4 * h * h;                     // [16][h] current h (bf16)
+  T* gx_s = reinterpret_cast<T*>(hb + 16 * h);   // [16][4h] staged gates_x
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int b0 = blockIdx.x * 16;
@@ -69,6 +70,31 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
 
   for (int step = 0; step < L; ++step) {
     const int t = reverse ? (L - 1 - step) : step;
+    // cooperative staging of gates_x[:,t,:] for the 16 rows (vectorized,
+    // overlaps the recurrent MFMAs below; barrier covers both)
+    {
+      const int nv = 16 * 4 * h / 8;
+      for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+        const int row = (i * 8) / (4 * h);
+        const int col = (i * 8) % (4 * h);
+        const int b = b0 + row;
+        s16x8 val{};
+        f32x2 dummy;
+        if (b < B) {
+          const long g = ((long)b * L + t) * 4 * h + col;
+          if (sizeof(T) == 2) {
+            reinterpret_cast<s16x8*>(gx_s)[i] =
+                *reinterpret_cast<const s16x8*>(gates_x + g);
+          } else {
+            reinterpret_cast<f32x4*>(gx_s)[i * 2] =
+                *reinterpret_cast<const f32x4*>(gates_x + g);
+            reinterpret_cast<f32x4*>(gx_s)[i * 2 + 1] =
+                *reinterpret_cast<const f32x4*>(gates_x + g + 4);
+          }
+        }
+        (void)val; (void)dummy;
+      }
+    }
     // gates = h_prev @ W_hh  (+ gates_x added in the epilogue)
     cfrag acc[4][2];
 #pragma unroll
@@ -98,10 +124,11 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
       for (int jj = 0; jj < 2; ++jj) {
         const int j = wid * 32 + jj * 16 + (lane & 15);
         if (j >= h) continue;
-        float gi = acc[0][jj][r] + to_f32(gates_x[gbase + 0 * h + j]);
-        float gf = acc[1][jj][r] + to_f32(gates_x[gbase + 1 * h + j]);
-        float gg = acc[2][jj][r] + to_f32(gates_x[gbase + 2 * h + j]);
-        float go = acc[3][jj][r] + to_f32(gates_x[gbase + 3 * h + j]);
+        const int lr4h = (lrow + r) * 4 * h;
+        float gi = acc[0][jj][r] + to_f32(gx_s[lr4h + 0 * h + j]);
+        float gf = acc[1][jj][r] + to_f32(gx_s[lr4h + 1 * h + j]);
+        float gg = acc[2][jj][r] + to_f32(gx_s[lr4h + 2 * h + j]);
+        float go = acc[3][jj][r] + to_f32(gx_s[lr4h + 3 * h + j]);
         gi = 1.f / (1.f + __expf(-gi));
         gf = 1.f / (1.f + __expf(-gf));
         go = 1.f / (1.f + __expf(-go));
@@ -251,7 +278,8 @@ std::vector<at::Tensor> lstm_fwd(const at::Tensor& gates_x,
   auto cs = at::empty({B, L, h}, gates_x.options().dtype(at::kFloat));
   auto gates = at::empty({B, L, h4}, gates_x.options().dtype(at::kFloat));
   auto w_t = w_hh.t().contiguous().to(at::kBFloat16);  // [4h,h]
-  const size_t smem = (size_t)(4 * h * h + 16 * h) * sizeof(bf16);
+  const size_t smem = (size_t)(4 * h * h + 16 * h) * sizeof(bf16)
+                      + (size_t)16 * 4 * h * gates_x.element_size();
   TORCH_CHECK(smem <= 150 * 1024, "lstm fwd LDS overflow");
   const int grid = (B + 15) / 16;
   auto stream = cur_stream(gates_x);

@@ -107,8 +107,9 @@ __global__ __launch_bounds__(256) void tener_fwd_kernel(
     const bfrag aq = lds_frag_t(qu_s, m0, D, 0);
 #pragma unroll
     for (int nf = 0; nf < TMAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
       acc[nf] = mfma16t(aq, lds_frag_t(k_s, nf * 16, D, 0), acc[nf]);
+      }
     }
     float row_lse[4];
 #pragma unroll
@@ -118,7 +119,7 @@ __global__ __launch_bounds__(256) void tener_fwd_kernel(
       float mx = -1e30f;
 #pragma unroll
       for (int nf = 0; nf < TMAXNF; ++nf) {
-        if (nf >= NF) break;
+        if (nf < NF) {
         const int col = nf * 16 + (lane & 15);
         float s = -1e30f;
         if (col < len && row < L) {
@@ -127,32 +128,36 @@ __global__ __launch_bounds__(256) void tener_fwd_kernel(
         }
         acc[nf][r] = s;
         mx = fmaxf(mx, s);
+        }
       }
       mx = group16_reduce_max(mx);
       float sum = 0.f;
 #pragma unroll
       for (int nf = 0; nf < TMAXNF; ++nf) {
-        if (nf >= NF) break;
+        if (nf < NF) {
         const float p = __expf(acc[nf][r] - mx);
         acc[nf][r] = p;
         sum += p;
+        }
       }
       sum = group16_reduce_sum(sum);
       const float inv = __frcp_rn(sum);
 #pragma unroll
       for (int nf = 0; nf < TMAXNF; ++nf) {
-        if (nf >= NF) break;
+        if (nf < NF) {
         acc[nf][r] *= inv;
+        }
       }
       row_lse[r] = mx + __logf(sum);
     }
 #pragma unroll
     for (int nf = 0; nf < TMAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int lr = ((lane >> 4) << 2) + r;
         pw[lr * Lpad + nf * 16 + (lane & 15)] = __float2bfloat16(acc[nf][r]);
+      }
       }
     }
     if ((lane & 15) == 0) {
@@ -262,12 +267,13 @@ __global__ __launch_bounds__(256) void tener_bwd_kernel(
     const bfrag aq = lds_frag_t(qu_s, m0, D, 0);
 #pragma unroll
     for (int nf = 0; nf < TMAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
       acc[nf] = mfma16t(aq, lds_frag_t(k_s, nf * 16, D, 0), acc[nf]);
+      }
     }
 #pragma unroll
     for (int nf = 0; nf < TMAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int lr = ((lane >> 4) << 2) + r;
@@ -280,6 +286,7 @@ __global__ __launch_bounds__(256) void tener_bwd_kernel(
           p = __expf(s - lse_s[row]);
         }
         pt_s[(long)col * Lpad + row] = __float2bfloat16(p);
+      }
       }
     }
   }
@@ -323,12 +330,13 @@ __global__ __launch_bounds__(256) void tener_bwd_kernel(
     const bfrag ado = lds_frag_t(do_s, m0, D, 0);
 #pragma unroll
     for (int nf = 0; nf < TMAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
       acc[nf] = mfma16t(ado, lds_frag_t(v_s, nf * 16, D, 0), acc[nf]);
+      }
     }
 #pragma unroll
     for (int nf = 0; nf < TMAXNF; ++nf) {
-      if (nf >= NF) break;
+      if (nf < NF) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = m0 + ((lane >> 4) << 2) + r;
@@ -336,6 +344,7 @@ __global__ __launch_bounds__(256) void tener_bwd_kernel(
         const float p = to_f32(pt_s[(long)col * Lpad + row]);
         const float ds = p * (acc[nf][r] - delta_s[row]);  // unscaled attn
         pt_s[(long)col * Lpad + row] = __float2bfloat16(ds);
+      }
       }
     }
   }
@@ -411,9 +420,10 @@ __global__ __launch_bounds__(256) void tener_bwd_kernel(
 #pragma unroll
       for (int lr = 0; lr < 16; ++lr) {
         const int row = m0 + lr;
-        if (row >= L) break;
+        if (row < L) {
         const int off = col - row + L - 1;
         bdw[(long)lr * 2 * Lpad + off] = pt_s[(long)col * Lpad + row];
+        }
       }
     }
     // dqv tile = dBD @ R : [16, 2Lpad] x [2Lpad, D]

@@ -34,18 +34,22 @@ def test_bert_bilstm_crf_train_steps():
     from chinesener_amd.models import build_model
     from chinesener_amd.train.optimizers import (AdamWeightDecay,
                                                  build_param_groups)
+    from chinesener_amd.train.optimizers import LrSchedule, clip_gradients
     model = build_model("bert_bilstm_crf", _small_bert_params("bert_bilstm_crf"))
     model.to("cuda")
-    opt = AdamWeightDecay(build_param_groups(model, 2e-4, 0.01), lr=2e-4)
+    opt = AdamWeightDecay(build_param_groups(model, 1e-4, 0.01), lr=1e-4)
+    sched = LrSchedule("bert", 1e-4, num_train_steps=200, warmup_ratio=0.2)
     batch = make_synthetic_batch(8, 128, 10, vocab_size=2000, device="cuda")
     losses = []
-    for _ in range(12):
+    for step in range(1, 25):
         opt.zero_grad(set_to_none=True)
         with torch.autocast("cuda", dtype=torch.bfloat16):
             out = model(batch)
         out.loss.backward()
+        clip_gradients(model, "bert")      # global-norm 1.0, as the trainer
+        sched.apply(opt, step)
         opt.step()
-        losses.append(float(out.loss))
+        losses.append(float(out.loss.detach()))
     assert all(torch.isfinite(torch.tensor(losses))), losses
     assert losses[-1] < losses[0], losses
 
