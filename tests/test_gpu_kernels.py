@@ -286,7 +286,9 @@ def test_multi_tensor_adamw_matches_cpu_math():
     ref_m = [m.clone() for m in ms]
     ref_v = [v.clone() for v in vs]
     lr, b1, b2, eps, wd = 1e-3, 0.9, 0.999, 1e-6, 0.01
-    ops.get_ext().multi_tensor_adamw(params, grads, ms, vs, lr, b1, b2, eps, wd)
+    ops.get_ext().multi_tensor_adamw(params, grads, ms, vs,
+                                     [lr] * len(params), [wd] * len(params),
+                                     b1, b2, eps)
     for p, g, m, v in zip(ref_p, grads, ref_m, ref_v):
         m.mul_(b1).add_(g, alpha=1 - b1)
         v.mul_(b2).addcmul_(g, g, value=1 - b2)
