@@ -1,0 +1,117 @@
+"""L5 evaluation layer: strict span extraction, entity/tag reports,
+prediction post-processing, BIO entity decode (reference
+evaluation.py / tools/predict_utils.py behaviour)."""
+import numpy as np
+import pytest
+
+from chinesener_amd.eval import (decode_prediction, extract_entity,
+                                 fix_tokens, process_prediction)
+from chinesener_amd.eval.entity_eval import (entity_report, extract_spans,
+                                             tag_report)
+from chinesener_amd.eval.predict_utils import bio_extract_entity
+
+
+def test_extract_spans_bio():
+    tags = ["O", "B-PER", "I-PER", "O", "B-LOC", "B-LOC", "I-ORG"]
+    assert extract_spans(tags) == [("PER", 1, 3), ("LOC", 4, 5),
+                                   ("LOC", 5, 6), ("ORG", 6, 7)]
+
+
+def test_extract_spans_bies_cws():
+    # msr CWS adapter uses B/M/E/S (reference data/msr/preprocess.py:7-53)
+    tags = ["B", "E", "S", "B", "M", "E"]
+    assert extract_spans(tags) == [("SEG", 0, 2), ("SEG", 2, 3), ("SEG", 3, 6)]
+
+
+def test_entity_report_exact():
+    y_true = [["B-PER", "I-PER", "O", "B-LOC"]]
+    y_pred = [["B-PER", "I-PER", "O", "B-PER"]]
+    rep = entity_report(y_true, y_pred)
+    # PER: 1 tp of 2 predicted, 1 true -> P=0.5 R=1.0
+    assert rep["PER"]["precision"] == 0.5
+    assert rep["PER"]["recall"] == 1.0
+    assert rep["LOC"]["recall"] == 0.0
+    assert rep["micro avg"]["precision"] == 0.5
+    assert rep["micro avg"]["recall"] == 0.5
+    assert rep["weighted avg"]["support"] == 2
+
+
+def test_entity_report_strict_boundary():
+    # boundary mismatch is NOT a hit under strict matching
+    rep = entity_report([["B-PER", "I-PER", "I-PER"]], [["B-PER", "I-PER", "O"]])
+    assert rep["PER"]["f1"] == 0.0
+
+
+def test_tag_report():
+    rep = tag_report([["O", "B-PER"]], [["O", "O"]])
+    assert rep["O"]["recall"] == 1.0
+    assert rep["O"]["precision"] == 0.5
+    assert rep["B-PER"]["recall"] == 0.0
+
+
+def test_process_prediction_strips_specials():
+    idx2tag = {0: "[PAD]", 1: "[CLS]", 2: "[SEP]", 3: "O", 4: "B-PER"}
+    row = {"pred_ids": np.array([1, 4, 3, 2, 0]),
+           "label_ids": np.array([1, 4, 4, 2, 0]),
+           "mask": np.array([1, 1, 1, 1, 0])}
+    out = process_prediction(row, idx2tag)
+    assert out["label_tags"] == ["B-PER", "B-PER"]
+    assert out["pred_tags"] == ["B-PER", "O"]
+
+
+def test_process_prediction_maps_special_pred_to_O():
+    idx2tag = {0: "[PAD]", 3: "O", 4: "B-PER"}
+    row = {"pred_ids": np.array([0]), "label_ids": np.array([4]),
+           "mask": np.array([1])}
+    assert process_prediction(row, idx2tag)["pred_tags"] == ["O"]
+
+
+def test_decode_prediction_entities():
+    toks = list("张三在北京")
+    tags = ["B-PER", "I-PER", "O", "B-LOC", "I-LOC"]
+    ents = decode_prediction(toks, tags)
+    assert ents == {"PER": ["张三"], "LOC": ["北京"]}
+
+
+def test_decode_prediction_error_marker():
+    # B-PER followed by I-LOC: type mismatch flagged (predict_utils.py:12-13)
+    ents = decode_prediction(["a", "b"], ["B-PER", "I-LOC"])
+    assert ents["PER"] == ["ab[ERROR]"]
+
+
+def test_fix_tokens_wordpiece():
+    assert fix_tokens(["北", "##京", "[UNK]"]) == ["北", "京", "?"]
+
+
+def test_extract_entity_dedup():
+    ents = extract_entity(list("北京北京"),
+                          ["B-LOC", "I-LOC", "B-LOC", "I-LOC"])
+    assert ents == {"LOC": {"北京"}}
+
+
+def test_bio_extract_entity():
+    # reference mrc/evaluation.py:8-24
+    assert bio_extract_entity("张三去北京", ["B", "I", "O", "B", "I"]) == ["张三", "北京"]
+
+
+def test_evaluation_cli_roundtrip(tmp_path):
+    """main.py-dumped pkl → evaluation.py SingleEval report."""
+    import pickle
+    data_dir = tmp_path / "msra"
+    data_dir.mkdir()
+    from chinesener_amd.data.datasets import get_spec
+    spec = get_spec("msra")
+    tag2idx = spec.tag2idx
+    rows = []
+    for _ in range(4):
+        ids = [tag2idx["[CLS]"], tag2idx["B-PER"], tag2idx["I-PER"],
+               tag2idx["O"], tag2idx["[SEP]"], tag2idx["[PAD]"]]
+        rows.append({"pred_ids": np.array(ids), "label_ids": np.array(ids),
+                     "mask": np.array([1, 1, 1, 1, 1, 0])})
+    with open(data_dir / "m_predict.pkl", "wb") as f:
+        pickle.dump(rows, f)
+    import evaluation
+    ev = evaluation.SingleEval("m", "msra", str(tmp_path))
+    rep = ev.gen_report()
+    assert rep["micro avg"]["f1"] == 1.0
+    assert rep["PER"]["support"] == 4
