@@ -1,0 +1,167 @@
+"""hipGraph-captured inference engine.
+
+The reference serves a frozen SavedModel through TF-Serving
+(server.sh:1-5, export signature tools/train_utils.py:173-185). The
+MI355X-native equivalent loads the exported state_dict and captures the
+whole PREDICT path — embedding → BERT encoder → BiLSTM → logits →
+CRF Viterbi — into a hipGraph (torch.cuda.CUDAGraph on ROCm *is*
+hipGraphLaunch under the hood) per supported batch size, so a request
+replays one pre-built graph: zero launch latency for ~100 small kernels.
+
+Batch shapes are fixed at capture time; warmup requests (warmup.py,
+mirroring the reference's TF-Serving warmup records) establish them.
+Requests are padded up to the nearest captured batch size.
+"""
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional, Sequence
+
+import numpy as np
+import torch
+
+from ..config import EXPORT_DIR
+from .export import load_exported
+
+log = logging.getLogger("chinesener_amd.serve")
+
+# feature dtypes the models consume (see data/preprocess.py feature build)
+_INT_KEYS = ("token_ids", "segment_ids", "mask", "label_ids", "task_ids",
+             "softword_ids", "softlexicon_ids", "bichar_ids")
+
+
+class _CapturedGraph:
+    def __init__(self, graph, static_in: Dict[str, torch.Tensor],
+                 static_out: torch.Tensor):
+        self.graph = graph
+        self.static_in = static_in
+        self.static_out = static_out
+
+
+class InferenceEngine:
+    """Runs PREDICT for one exported model. GPU: hipGraph replay per
+    batch-size bucket; CPU (tests): eager no_grad."""
+
+    def __init__(self, name: str, export_root: str = EXPORT_DIR,
+                 version: Optional[int] = None,
+                 batch_sizes: Sequence[int] = (1, 4, 8),
+                 max_seq_len: Optional[int] = None,
+                 use_graph: Optional[bool] = None,
+                 device: Optional[str] = None):
+        self.name = name
+        self.device = device or ("cuda" if torch.cuda.is_available() else "cpu")
+        self.model, self.params = load_exported(name, export_root,
+                                                device=self.device,
+                                                version=version)
+        self.model.eval()
+        self.max_seq_len = max_seq_len or self.params.get("max_seq_len", 150)
+        self.batch_sizes = sorted(batch_sizes)
+        self.use_graph = (use_graph if use_graph is not None
+                          else self.device.startswith("cuda"))
+        self._graphs: Dict[int, _CapturedGraph] = {}
+        self.n_requests = 0
+
+    # ---------------------------------------------------------- capture
+    def _example_features(self, batch: int) -> Dict[str, torch.Tensor]:
+        """Static input buffers for one bucket, shaped like the exported
+        model's feature schema (word-enhance keys included when the model
+        name demands them, reference base_preprocess.py:22-33)."""
+        L = self.max_seq_len
+        feats = {
+            "token_ids": torch.ones(batch, L, dtype=torch.long),
+            "segment_ids": torch.zeros(batch, L, dtype=torch.long),
+            "mask": torch.ones(batch, L, dtype=torch.long),
+        }
+        name = self.name
+        if "softword" in name and "ex_softword" not in name:
+            feats["softword_ids"] = torch.zeros(batch, L, dtype=torch.long)
+        if "ex_softword" in name:
+            feats["ex_softword_ids"] = torch.zeros(batch, L, 5)
+        if "softlexicon" in name:
+            feats["softlexicon_ids"] = torch.zeros(batch, L, 40, dtype=torch.long)
+            feats["softlexicon_weights"] = torch.zeros(batch, L, 40)
+        if "bichar" in name:
+            feats["bichar_ids"] = torch.zeros(batch, L, dtype=torch.long)
+        if "mtl" in name or "adv" in name:
+            feats["task_ids"] = torch.ones(batch, L, dtype=torch.long)
+        return {k: v.to(self.device) for k, v in feats.items()}
+
+    @torch.no_grad()
+    def _capture(self, batch: int) -> _CapturedGraph:
+        static_in = self._example_features(batch)
+        # two eager warmup passes on a side stream (allocator steady-state)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                out = self.model(static_in, compute_pred=True)
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            out = self.model(static_in, compute_pred=True)
+            static_out = out.pred_ids
+        log.info("hipGraph captured for %s batch=%d L=%d", self.name, batch,
+                 self.max_seq_len)
+        return _CapturedGraph(graph, static_in, static_out)
+
+    def warmup(self, requests: Optional[List[Dict[str, np.ndarray]]] = None):
+        """Capture all batch buckets (optionally driven by recorded warmup
+        requests, whose batch sizes define the buckets — the reference's
+        assets.extra/tf_serving_warmup_requests role)."""
+        if requests:
+            sizes = sorted({r["token_ids"].shape[0] for r in requests})
+            self.batch_sizes = sorted(set(self.batch_sizes) | set(sizes))
+        if self.use_graph:
+            for b in self.batch_sizes:
+                if b not in self._graphs:
+                    self._graphs[b] = self._capture(b)
+        if requests:
+            for r in requests:
+                self.predict(r)
+
+    # ---------------------------------------------------------- predict
+    def _bucket(self, batch: int) -> Optional[int]:
+        for b in self.batch_sizes:
+            if b >= batch:
+                return b
+        return None
+
+    @torch.no_grad()
+    def predict(self, features: Dict[str, np.ndarray]) -> np.ndarray:
+        """features: numpy arrays [B, ...] (as built by
+        data.preprocess.BasicProc.build_seq_feature, stacked).
+        Returns pred_ids [B, L]."""
+        self.n_requests += 1
+        batch = features["token_ids"].shape[0]
+        tensors = {}
+        for k, v in features.items():
+            if k == "label_ids":
+                continue  # PREDICT path never consumes labels
+            a = np.asarray(v)
+            t = torch.as_tensor(a)
+            if k in _INT_KEYS or a.dtype.kind in "iu":
+                t = t.long()
+            else:
+                t = t.float()
+            tensors[k] = t
+
+        bucket = self._bucket(batch) if self.use_graph else None
+        if bucket is not None and bucket in self._graphs:
+            g = self._graphs[bucket]
+            for k, t in tensors.items():
+                if k not in g.static_in:
+                    continue
+                buf = g.static_in[k]
+                buf.zero_()
+                sl = [slice(0, s) for s in t.shape]
+                buf[tuple(sl)].copy_(t.to(self.device), non_blocking=True)
+            if "mask" not in tensors:
+                g.static_in["mask"].zero_()
+            g.graph.replay()
+            torch.cuda.synchronize()
+            return g.static_out[:batch].cpu().numpy()
+
+        # eager path (CPU, or batch larger than any bucket)
+        dev = {k: t.to(self.device) for k, t in tensors.items()}
+        out = self.model(dev, compute_pred=True)
+        return out.pred_ids.cpu().numpy()

@@ -47,8 +47,12 @@ def latest_version_dir(name: str, export_root: str = EXPORT_DIR) -> str:
 
 
 def load_exported(name: str, export_root: str = EXPORT_DIR,
-                  device: str = "cpu") -> Tuple[torch.nn.Module, Dict]:
-    vdir = latest_version_dir(name, export_root)
+                  device: str = "cpu",
+                  version: Optional[int] = None) -> Tuple[torch.nn.Module, Dict]:
+    if version is not None:
+        vdir = os.path.join(export_root, name, str(version))
+    else:
+        vdir = latest_version_dir(name, export_root)
     with open(os.path.join(vdir, "params.pkl"), "rb") as f:
         meta = pickle.load(f)
     model_name = meta["model_name"]
