@@ -59,6 +59,11 @@ def main():
         "num_train_steps": 100000, "step_per_epoch": 1000,
     })
     model = build_model(args.model, params).to(device)
+    use_bf16 = use_gpu and params.get("dtype", "bf16") == "bf16"
+    if use_bf16:
+        # pure-bf16 weights + fp32 masters in the fused optimizer
+        from chinesener_amd.train.precision import convert_bf16_mixed
+        convert_bf16_mixed(model)
     dp = None
     if world > 1:
         from chinesener_amd.dist import BucketedDataParallel
@@ -73,8 +78,13 @@ def main():
                                     device=device)
                for i in range(8)]
 
-    use_bf16 = use_gpu and params.get("dtype", "bf16") == "bf16"
     step_num = 0
+
+    def cast(b):
+        if not use_bf16:
+            return b
+        return {k: v.to(torch.bfloat16) if v.is_floating_point() else v
+                for k, v in b.items()}
 
     def train_step(batch):
         nonlocal step_num
@@ -82,11 +92,7 @@ def main():
             dp.zero_grad()
         else:
             opt.zero_grad(set_to_none=True)
-        if use_bf16:
-            with torch.autocast("cuda", dtype=torch.bfloat16):
-                out = model(batch)
-        else:
-            out = model(batch)
+        out = model(cast(batch))
         out.loss.backward()
         if dp is not None:
             dp.finalize_backward()

@@ -74,7 +74,9 @@ class AdamWeightDecay(torch.optim.Optimizer):
     @torch.no_grad()
     def step(self, closure=None):
         loss = closure() if closure is not None else None
-        fused = []   # (params, grads, ms, vs, lrs, wds) across ALL groups
+        # fused bins by param dtype: bf16 params carry an fp32 master copy
+        # (pure-bf16 training mode — no autocast weight casts per step)
+        fused = {torch.bfloat16: [], torch.float32: []}
         for group in self.param_groups:
             params = [p for p in group["params"] if p.grad is not None]
             if not params:
@@ -83,16 +85,33 @@ class AdamWeightDecay(torch.optim.Optimizer):
             states = [self.state[p] for p in params]
             for p, s in zip(params, states):
                 if "m" not in s:
-                    s["m"] = torch.zeros_like(p)
-                    s["v"] = torch.zeros_like(p)
+                    s["m"] = torch.zeros(p.shape, dtype=torch.float32,
+                                         device=p.device)
+                    s["v"] = torch.zeros_like(s["m"])
+                    if p.dtype == torch.bfloat16:
+                        s["master"] = p.detach().float().clone()
             ms = [s["m"] for s in states]
             vs = [s["v"] for s in states]
             lr = group["lr"] * group.get("lr_scale", 1.0)
             b1, b2 = group["betas"]
             eps, wd = group["eps"], group["weight_decay"]
             if params[0].is_cuda and ops.ext_available():
-                fused.append((params, grads, ms, vs,
-                              [lr] * len(params), [wd] * len(params)))
+                masters = ([s["master"] for s in states]
+                           if params[0].dtype == torch.bfloat16 else [])
+                fused[params[0].dtype].append(
+                    (params, [g.contiguous() for g in grads], masters, ms, vs,
+                     [lr] * len(params), [wd] * len(params)))
+            elif params[0].dtype == torch.bfloat16:
+                # eager master-weight path (CPU/debug)
+                for p, g, s in zip(params, grads, states):
+                    gf = g.float()
+                    s["m"].mul_(b1).add_(gf, alpha=1 - b1)
+                    s["v"].mul_(b2).addcmul_(gf, gf, value=1 - b2)
+                    upd = s["m"] / (s["v"].sqrt() + eps)
+                    if wd:
+                        upd = upd + wd * s["master"]
+                    s["master"].add_(upd, alpha=-lr)
+                    p.copy_(s["master"].to(p.dtype))
             else:
                 torch._foreach_mul_(ms, b1)
                 torch._foreach_add_(ms, grads, alpha=1 - b1)
@@ -104,13 +123,16 @@ class AdamWeightDecay(torch.optim.Optimizer):
                 if wd:
                     torch._foreach_add_(update, params, alpha=wd)
                 torch._foreach_add_(params, update, alpha=-lr)
-        if fused:
-            # one multi-tensor kernel launch for every group (per-tensor lr)
+        for dtype, bins in fused.items():
+            if not bins:
+                continue
+            # one multi-tensor kernel launch per dtype (per-tensor lr)
             b1, b2 = self.param_groups[0]["betas"]
             eps = self.param_groups[0]["eps"]
-            cat = [sum((f[i] for f in fused), []) for i in range(6)]
+            cat = [sum((f[i] for f in bins), []) for i in range(7)]
             ops.get_ext().multi_tensor_adamw(cat[0], cat[1], cat[2], cat[3],
-                                             cat[4], cat[5], b1, b2, eps)
+                                             cat[4], cat[5], cat[6],
+                                             b1, b2, eps)
         return loss
 
 
@@ -152,11 +174,29 @@ class LrSchedule:
         return lr
 
 
-def clip_gradients(model: torch.nn.Module, family: str) -> Optional[float]:
+def clip_gradients(model: torch.nn.Module, family: str,
+                   max_norm: float = 1.0) -> Optional[torch.Tensor]:
     """bert/transformer: clip_by_global_norm 1.0 (:315,334); custom:
-    clip_by_value +-5 (:379-390). Returns grad-norm if computed."""
+    clip_by_value +-5 (:379-390).
+
+    GPU path is fully device-side (fused multi-tensor sumsq + scale, no
+    host sync in the step loop); returns the squared-norm tensor."""
     if family in ("bert", "transformer"):
-        return float(torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0))
+        ps = [p for p in model.parameters() if p.grad is not None]
+        if not ps:
+            return None
+        for p in ps:            # scale must hit the real grad storage
+            if not p.grad.is_contiguous():
+                p.grad = p.grad.contiguous()
+        grads = [p.grad for p in ps]
+        if grads[0].is_cuda and ops.ext_available():
+            ext = ops.get_ext()
+            sumsq = ext.multi_tensor_sumsq(grads)
+            norm = sumsq.sqrt()
+            coef = (max_norm / (norm + 1e-6)).clamp(max=1.0)
+            ext.multi_tensor_scale(grads, coef)
+            return sumsq
+        return torch.nn.utils.clip_grad_norm_(model.parameters(), max_norm)
     torch.nn.utils.clip_grad_value_(model.parameters(), 5.0)
     return None
 

@@ -17,6 +17,7 @@ from ..models import ModelOutput, NerModel, optimizer_family
 from .checkpoints import CheckpointManager
 from .metrics import TagMetrics
 from .optimizers import build_optimizer, clip_gradients
+from .precision import convert_bf16_mixed, wants_pure_bf16
 
 log = logging.getLogger("chinesener_amd")
 
@@ -31,17 +32,27 @@ class Trainer:
         self.rank = rank
         self.device = device or ("cuda" if torch.cuda.is_available() else "cpu")
         self.model.to(self.device)
+        # pure-bf16 weights + fp32 masters in the fused optimizer (no
+        # autocast weight-cast kernels; see train/precision.py)
+        self.pure_bf16 = wants_pure_bf16(params, self.device)
+        if self.pure_bf16:
+            convert_bf16_mixed(self.model)
         self.family = optimizer_family(model_name)
         self.optimizer, self.schedule = build_optimizer(model, self.family, params)
         self.ckpt = CheckpointManager(ckpt_dir,
                                       params.get("keep_checkpoint_max", 3))
         self.dp = dp_engine
-        self.use_bf16 = (self.device.startswith("cuda")
-                         and params.get("dtype", "bf16") == "bf16")
+        self.use_bf16 = False  # autocast replaced by pure-bf16 weights
         self.step = self.ckpt.restore(model, self.optimizer,
                                       map_location=self.device)
 
     # ------------------------------------------------------------- train
+    def _cast(self, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+        if not self.pure_bf16:
+            return batch
+        return {k: v.to(torch.bfloat16) if v.is_floating_point() else v
+                for k, v in batch.items()}
+
     def _forward(self, batch: Dict[str, torch.Tensor]) -> ModelOutput:
         if self.use_bf16:
             with torch.autocast("cuda", dtype=torch.bfloat16):
@@ -49,7 +60,8 @@ class Trainer:
         return self.model(batch)
 
     def train_step(self, batch: Dict[str, torch.Tensor]) -> float:
-        batch = {k: v.to(self.device, non_blocking=True) for k, v in batch.items()}
+        batch = self._cast({k: v.to(self.device, non_blocking=True)
+                            for k, v in batch.items()})
         if self.dp is not None:
             self.dp.zero_grad()
         else:
@@ -110,7 +122,7 @@ class Trainer:
                              idx2tag or self.params.get("idx2tag"))
         total_loss, n = 0.0, 0
         for batch in batches:
-            batch = {k: v.to(self.device) for k, v in batch.items()}
+            batch = self._cast({k: v.to(self.device) for k, v in batch.items()})
             if self.use_bf16:
                 with torch.autocast("cuda", dtype=torch.bfloat16):
                     out = self.model(batch, compute_pred=True)
@@ -133,7 +145,7 @@ class Trainer:
         self.model.eval()
         out_rows = []
         for batch in batches:
-            dev = {k: v.to(self.device) for k, v in batch.items()}
+            dev = self._cast({k: v.to(self.device) for k, v in batch.items()})
             if self.use_bf16:
                 with torch.autocast("cuda", dtype=torch.bfloat16):
                     out = self.model(dev, compute_pred=True)

@@ -3,13 +3,21 @@
 // decoupled weight decay — reference tools/train_utils.py:276-284):
 //   m = b1*m + (1-b1)*g ; v = b2*v + (1-b2)*g^2
 //   p -= lr * (m / (sqrt(v)+eps) + wd*p)
+//
+// Two parameter modes:
+//  - fp32 params: p/g/m/v all fp32 (CPU-parity path)
+//  - bf16 params: model weights + grads bf16, fp32 MASTER copy updated in
+//    fp32 and rounded back to the bf16 weight — the MI355X-native "pure
+//    bf16" training mode (no autocast weight-cast kernels per step).
+// Plus a fused multi-tensor L2 norm for the global-norm gradient clip.
 #include "common.h"
 
 #define MAX_TENSORS 512
 
 struct ChunkMeta {
-  float* p;
-  float* g;
+  void* p;        // bf16 or fp32 weight
+  const void* g;  // grad, same dtype as p
+  float* master;  // fp32 master (nullptr when p is fp32)
   float* m;
   float* v;
   long n;
@@ -17,54 +25,183 @@ struct ChunkMeta {
   float wd;
 };
 
+template <typename T>
 __global__ void multi_tensor_adamw_kernel(ChunkMeta* metas, int n_tensors,
                                           float b1, float b2, float eps) {
-  // grid-stride over (tensor, element) pairs: block handles slices of one
-  // tensor chosen by blockIdx.y-style flattening
   for (int ti = blockIdx.y; ti < n_tensors; ti += gridDim.y) {
     const ChunkMeta mt = metas[ti];
     const float lr = mt.lr;
     const float wd = mt.wd;
+    T* p = reinterpret_cast<T*>(mt.p);
+    const T* g = reinterpret_cast<const T*>(mt.g);
     const long stride = (long)gridDim.x * blockDim.x;
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < mt.n;
          i += stride) {
-      const float g = mt.g[i];
-      const float m = b1 * mt.m[i] + (1.f - b1) * g;
-      const float v = b2 * mt.v[i] + (1.f - b2) * g * g;
+      const float gv = to_f32(g[i]);
+      const float m = b1 * mt.m[i] + (1.f - b1) * gv;
+      const float v = b2 * mt.v[i] + (1.f - b2) * gv * gv;
       mt.m[i] = m;
       mt.v[i] = v;
-      mt.p[i] -= lr * (m * __frcp_rn(sqrtf(v) + eps) + wd * mt.p[i]);
+      const float pv = mt.master ? mt.master[i] : to_f32(p[i]);
+      const float upd = pv - lr * (m * __frcp_rn(sqrtf(v) + eps) + wd * pv);
+      if (mt.master) mt.master[i] = upd;
+      from_f32(upd, &p[i]);
     }
   }
 }
 
+static void launch_adamw(std::vector<ChunkMeta>& metas, long total,
+                         bool is_bf16, const at::Tensor& ref, double b1,
+                         double b2, double eps) {
+  const int n = metas.size();
+  auto meta_blob = at::from_blob(metas.data(), {(long)(n * sizeof(ChunkMeta))},
+                                 at::TensorOptions().dtype(at::kByte))
+                       .to(ref.device(), /*non_blocking=*/false);
+  const int block = 256;
+  const int gx = std::min<long>((total / n + block - 1) / block, 512);
+  dim3 grid(std::max(gx, 1), std::min(n, 64));
+  auto stream = cur_stream(ref);
+  if (is_bf16) {
+    hipLaunchKernelGGL(multi_tensor_adamw_kernel<bf16>, grid, dim3(block), 0,
+                       stream, (ChunkMeta*)meta_blob.data_ptr(), n, (float)b1,
+                       (float)b2, (float)eps);
+  } else {
+    hipLaunchKernelGGL(multi_tensor_adamw_kernel<float>, grid, dim3(block), 0,
+                       stream, (ChunkMeta*)meta_blob.data_ptr(), n, (float)b1,
+                       (float)b2, (float)eps);
+  }
+  HIP_CHECK_LAST();
+}
+
+// fp32 path (masters empty) or bf16 path (masters[i] fp32, same numel).
 void multi_tensor_adamw(std::vector<at::Tensor> params,
                         std::vector<at::Tensor> grads,
+                        std::vector<at::Tensor> masters,
                         std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
                         std::vector<double> lrs, std::vector<double> wds,
                         double b1, double b2, double eps) {
   const int n = params.size();
   TORCH_CHECK(n > 0 && n <= MAX_TENSORS, "multi_tensor_adamw: bad tensor count");
+  const bool has_master = !masters.empty();
+  const bool is_bf16 = params[0].scalar_type() == at::kBFloat16;
+  TORCH_CHECK(!is_bf16 || has_master,
+              "adamw: bf16 params need fp32 master weights");
   std::vector<ChunkMeta> metas(n);
   long total = 0;
   for (int i = 0; i < n; ++i) {
-    TORCH_CHECK(params[i].scalar_type() == at::kFloat,
-                "adamw: fp32 master params expected");
-    TORCH_CHECK(grads[i].is_contiguous(), "adamw: non-contiguous grad");
-    metas[i] = {params[i].data_ptr<float>(), grads[i].data_ptr<float>(),
+    TORCH_CHECK(params[i].scalar_type() ==
+                    (is_bf16 ? at::kBFloat16 : at::kFloat),
+                "adamw: mixed param dtypes in one call");
+    TORCH_CHECK(grads[i].is_contiguous() &&
+                    grads[i].scalar_type() == params[i].scalar_type(),
+                "adamw: grad must be contiguous, same dtype as param");
+    metas[i] = {params[i].data_ptr(), grads[i].data_ptr(),
+                has_master ? masters[i].data_ptr<float>() : nullptr,
                 ms[i].data_ptr<float>(), vs[i].data_ptr<float>(),
                 params[i].numel(), (float)lrs[i], (float)wds[i]};
     total += metas[i].n;
   }
-  auto meta_blob = at::from_blob(metas.data(), {(long)(n * sizeof(ChunkMeta))},
+  launch_adamw(metas, total, is_bf16, params[0], b1, b2, eps);
+}
+
+// ------------------------------------------------------- global L2 norm
+struct NormMeta {
+  const void* g;
+  long n;
+  int is_bf16;
+};
+
+__global__ void multi_tensor_l2_kernel(NormMeta* metas, int n_tensors,
+                                       float* out) {
+  float acc = 0.f;
+  for (int ti = blockIdx.y; ti < n_tensors; ti += gridDim.y) {
+    const NormMeta mt = metas[ti];
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < mt.n;
+         i += stride) {
+      const float v = mt.is_bf16
+                          ? to_f32(reinterpret_cast<const bf16*>(mt.g)[i])
+                          : reinterpret_cast<const float*>(mt.g)[i];
+      acc += v * v;
+    }
+  }
+  // wave reduce then one atomic per wave
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, WAVE);
+  if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(out, acc);
+}
+
+// Returns a 1-element fp32 tensor holding sum of squares (caller sqrts).
+at::Tensor multi_tensor_sumsq(std::vector<at::Tensor> grads) {
+  const int n = grads.size();
+  TORCH_CHECK(n > 0 && n <= MAX_TENSORS, "multi_tensor_sumsq: bad count");
+  std::vector<NormMeta> metas(n);
+  long total = 0;
+  for (int i = 0; i < n; ++i) {
+    TORCH_CHECK(grads[i].is_contiguous(), "sumsq: non-contiguous grad");
+    metas[i] = {grads[i].data_ptr(), grads[i].numel(),
+                grads[i].scalar_type() == at::kBFloat16 ? 1 : 0};
+    total += metas[i].n;
+  }
+  auto out = at::zeros({1}, grads[0].options().dtype(at::kFloat));
+  auto meta_blob = at::from_blob(metas.data(), {(long)(n * sizeof(NormMeta))},
                                  at::TensorOptions().dtype(at::kByte))
-                       .to(params[0].device(), /*non_blocking=*/false);
+                       .to(grads[0].device(), false);
   const int block = 256;
-  const int gx = std::min<long>((total / n + block - 1) / block, 512);
+  const int gx = std::min<long>((total / n + block - 1) / block, 256);
   dim3 grid(std::max(gx, 1), std::min(n, 64));
-  hipLaunchKernelGGL(multi_tensor_adamw_kernel, grid, dim3(block), 0,
-                     cur_stream(params[0]),
-                     (ChunkMeta*)meta_blob.data_ptr(), n, (float)b1,
-                     (float)b2, (float)eps);
+  hipLaunchKernelGGL(multi_tensor_l2_kernel, grid, dim3(block), 0,
+                     cur_stream(grads[0]), (NormMeta*)meta_blob.data_ptr(), n,
+                     out.data_ptr<float>());
+  HIP_CHECK_LAST();
+  return out;
+}
+
+// Scale every grad in-place by *scale (1-element fp32 tensor, device side —
+// no host sync in the clip path).
+struct ScaleMeta {
+  void* g;
+  long n;
+  int is_bf16;
+};
+
+__global__ void multi_tensor_scale_kernel(ScaleMeta* metas, int n_tensors,
+                                          const float* scale) {
+  const float s = *scale;
+  for (int ti = blockIdx.y; ti < n_tensors; ti += gridDim.y) {
+    const ScaleMeta mt = metas[ti];
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < mt.n;
+         i += stride) {
+      if (mt.is_bf16) {
+        bf16* g = reinterpret_cast<bf16*>(mt.g);
+        g[i] = __float2bfloat16(to_f32(g[i]) * s);
+      } else {
+        reinterpret_cast<float*>(mt.g)[i] *= s;
+      }
+    }
+  }
+}
+
+void multi_tensor_scale(std::vector<at::Tensor> grads,
+                        const at::Tensor& scale) {
+  const int n = grads.size();
+  TORCH_CHECK(n > 0 && n <= MAX_TENSORS, "multi_tensor_scale: bad count");
+  std::vector<ScaleMeta> metas(n);
+  long total = 0;
+  for (int i = 0; i < n; ++i) {
+    metas[i] = {grads[i].data_ptr(), grads[i].numel(),
+                grads[i].scalar_type() == at::kBFloat16 ? 1 : 0};
+    total += metas[i].n;
+  }
+  auto meta_blob = at::from_blob(metas.data(), {(long)(n * sizeof(ScaleMeta))},
+                                 at::TensorOptions().dtype(at::kByte))
+                       .to(grads[0].device(), false);
+  const int block = 256;
+  const int gx = std::min<long>((total / n + block - 1) / block, 256);
+  dim3 grid(std::max(gx, 1), std::min(n, 64));
+  hipLaunchKernelGGL(multi_tensor_scale_kernel, grid, dim3(block), 0,
+                     cur_stream(grads[0]), (ScaleMeta*)meta_blob.data_ptr(), n,
+                     scale.data_ptr<float>());
   HIP_CHECK_LAST();
 }

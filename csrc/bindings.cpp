@@ -33,8 +33,10 @@ std::vector<at::Tensor> softlexicon_bwd(const at::Tensor&, const at::Tensor&,
 // adam.hip
 void multi_tensor_adamw(std::vector<at::Tensor>, std::vector<at::Tensor>,
                         std::vector<at::Tensor>, std::vector<at::Tensor>,
-                        std::vector<double>, std::vector<double>, double,
-                        double, double);
+                        std::vector<at::Tensor>, std::vector<double>,
+                        std::vector<double>, double, double, double);
+at::Tensor multi_tensor_sumsq(std::vector<at::Tensor>);
+void multi_tensor_scale(std::vector<at::Tensor>, const at::Tensor&);
 // attention.hip
 std::vector<at::Tensor> attn_fwd(const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&, double);
@@ -65,6 +67,11 @@ std::vector<at::Tensor> lstm_bwd(const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&, bool,
                                  bool);
+std::vector<at::Tensor> bilstm_fwd_l(const at::Tensor&, const at::Tensor&,
+                                     const at::Tensor&, bool);
+at::Tensor bilstm_bwd_l(const at::Tensor&, const at::Tensor&,
+                        const at::Tensor&, const at::Tensor&,
+                        const at::Tensor&, bool);
 
 PYBIND11_MODULE(_hip_ops, m) {
   m.doc() = "chinesener_amd gfx950 HIP kernels";
@@ -80,6 +87,8 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("softlexicon_fwd", &softlexicon_fwd);
   m.def("softlexicon_bwd", &softlexicon_bwd);
   m.def("multi_tensor_adamw", &multi_tensor_adamw);
+  m.def("multi_tensor_sumsq", &multi_tensor_sumsq);
+  m.def("multi_tensor_scale", &multi_tensor_scale);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("attn_fwd_qkv", &attn_fwd_qkv);
@@ -89,4 +98,6 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("mfma_probe", &mfma_probe);
   m.def("lstm_fwd", &lstm_fwd);
   m.def("lstm_bwd", &lstm_bwd);
+  m.def("bilstm_fwd", &bilstm_fwd_l);
+  m.def("bilstm_bwd", &bilstm_bwd_l);
 }

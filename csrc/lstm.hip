@@ -1,10 +1,15 @@
-// Persistent LSTM recurrence (the "BiLSTM tagger" kernel
-// named in BASELINE.json). The x-projection (x @ W_ih + b) is a single
-// library GEMM done by the Python wrapper; these kernels own the
+// Persistent BiLSTM recurrence (the "BiLSTM tagger" kernel
+// named in BASELINE.json). The x-projection (x @ [W_ih_f|W_ih_b] + b) is a
+// single library GEMM done by the Python wrapper; these kernels own the
 // sequential part: per step  gates = gates_x[t] + h_{t-1} @ W_hh,
 // LSTM cell update, variable-length masking.
 //
-// Design: recurrent weights stay resident in LDS for the whole sequence
+// Design: BOTH directions run in ONE launch (blockIdx.y = direction) so
+// their sequence loops overlap on different CUs instead of serializing on
+// the stream — with B=64 each direction is only B/16 workgroups, far under
+// the 256 CUs. Inputs/outputs are strided so the python-side layout is
+// [B,L,8h] gates (fw|bw halves) and [B,L,2h] hidden (the BiLSTM concat is
+// free). Recurrent weights stay resident in LDS for the whole sequence
 // (W_hh bf16 <= 128 KiB for h <= 128); one workgroup (4 waves) owns a
 // 16-row batch tile and loops time in-kernel — no per-step launches.
 // MFMA 16x16x32 bf16 computes [16,4h] gates; wave w owns hidden slice
@@ -34,14 +39,18 @@ __device__ __forceinline__ float dact_from_out(float y, bool relu) {
   return relu ? (y > 0.f ? 1.f : 0.f) : (1.f - y * y);
 }
 
+// dir = blockIdx.y. gates_x rows have stride gxs with this direction's
+// slice at column offset dir*4h; hs rows have stride hss, offset dir*h.
+// cs/gates_out are per-direction contiguous ([D,B,L,h] / [D,B,L,4h]).
+// Direction 1 (when D=2) scans the sequence reversed.
 template <typename T>
 __global__ __launch_bounds__(256) void lstm_fwd_kernel(
-    const T* __restrict__ gates_x,  // [B,L,4h]
-    const bf16* __restrict__ w_hh_t,  // [4h,h] (transposed by wrapper)
-    const int* __restrict__ lens, T* __restrict__ hs,  // [B,L,h]
-    float* __restrict__ cs,                            // [B,L,h]
-    float* __restrict__ gates_out,                     // [B,L,4h] activated
-    int B, int L, int h, bool reverse, bool relu) {
+    const T* __restrict__ gates_x,    // [B,L,gxs]
+    const bf16* __restrict__ w_hh_t,  // [D,4h,h] (transposed by wrapper)
+    const int* __restrict__ lens, T* __restrict__ hs,  // [B,L,hss]
+    float* __restrict__ cs,                            // [D,B,L,h]
+    float* __restrict__ gates_out,                     // [D,B,L,4h] activated
+    int B, int L, int h, int gxs, int hss, bool rev0, bool relu) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   bf16* wT = reinterpret_cast<bf16*>(smem_raw);  // [4h][h]
   bf16* hb = wT + 4 * h * h;                     // [16][h] current h (bf16)
@@ -49,9 +58,15 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int b0 = blockIdx.x * 16;
+  const int dir = blockIdx.y;
+  const bool reverse = dir ? true : rev0;
+  const int gx_off = dir * 4 * h;
+  const int hs_off = dir * h;
+  const long dbase = (long)dir * B * L;  // row offset into cs/gates_out
 
   for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x)
-    reinterpret_cast<s16x8*>(wT)[i] = reinterpret_cast<const s16x8*>(w_hh_t)[i];
+    reinterpret_cast<s16x8*>(wT)[i] =
+        reinterpret_cast<const s16x8*>(w_hh_t + (long)dir * 4 * h * h)[i];
   for (int i = threadIdx.x; i < 16 * h / 8; i += blockDim.x)
     reinterpret_cast<s16x8*>(hb)[i] = s16x8{};
   __syncthreads();
@@ -70,18 +85,15 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
 
   for (int step = 0; step < L; ++step) {
     const int t = reverse ? (L - 1 - step) : step;
-    // cooperative staging of gates_x[:,t,:] for the 16 rows (vectorized,
-    // overlaps the recurrent MFMAs below; barrier covers both)
+    // cooperative staging of gates_x[:,t,dir-slice] for the 16 rows
     {
       const int nv = 16 * 4 * h / 8;
       for (int i = threadIdx.x; i < nv; i += blockDim.x) {
         const int row = (i * 8) / (4 * h);
         const int col = (i * 8) % (4 * h);
         const int b = b0 + row;
-        s16x8 val{};
-        f32x2 dummy;
         if (b < B) {
-          const long g = ((long)b * L + t) * 4 * h + col;
+          const long g = ((long)b * L + t) * gxs + gx_off + col;
           if (sizeof(T) == 2) {
             reinterpret_cast<s16x8*>(gx_s)[i] =
                 *reinterpret_cast<const s16x8*>(gates_x + g);
@@ -92,7 +104,6 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
                 *reinterpret_cast<const f32x4*>(gates_x + g + 4);
           }
         }
-        (void)val; (void)dummy;
       }
     }
     // gates = h_prev @ W_hh  (+ gates_x added in the epilogue)
@@ -119,7 +130,7 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
       const int b = b0 + lrow + r;
       if (b >= B) continue;
       const bool valid = t < mylen[r];
-      const long gbase = ((long)b * L + t) * 4 * h;
+      const long gbase = (dbase + (long)b * L + t) * 4 * h;
 #pragma unroll
       for (int jj = 0; jj < 2; ++jj) {
         const int j = wid * 32 + jj * 16 + (lane & 15);
@@ -136,9 +147,9 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
         const float c_new = gf * c_reg[r][jj] + gi * gg;
         const float h_new = go * act_f(c_new, relu);
         if (valid) c_reg[r][jj] = c_new;
-        const long obase = ((long)b * L + t) * h + j;
+        const long obase = ((long)b * L + t) * hss + hs_off + j;
         from_f32(valid ? h_new : 0.f, &hs[obase]);
-        cs[obase] = c_reg[r][jj];
+        cs[(dbase + (long)b * L + t) * h + j] = c_reg[r][jj];
         gates_out[gbase + 0 * h + j] = gi;
         gates_out[gbase + 1 * h + j] = gf;
         gates_out[gbase + 2 * h + j] = gg;
@@ -153,20 +164,26 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
 
 template <typename T>
 __global__ __launch_bounds__(256) void lstm_bwd_kernel(
-    const T* __restrict__ dhs,        // [B,L,h] upstream grad
-    const float* __restrict__ cs,     // [B,L,h] carried cell states
-    const float* __restrict__ gates,  // [B,L,4h] activated
-    const bf16* __restrict__ w_hh,    // [h,4h] (original layout)
-    const int* __restrict__ lens, T* __restrict__ dgates_x,  // [B,L,4h]
-    int B, int L, int h, bool reverse, bool relu) {
+    const T* __restrict__ dhs,        // [B,L,hss] upstream grad (strided)
+    const float* __restrict__ cs,     // [D,B,L,h] carried cell states
+    const float* __restrict__ gates,  // [D,B,L,4h] activated
+    const bf16* __restrict__ w_hh,    // [D,h,4h] (original layout)
+    const int* __restrict__ lens, T* __restrict__ dgates_x,  // [B,L,gxs]
+    int B, int L, int h, int gxs, int hss, bool rev0, bool relu) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   bf16* w_s = reinterpret_cast<bf16*>(smem_raw);  // [h][4h]
   bf16* dg_s = w_s + 4 * h * h;                   // [16][4h]
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int b0 = blockIdx.x * 16;
+  const int dir = blockIdx.y;
+  const bool reverse = dir ? true : rev0;
+  const int gx_off = dir * 4 * h;
+  const int hs_off = dir * h;
+  const long dbase = (long)dir * B * L;
   for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x)
-    reinterpret_cast<s16x8*>(w_s)[i] = reinterpret_cast<const s16x8*>(w_hh)[i];
+    reinterpret_cast<s16x8*>(w_s)[i] =
+        reinterpret_cast<const s16x8*>(w_hh + (long)dir * 4 * h * h)[i];
   __syncthreads();
 
   const int lrow = ((lane >> 4) << 2);
@@ -187,14 +204,15 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
       const int b = b0 + lrow + r;
       const bool inb = b < B;
       const bool valid = inb && (t < mylen[r]);
-      const long gbase = inb ? ((long)b * L + t) * 4 * h : 0;
+      const long gbase = inb ? ((dbase + (long)b * L + t) * 4 * h) : 0;
+      const long xbase = inb ? (((long)b * L + t) * gxs + gx_off) : 0;
 #pragma unroll
       for (int jj = 0; jj < 2; ++jj) {
         const int j = wid * 32 + jj * 16 + (lane & 15);
         if (j >= h) continue;
         float dgi = 0.f, dgf = 0.f, dgg = 0.f, dgo = 0.f;
         if (valid) {
-          const long obase = ((long)b * L + t) * h + j;
+          const long obase = (dbase + (long)b * L + t) * h + j;
           const float gi = gates[gbase + 0 * h + j];
           const float gf = gates[gbase + 1 * h + j];
           const float gg = gates[gbase + 2 * h + j];
@@ -202,10 +220,11 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
           const float c_t = cs[obase];
           const float c_prev =
               (t_prev >= 0 && t_prev < L && t_prev < mylen[r])
-                  ? cs[((long)b * L + t_prev) * h + j]
+                  ? cs[(dbase + (long)b * L + t_prev) * h + j]
                   : 0.f;
           const float ac = act_f(c_t, relu);
-          const float dh = dh_reg[r][jj] + to_f32(dhs[obase]);
+          const float dh =
+              dh_reg[r][jj] + to_f32(dhs[((long)b * L + t) * hss + hs_off + j]);
           float dc = dc_reg[r][jj] + dh * go * dact_from_out(ac, relu);
           dgo = dh * ac * go * (1.f - go);
           dgi = dc * gg * gi * (1.f - gi);
@@ -215,10 +234,10 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
         }
         // write pre-activation gate grads (global + LDS for the MFMA)
         if (inb) {
-          from_f32(dgi, &dgates_x[gbase + 0 * h + j]);
-          from_f32(dgf, &dgates_x[gbase + 1 * h + j]);
-          from_f32(dgg, &dgates_x[gbase + 2 * h + j]);
-          from_f32(dgo, &dgates_x[gbase + 3 * h + j]);
+          from_f32(dgi, &dgates_x[xbase + 0 * h + j]);
+          from_f32(dgf, &dgates_x[xbase + 1 * h + j]);
+          from_f32(dgg, &dgates_x[xbase + 2 * h + j]);
+          from_f32(dgo, &dgates_x[xbase + 3 * h + j]);
         }
         const int lr = lrow + r;
         dg_s[lr * 4 * h + 0 * h + j] = __float2bfloat16(dgi);
@@ -256,80 +275,139 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
 }
 
 // ===================================================================== host
-static void lstm_checks(const at::Tensor& gates_x, int& B, int& L, int& h4) {
+
+// Bidirectional fused path: gates_x [B,L,8h] (fw|bw), w_hh_t2 [2,4h,h].
+// Returns hs [B,L,2h] (concat free), cs [2,B,L,h], gates [2,B,L,4h].
+std::vector<at::Tensor> bilstm_fwd_l(const at::Tensor& gates_x,
+                                     const at::Tensor& w_hh_t2,
+                                     const at::Tensor& lens, bool relu) {
   CHECK_CUDA_CONTIG(gates_x);
-  B = gates_x.size(0);
-  L = gates_x.size(1);
-  h4 = gates_x.size(2);
-  const int h = h4 / 4;
+  const int B = gates_x.size(0), L = gates_x.size(1);
+  const int h = gates_x.size(2) / 8;
   TORCH_CHECK(h % 32 == 0 && h <= 128,
-              "lstm kernel: hidden must be a multiple of 32 and <= 128 "
-              "(wrapper falls back otherwise), got ", h);
+              "bilstm kernel: hidden must be a multiple of 32 and <= 128, got ",
+              h);
+  TORCH_CHECK(w_hh_t2.size(0) == 2 && w_hh_t2.scalar_type() == at::kBFloat16,
+              "w_hh_t2 must be bf16 [2,4h,h]");
+  auto hs = at::empty({B, L, 2 * h}, gates_x.options());
+  auto cs = at::empty({2, B, L, h}, gates_x.options().dtype(at::kFloat));
+  auto gates = at::empty({2, B, L, 4 * h},
+                         gates_x.options().dtype(at::kFloat));
+  const size_t smem = (size_t)(4 * h * h + 16 * h) * sizeof(bf16)
+                      + (size_t)16 * 4 * h * gates_x.element_size();
+  TORCH_CHECK(smem <= 160 * 1024, "lstm fwd LDS overflow");
+  const dim3 grid((B + 15) / 16, 2);
+  auto stream = cur_stream(gates_x);
+  if (gates_x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(lstm_fwd_kernel<bf16>, grid, dim3(256), smem, stream,
+                       (const bf16*)gates_x.data_ptr(),
+                       (const bf16*)w_hh_t2.data_ptr(), lens.data_ptr<int>(),
+                       (bf16*)hs.data_ptr(), cs.data_ptr<float>(),
+                       gates.data_ptr<float>(), B, L, h, 8 * h, 2 * h,
+                       /*rev0=*/false, relu);
+  } else {
+    hipLaunchKernelGGL(lstm_fwd_kernel<float>, grid, dim3(256), smem, stream,
+                       gates_x.data_ptr<float>(),
+                       (const bf16*)w_hh_t2.data_ptr(), lens.data_ptr<int>(),
+                       hs.data_ptr<float>(), cs.data_ptr<float>(),
+                       gates.data_ptr<float>(), B, L, h, 8 * h, 2 * h,
+                       /*rev0=*/false, relu);
+  }
+  HIP_CHECK_LAST();
+  return {hs, cs, gates};
 }
 
+at::Tensor bilstm_bwd_l(const at::Tensor& dhs, const at::Tensor& cs,
+                        const at::Tensor& gates, const at::Tensor& w_hh2,
+                        const at::Tensor& lens, bool relu) {
+  const int B = dhs.size(0), L = dhs.size(1);
+  const int h = dhs.size(2) / 2;
+  TORCH_CHECK(w_hh2.size(0) == 2 && w_hh2.scalar_type() == at::kBFloat16,
+              "w_hh2 must be bf16 [2,h,4h]");
+  auto dhs_c = dhs.contiguous();
+  auto dgates_x = at::empty({B, L, 8 * h}, dhs.options());
+  const size_t smem = (size_t)(4 * h * h + 16 * 4 * h) * sizeof(bf16);
+  TORCH_CHECK(smem <= 160 * 1024, "lstm bwd LDS overflow");
+  const dim3 grid((B + 15) / 16, 2);
+  auto stream = cur_stream(dhs);
+  if (dhs.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(lstm_bwd_kernel<bf16>, grid, dim3(256), smem, stream,
+                       (const bf16*)dhs_c.data_ptr(), cs.data_ptr<float>(),
+                       gates.data_ptr<float>(), (const bf16*)w_hh2.data_ptr(),
+                       lens.data_ptr<int>(), (bf16*)dgates_x.data_ptr(), B, L,
+                       h, 8 * h, 2 * h, /*rev0=*/false, relu);
+  } else {
+    hipLaunchKernelGGL(lstm_bwd_kernel<float>, grid, dim3(256), smem, stream,
+                       dhs_c.data_ptr<float>(), cs.data_ptr<float>(),
+                       gates.data_ptr<float>(), (const bf16*)w_hh2.data_ptr(),
+                       lens.data_ptr<int>(), dgates_x.data_ptr<float>(), B, L,
+                       h, 8 * h, 2 * h, /*rev0=*/false, relu);
+  }
+  HIP_CHECK_LAST();
+  return dgates_x;
+}
+
+// Single-direction path (kept for the kernel unit tests: one direction,
+// reverse selectable, gates_x [B,L,4h] contiguous).
 std::vector<at::Tensor> lstm_fwd(const at::Tensor& gates_x,
                                  const at::Tensor& w_hh,
                                  const at::Tensor& lens, bool reverse,
                                  bool relu) {
-  int B, L, h4;
-  lstm_checks(gates_x, B, L, h4);
-  const int h = h4 / 4;
+  CHECK_CUDA_CONTIG(gates_x);
+  const int B = gates_x.size(0), L = gates_x.size(1);
+  const int h4 = gates_x.size(2), h = h4 / 4;
+  TORCH_CHECK(h % 32 == 0 && h <= 128, "lstm kernel: bad hidden ", h);
   auto hs = at::empty({B, L, h}, gates_x.options());
-  auto cs = at::empty({B, L, h}, gates_x.options().dtype(at::kFloat));
-  auto gates = at::empty({B, L, h4}, gates_x.options().dtype(at::kFloat));
+  auto cs = at::empty({1, B, L, h}, gates_x.options().dtype(at::kFloat));
+  auto gates = at::empty({1, B, L, h4}, gates_x.options().dtype(at::kFloat));
   auto w_t = w_hh.t().contiguous().to(at::kBFloat16);  // [4h,h]
   const size_t smem = (size_t)(4 * h * h + 16 * h) * sizeof(bf16)
                       + (size_t)16 * 4 * h * gates_x.element_size();
-  TORCH_CHECK(smem <= 150 * 1024, "lstm fwd LDS overflow");
-  const int grid = (B + 15) / 16;
+  TORCH_CHECK(smem <= 160 * 1024, "lstm fwd LDS overflow");
+  const dim3 grid((B + 15) / 16, 1);
   auto stream = cur_stream(gates_x);
   if (gates_x.scalar_type() == at::kBFloat16) {
-    hipLaunchKernelGGL(lstm_fwd_kernel<bf16>, dim3(grid), dim3(256), smem,
-                       stream, (const bf16*)gates_x.data_ptr(),
+    hipLaunchKernelGGL(lstm_fwd_kernel<bf16>, grid, dim3(256), smem, stream,
+                       (const bf16*)gates_x.data_ptr(),
                        (const bf16*)w_t.data_ptr(), lens.data_ptr<int>(),
                        (bf16*)hs.data_ptr(), cs.data_ptr<float>(),
-                       gates.data_ptr<float>(), B, L, h, reverse, relu);
+                       gates.data_ptr<float>(), B, L, h, h4, h, reverse, relu);
   } else {
-    hipLaunchKernelGGL(lstm_fwd_kernel<float>, dim3(grid), dim3(256), smem,
-                       stream, gates_x.data_ptr<float>(),
-                       (const bf16*)w_t.data_ptr(), lens.data_ptr<int>(),
-                       hs.data_ptr<float>(), cs.data_ptr<float>(),
-                       gates.data_ptr<float>(), B, L, h, reverse, relu);
+    hipLaunchKernelGGL(lstm_fwd_kernel<float>, grid, dim3(256), smem, stream,
+                       gates_x.data_ptr<float>(), (const bf16*)w_t.data_ptr(),
+                       lens.data_ptr<int>(), hs.data_ptr<float>(),
+                       cs.data_ptr<float>(), gates.data_ptr<float>(), B, L, h,
+                       h4, h, reverse, relu);
   }
   HIP_CHECK_LAST();
-  return {hs, cs, gates};
+  return {hs, cs.squeeze(0), gates.squeeze(0)};
 }
 
 std::vector<at::Tensor> lstm_bwd(const at::Tensor& dhs, const at::Tensor& hs,
                                  const at::Tensor& cs, const at::Tensor& gates,
                                  const at::Tensor& w_hh, const at::Tensor& lens,
                                  bool reverse, bool relu) {
-  int B, L, h4;
-  B = dhs.size(0);
-  L = dhs.size(1);
-  const int h = dhs.size(2);
-  h4 = 4 * h;
-  auto dgates_x = at::empty({B, L, h4}, dhs.options());
+  const int B = dhs.size(0), L = dhs.size(1), h = dhs.size(2);
+  auto dgates_x = at::empty({B, L, 4 * h}, dhs.options());
   auto w_b = w_hh.contiguous().to(at::kBFloat16);  // [h,4h]
   const size_t smem = (size_t)(4 * h * h + 16 * 4 * h) * sizeof(bf16);
-  TORCH_CHECK(smem <= 150 * 1024, "lstm bwd LDS overflow");
-  const int grid = (B + 15) / 16;
+  TORCH_CHECK(smem <= 160 * 1024, "lstm bwd LDS overflow");
+  const dim3 grid((B + 15) / 16, 1);
   auto stream = cur_stream(dhs);
   auto dhs_c = dhs.contiguous();
   if (dhs.scalar_type() == at::kBFloat16) {
-    hipLaunchKernelGGL(lstm_bwd_kernel<bf16>, dim3(grid), dim3(256), smem,
-                       stream, (const bf16*)dhs_c.data_ptr(),
-                       cs.data_ptr<float>(), gates.data_ptr<float>(),
-                       (const bf16*)w_b.data_ptr(), lens.data_ptr<int>(),
-                       (bf16*)dgates_x.data_ptr(), B, L, h, reverse, relu);
+    hipLaunchKernelGGL(lstm_bwd_kernel<bf16>, grid, dim3(256), smem, stream,
+                       (const bf16*)dhs_c.data_ptr(), cs.data_ptr<float>(),
+                       gates.data_ptr<float>(), (const bf16*)w_b.data_ptr(),
+                       lens.data_ptr<int>(), (bf16*)dgates_x.data_ptr(), B, L,
+                       h, 4 * h, h, reverse, relu);
   } else {
-    hipLaunchKernelGGL(lstm_bwd_kernel<float>, dim3(grid), dim3(256), smem,
-                       stream, dhs_c.data_ptr<float>(),
-                       cs.data_ptr<float>(), gates.data_ptr<float>(),
-                       (const bf16*)w_b.data_ptr(), lens.data_ptr<int>(),
-                       dgates_x.data_ptr<float>(), B, L, h, reverse, relu);
+    hipLaunchKernelGGL(lstm_bwd_kernel<float>, grid, dim3(256), smem, stream,
+                       dhs_c.data_ptr<float>(), cs.data_ptr<float>(),
+                       gates.data_ptr<float>(), (const bf16*)w_b.data_ptr(),
+                       lens.data_ptr<int>(), dgates_x.data_ptr<float>(), B, L,
+                       h, 4 * h, h, reverse, relu);
   }
   HIP_CHECK_LAST();
-  // dW_hh etc. are library GEMMs in the python wrapper (needs h_{t-1})
   return {dgates_x};
 }

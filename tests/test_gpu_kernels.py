@@ -286,7 +286,7 @@ def test_multi_tensor_adamw_matches_cpu_math():
     ref_m = [m.clone() for m in ms]
     ref_v = [v.clone() for v in vs]
     lr, b1, b2, eps, wd = 1e-3, 0.9, 0.999, 1e-6, 0.01
-    ops.get_ext().multi_tensor_adamw(params, grads, ms, vs,
+    ops.get_ext().multi_tensor_adamw(params, grads, [], ms, vs,
                                      [lr] * len(params), [wd] * len(params),
                                      b1, b2, eps)
     for p, g, m, v in zip(ref_p, grads, ref_m, ref_v):
@@ -295,6 +295,52 @@ def test_multi_tensor_adamw_matches_cpu_math():
         p.add_(-lr * (m / (v.sqrt() + eps) + wd * p))
     for a, b in zip(params, ref_p):
         torch.testing.assert_close(a, b, atol=1e-5, rtol=1e-5)
+
+
+def test_multi_tensor_adamw_bf16_master():
+    """bf16 params + fp32 masters: master carries precision, bf16
+    weight tracks round(master)."""
+    _cuda()
+    torch.manual_seed(12)
+    shapes = [(64,), (16, 32)]
+    masters = [torch.randn(s, device="cuda") for s in shapes]
+    params = [m.to(torch.bfloat16) for m in masters]
+    grads = [torch.randn(s, device="cuda").to(torch.bfloat16) for s in shapes]
+    ms = [torch.zeros(s, device="cuda") for s in shapes]
+    vs = [torch.zeros(s, device="cuda") for s in shapes]
+    ref_master = [m.clone() for m in masters]
+    lr, b1, b2, eps, wd = 1e-3, 0.9, 0.999, 1e-6, 0.01
+    for _ in range(3):
+        ops.get_ext().multi_tensor_adamw(params, grads, masters, ms, vs,
+                                         [lr] * 2, [wd] * 2, b1, b2, eps)
+    rm = [torch.zeros_like(m) for m in ref_master]
+    rv = [torch.zeros_like(m) for m in ref_master]
+    for _ in range(3):
+        for p, g, m, v in zip(ref_master, grads, rm, rv):
+            gf = g.float()
+            m.mul_(b1).add_(gf, alpha=1 - b1)
+            v.mul_(b2).addcmul_(gf, gf, value=1 - b2)
+            p.add_(-lr * (m / (v.sqrt() + eps) + wd * p))
+    for a, b in zip(masters, ref_master):
+        torch.testing.assert_close(a, b, atol=1e-5, rtol=1e-5)
+    for p, mstr in zip(params, masters):
+        torch.testing.assert_close(p, mstr.to(torch.bfloat16))
+
+
+def test_multi_tensor_sumsq_and_scale():
+    _cuda()
+    torch.manual_seed(13)
+    grads = [torch.randn(33, device="cuda"),
+             torch.randn(8, 9, device="cuda").to(torch.bfloat16)]
+    expect = sum(float(g.float().pow(2).sum()) for g in grads)
+    ss = ops.get_ext().multi_tensor_sumsq(grads)
+    assert abs(float(ss) - expect) < 1e-2 * max(1.0, expect)
+    coef = torch.tensor([0.5], device="cuda")
+    before = [g.clone() for g in grads]
+    ops.get_ext().multi_tensor_scale(grads, coef)
+    for g, b in zip(grads, before):
+        torch.testing.assert_close(g.float(), b.float() * 0.5,
+                                   atol=1e-2, rtol=1e-2)
 
 
 # ----------------------------------------------------- dispatch is native
