@@ -116,6 +116,27 @@ class BertModel(nn.Module):
         return x
 
 
+class BertMlmHead(nn.Module):
+    """MLM head: transform dense + layernorm + tied-embedding logits
+    (reference get_masked_lm_output, augment_mlm.py:49-72)."""
+
+    def __init__(self, bert: BertModel):
+        super().__init__()
+        cfg = bert.cfg
+        self.transform = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        self.ln_w = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.ln_b = nn.Parameter(torch.zeros(cfg.hidden_size))
+        self.word_embedding = bert.embeddings.word.weight  # tied
+        self.output_bias = nn.Parameter(torch.zeros(cfg.vocab_size))
+        self.eps = cfg.layer_norm_eps
+
+    def forward(self, seq_out: torch.Tensor) -> torch.Tensor:
+        x = torch.nn.functional.gelu(self.transform(seq_out))
+        x = ops.layernorm(x, self.ln_w, self.ln_b, self.eps)
+        return x @ self.word_embedding.to(x.dtype).T + \
+            self.output_bias.to(x.dtype)
+
+
 class BertEmbeddingOnly(nn.Module):
     """Wordpiece embedding + position postprocessor only, no encoder —
     the reference's bert_token_embedding (tools/layer.py:84-109), used by

@@ -25,14 +25,24 @@ class BiLSTM(nn.Module):
         self.hidden_size = hidden_size
         self.activation = activation
         self.dropout = nn.Dropout(1.0 - keep_prob)
-        k = 1.0 / (hidden_size ** 0.5)
 
-        def p(*shape):
-            return nn.Parameter(torch.empty(*shape).uniform_(-k, k))
-        self.w_ih_f, self.w_hh_f, self.b_f = p(input_size, 4 * hidden_size), \
-            p(hidden_size, 4 * hidden_size), p(4 * hidden_size)
-        self.w_ih_b, self.w_hh_b, self.b_b = p(input_size, 4 * hidden_size), \
-            p(hidden_size, 4 * hidden_size), p(4 * hidden_size)
+        # TF LSTMCell defaults the reference builds on: glorot-uniform
+        # kernels, zero bias with forget-gate bias 1.0 — keeps the (relu)
+        # recurrence away from the exponential-growth regime over L=150
+        # steps (a uniform 1/sqrt(h) init sits near the stability cliff).
+        def w(*shape):
+            t = torch.empty(*shape)
+            nn.init.xavier_uniform_(t)
+            return nn.Parameter(t)
+
+        def b():
+            t = torch.zeros(4 * hidden_size)
+            t[hidden_size:2 * hidden_size] = 1.0  # forget gate (i,f,g,o)
+            return nn.Parameter(t)
+        self.w_ih_f, self.w_hh_f, self.b_f = w(input_size, 4 * hidden_size), \
+            w(hidden_size, 4 * hidden_size), b()
+        self.w_ih_b, self.w_hh_b, self.b_b = w(input_size, 4 * hidden_size), \
+            w(hidden_size, 4 * hidden_size), b()
 
     def forward(self, x: torch.Tensor, lens: torch.Tensor) -> torch.Tensor:
         out = ops.bilstm(x, self.w_ih_f, self.w_hh_f, self.b_f,

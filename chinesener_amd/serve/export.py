@@ -26,9 +26,17 @@ def export_model(model: torch.nn.Module, name: str, params: Dict,
     out_dir = os.path.join(export_root, name, str(version))
     os.makedirs(out_dir, exist_ok=True)
     torch.save(model.state_dict(), os.path.join(out_dir, "model.pt"))
-    export_params = {k: v for k, v in params.items()
-                     if isinstance(v, (int, float, str, bool, list, tuple, dict,
-                                       type(None)))}
+    # keep everything picklable (incl. BertConfig and embedding matrices —
+    # the server must rebuild the exact graph, serve/engine.py)
+    export_params = {}
+    for k, v in params.items():
+        if isinstance(v, torch.Tensor):
+            v = v.detach().cpu()
+        try:
+            pickle.dumps(v)
+        except Exception:
+            continue
+        export_params[k] = v
     with open(os.path.join(out_dir, "params.pkl"), "wb") as f:
         pickle.dump({"model_name": params.get("model_name", name),
                      "params": export_params}, f)

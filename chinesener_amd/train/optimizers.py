@@ -77,10 +77,17 @@ class AdamWeightDecay(torch.optim.Optimizer):
         # fused bins by param dtype: bf16 params carry an fp32 master copy
         # (pure-bf16 training mode — no autocast weight casts per step)
         fused = {torch.bfloat16: [], torch.float32: []}
+        # a group may mix dtypes (e.g. fp32 LN affines + bf16 biases in the
+        # no-decay group under pure-bf16 mode): process each dtype slice
+        # separately so masters exist exactly for the bf16 params
+        split_groups = []
         for group in self.param_groups:
-            params = [p for p in group["params"] if p.grad is not None]
-            if not params:
-                continue
+            for dt in (torch.float32, torch.bfloat16):
+                params = [p for p in group["params"]
+                          if p.grad is not None and p.dtype == dt]
+                if params:
+                    split_groups.append((group, params))
+        for group, params in split_groups:
             grads = [p.grad for p in params]
             states = [self.state[p] for p in params]
             for p, s in zip(params, states):
