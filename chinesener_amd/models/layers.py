@@ -20,10 +20,14 @@ class BiLSTM(nn.Module):
     tools/layer.py:10-41, model/bilstm_crf.py:48-52)."""
 
     def __init__(self, input_size: int, hidden_size: int, activation: str = "tanh",
-                 keep_prob: float = 0.8):
+                 keep_prob: float = 0.8, cell_clip: float = None):
         super().__init__()
         self.hidden_size = hidden_size
         self.activation = activation
+        # relu cells are unbounded; clip the cell state like TF's
+        # LSTMCell(cell_clip=...) to keep the recurrence stable
+        self.cell_clip = (10.0 if activation == "relu" else 0.0) \
+            if cell_clip is None else cell_clip
         self.dropout = nn.Dropout(1.0 - keep_prob)
 
         # TF LSTMCell defaults the reference builds on: glorot-uniform
@@ -47,7 +51,7 @@ class BiLSTM(nn.Module):
     def forward(self, x: torch.Tensor, lens: torch.Tensor) -> torch.Tensor:
         out = ops.bilstm(x, self.w_ih_f, self.w_hh_f, self.b_f,
                          self.w_ih_b, self.w_hh_b, self.b_b, lens,
-                         self.activation)
+                         self.activation, cell_clip=self.cell_clip)
         return self.dropout(out)
 
 

@@ -297,12 +297,14 @@ class _BiLstmFn(torch.autograd.Function):
     concat is produced directly by the strided kernel."""
 
     @staticmethod
-    def forward(ctx, gates_x, w_hh_f, w_hh_b, lens, act_relu):
+    def forward(ctx, gates_x, w_hh_f, w_hh_b, lens, act_relu, cell_clip):
         w_hh_t2 = torch.stack([w_hh_f.t(), w_hh_b.t()]) \
             .to(torch.bfloat16).contiguous()
-        hs, cs, gates = get_ext().bilstm_fwd(gates_x, w_hh_t2, lens, act_relu)
+        hs, cs, gates = get_ext().bilstm_fwd(gates_x, w_hh_t2, lens, act_relu,
+                                             cell_clip)
         ctx.save_for_backward(hs, cs, gates, w_hh_f, w_hh_b, lens)
         ctx.relu = act_relu
+        ctx.cell_clip = cell_clip
         return hs
 
     @staticmethod
@@ -310,7 +312,7 @@ class _BiLstmFn(torch.autograd.Function):
         hs, cs, gates, w_hh_f, w_hh_b, lens = ctx.saved_tensors
         w_hh2 = torch.stack([w_hh_f, w_hh_b]).to(torch.bfloat16).contiguous()
         dgates_x = get_ext().bilstm_bwd(dhs.contiguous(), cs, gates, w_hh2,
-                                        lens, ctx.relu)
+                                        lens, ctx.relu, ctx.cell_clip)
         h = hs.shape[-1] // 2
         # dW_hh = sum_t h_{t-1}^T dgates_t per direction — library GEMMs.
         # hs holds the carried state at valid steps and 0 elsewhere; dgates
@@ -323,15 +325,18 @@ class _BiLstmFn(torch.autograd.Function):
         dw_f = h_prev_f.reshape(-1, h).T @ dg_f.reshape(-1, 4 * h)
         dw_b = h_prev_b.reshape(-1, h).T @ dg_b.reshape(-1, 4 * h)
         return (dgates_x, dw_f.to(w_hh_f.dtype), dw_b.to(w_hh_b.dtype),
-                None, None)
+                None, None, None)
 
 
 def bilstm(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b, lens,
-           activation: str = "tanh", state_dropout=None):
+           activation: str = "tanh", state_dropout=None,
+           cell_clip: float = 0.0):
     """BiLSTM over padded [B,L,E] -> [B,L,2h].
 
     HIP path: hidden % 32 == 0 and hidden <= 128 (LDS-resident W_hh);
-    other sizes run the torch recurrence (slow path, logged once)."""
+    other sizes run the torch recurrence (slow path, logged once).
+    cell_clip > 0 bounds the cell state (TF LSTMCell cell_clip) — the
+    relu recurrence needs it to stay off the exponential-growth regime."""
     h = w_hh_f.shape[0]
     if hip_enabled(x) and h % 32 == 0 and h <= 128:
         # single fused x-projection GEMM for both directions
@@ -339,7 +344,8 @@ def bilstm(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b, lens,
         b2 = torch.cat([b_f, b_b]).to(x.dtype)
         gates_x = (x @ w_ih2 + b2).contiguous()
         return _BiLstmFn.apply(gates_x, w_hh_f, w_hh_b,
-                               lens.to(torch.int32), activation == "relu")
+                               lens.to(torch.int32), activation == "relu",
+                               float(cell_clip))
     if x.is_cuda:
         global _LSTM_FALLBACK_WARNED
         if not _LSTM_FALLBACK_WARNED:
@@ -349,7 +355,7 @@ def bilstm(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b, lens,
                 " using torch recurrence", h)
             _LSTM_FALLBACK_WARNED = True
     return ref.bilstm_forward(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b,
-                              lens, activation, state_dropout)
+                              lens, activation, state_dropout, cell_clip)
 
 
 _LSTM_FALLBACK_WARNED = False

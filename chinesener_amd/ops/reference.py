@@ -169,7 +169,8 @@ def crf_decode(emissions: torch.Tensor, mask: torch.Tensor,
 def lstm_forward(x: torch.Tensor, w_ih: torch.Tensor, w_hh: torch.Tensor,
                  b: torch.Tensor, lens: torch.Tensor, reverse: bool = False,
                  activation: str = "tanh",
-                 state_dropout: Optional[torch.Tensor] = None) -> torch.Tensor:
+                 state_dropout: Optional[torch.Tensor] = None,
+                 cell_clip: float = 0.0) -> torch.Tensor:
     """Single-direction LSTM over padded [B,L,E] -> [B,L,h].
 
     Gate order i,f,g,o (torch convention). `activation` applies to the
@@ -193,6 +194,9 @@ def lstm_forward(x: torch.Tensor, w_ih: torch.Tensor, w_hh: torch.Tensor,
         i, f, gc, o = g.split(h, dim=-1)
         i, f, o = torch.sigmoid(i), torch.sigmoid(f), torch.sigmoid(o)
         c_new = f * ct + i * act(gc)
+        if cell_clip > 0:
+            # TF LSTMCell cell_clip: bounds the relu recurrence
+            c_new = c_new.clamp(-cell_clip, cell_clip)
         h_new = o * act(c_new)
         valid = (t < lens).to(x.dtype)[:, None]                 # [B,1]
         ct = valid * c_new + (1 - valid) * ct
@@ -205,9 +209,12 @@ def lstm_forward(x: torch.Tensor, w_ih: torch.Tensor, w_hh: torch.Tensor,
 
 
 def bilstm_forward(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b, lens,
-                   activation="tanh", state_dropout=None) -> torch.Tensor:
-    fw = lstm_forward(x, w_ih_f, w_hh_f, b_f, lens, False, activation, state_dropout)
-    bw = lstm_forward(x, w_ih_b, w_hh_b, b_b, lens, True, activation, state_dropout)
+                   activation="tanh", state_dropout=None,
+                   cell_clip: float = 0.0) -> torch.Tensor:
+    fw = lstm_forward(x, w_ih_f, w_hh_f, b_f, lens, False, activation,
+                      state_dropout, cell_clip)
+    bw = lstm_forward(x, w_ih_b, w_hh_b, b_b, lens, True, activation,
+                      state_dropout, cell_clip)
     return torch.cat([fw, bw], dim=-1)
 
 

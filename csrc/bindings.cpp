@@ -68,10 +68,10 @@ std::vector<at::Tensor> lstm_bwd(const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&, bool,
                                  bool);
 std::vector<at::Tensor> bilstm_fwd_l(const at::Tensor&, const at::Tensor&,
-                                     const at::Tensor&, bool);
+                                     const at::Tensor&, bool, double);
 at::Tensor bilstm_bwd_l(const at::Tensor&, const at::Tensor&,
                         const at::Tensor&, const at::Tensor&,
-                        const at::Tensor&, bool);
+                        const at::Tensor&, bool, double);
 
 PYBIND11_MODULE(_hip_ops, m) {
   m.doc() = "chinesener_amd gfx950 HIP kernels";

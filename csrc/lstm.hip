@@ -50,11 +50,13 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     const int* __restrict__ lens, T* __restrict__ hs,  // [B,L,hss]
     float* __restrict__ cs,                            // [D,B,L,h]
     float* __restrict__ gates_out,                     // [D,B,L,4h] activated
-    int B, int L, int h, int gxs, int hss, bool rev0, bool relu) {
+    int B, int L, int h, int gxs, int hss, bool rev0, bool relu,
+    float cell_clip) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   bf16* wT = reinterpret_cast<bf16*>(smem_raw);  // [4h][h]
   bf16* hb = wT + 4 * h * h;                     // [16][h] current h (bf16)
-  T* gx_s = reinterpret_cast<T*>(hb + 16 * h);   // [16][4h] staged gates_x
+  bf16* gx_s = hb + 16 * h;  // [16][4h] staged gates_x (bf16 also for fp32
+                             // inputs: halves the LDS footprint)
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int b0 = blockIdx.x * 16;
@@ -98,10 +100,16 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
             reinterpret_cast<s16x8*>(gx_s)[i] =
                 *reinterpret_cast<const s16x8*>(gates_x + g);
           } else {
-            reinterpret_cast<f32x4*>(gx_s)[i * 2] =
-                *reinterpret_cast<const f32x4*>(gates_x + g);
-            reinterpret_cast<f32x4*>(gx_s)[i * 2 + 1] =
-                *reinterpret_cast<const f32x4*>(gates_x + g + 4);
+            const f32x4 lo = *reinterpret_cast<const f32x4*>(gates_x + g);
+            const f32x4 hi = *reinterpret_cast<const f32x4*>(gates_x + g + 4);
+            bf16 packed[8];
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+              packed[e] = __float2bfloat16(lo[e]);
+              packed[e + 4] = __float2bfloat16(hi[e]);
+            }
+            reinterpret_cast<s16x8*>(gx_s)[i] =
+                *reinterpret_cast<const s16x8*>(packed);
           }
         }
       }
@@ -144,7 +152,10 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
         gf = 1.f / (1.f + __expf(-gf));
         go = 1.f / (1.f + __expf(-go));
         gg = act_f(gg, relu);
-        const float c_new = gf * c_reg[r][jj] + gi * gg;
+        float c_new = gf * c_reg[r][jj] + gi * gg;
+        // TF LSTMCell cell_clip: bounds the (relu) recurrence
+        if (cell_clip > 0.f)
+          c_new = fminf(fmaxf(c_new, -cell_clip), cell_clip);
         const float h_new = go * act_f(c_new, relu);
         if (valid) c_reg[r][jj] = c_new;
         const long obase = ((long)b * L + t) * hss + hs_off + j;
@@ -169,7 +180,8 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
     const float* __restrict__ gates,  // [D,B,L,4h] activated
     const bf16* __restrict__ w_hh,    // [D,h,4h] (original layout)
     const int* __restrict__ lens, T* __restrict__ dgates_x,  // [B,L,gxs]
-    int B, int L, int h, int gxs, int hss, bool rev0, bool relu) {
+    int B, int L, int h, int gxs, int hss, bool rev0, bool relu,
+    float cell_clip) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   bf16* w_s = reinterpret_cast<bf16*>(smem_raw);  // [h][4h]
   bf16* dg_s = w_s + 4 * h * h;                   // [16][4h]
@@ -227,6 +239,8 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
               dh_reg[r][jj] + to_f32(dhs[((long)b * L + t) * hss + hs_off + j]);
           float dc = dc_reg[r][jj] + dh * go * dact_from_out(ac, relu);
           dgo = dh * ac * go * (1.f - go);
+          // clamp boundary: no grad through a clipped cell state
+          if (cell_clip > 0.f && fabsf(c_t) >= cell_clip) dc = 0.f;
           dgi = dc * gg * gi * (1.f - gi);
           dgf = dc * c_prev * gf * (1.f - gf);
           dgg = dc * gi * dact_from_out(gg, relu);
@@ -280,7 +294,8 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
 // Returns hs [B,L,2h] (concat free), cs [2,B,L,h], gates [2,B,L,4h].
 std::vector<at::Tensor> bilstm_fwd_l(const at::Tensor& gates_x,
                                      const at::Tensor& w_hh_t2,
-                                     const at::Tensor& lens, bool relu) {
+                                     const at::Tensor& lens, bool relu,
+                                     double cell_clip) {
   CHECK_CUDA_CONTIG(gates_x);
   const int B = gates_x.size(0), L = gates_x.size(1);
   const int h = gates_x.size(2) / 8;
@@ -293,8 +308,7 @@ std::vector<at::Tensor> bilstm_fwd_l(const at::Tensor& gates_x,
   auto cs = at::empty({2, B, L, h}, gates_x.options().dtype(at::kFloat));
   auto gates = at::empty({2, B, L, 4 * h},
                          gates_x.options().dtype(at::kFloat));
-  const size_t smem = (size_t)(4 * h * h + 16 * h) * sizeof(bf16)
-                      + (size_t)16 * 4 * h * gates_x.element_size();
+  const size_t smem = (size_t)(4 * h * h + 16 * h + 16 * 4 * h) * sizeof(bf16);
   TORCH_CHECK(smem <= 160 * 1024, "lstm fwd LDS overflow");
   const dim3 grid((B + 15) / 16, 2);
   auto stream = cur_stream(gates_x);
@@ -304,14 +318,14 @@ std::vector<at::Tensor> bilstm_fwd_l(const at::Tensor& gates_x,
                        (const bf16*)w_hh_t2.data_ptr(), lens.data_ptr<int>(),
                        (bf16*)hs.data_ptr(), cs.data_ptr<float>(),
                        gates.data_ptr<float>(), B, L, h, 8 * h, 2 * h,
-                       /*rev0=*/false, relu);
+                       /*rev0=*/false, relu, (float)cell_clip);
   } else {
     hipLaunchKernelGGL(lstm_fwd_kernel<float>, grid, dim3(256), smem, stream,
                        gates_x.data_ptr<float>(),
                        (const bf16*)w_hh_t2.data_ptr(), lens.data_ptr<int>(),
                        hs.data_ptr<float>(), cs.data_ptr<float>(),
                        gates.data_ptr<float>(), B, L, h, 8 * h, 2 * h,
-                       /*rev0=*/false, relu);
+                       /*rev0=*/false, relu, (float)cell_clip);
   }
   HIP_CHECK_LAST();
   return {hs, cs, gates};
@@ -319,7 +333,7 @@ std::vector<at::Tensor> bilstm_fwd_l(const at::Tensor& gates_x,
 
 at::Tensor bilstm_bwd_l(const at::Tensor& dhs, const at::Tensor& cs,
                         const at::Tensor& gates, const at::Tensor& w_hh2,
-                        const at::Tensor& lens, bool relu) {
+                        const at::Tensor& lens, bool relu, double cell_clip) {
   const int B = dhs.size(0), L = dhs.size(1);
   const int h = dhs.size(2) / 2;
   TORCH_CHECK(w_hh2.size(0) == 2 && w_hh2.scalar_type() == at::kBFloat16,
@@ -335,13 +349,13 @@ at::Tensor bilstm_bwd_l(const at::Tensor& dhs, const at::Tensor& cs,
                        (const bf16*)dhs_c.data_ptr(), cs.data_ptr<float>(),
                        gates.data_ptr<float>(), (const bf16*)w_hh2.data_ptr(),
                        lens.data_ptr<int>(), (bf16*)dgates_x.data_ptr(), B, L,
-                       h, 8 * h, 2 * h, /*rev0=*/false, relu);
+                       h, 8 * h, 2 * h, /*rev0=*/false, relu, (float)cell_clip);
   } else {
     hipLaunchKernelGGL(lstm_bwd_kernel<float>, grid, dim3(256), smem, stream,
                        dhs_c.data_ptr<float>(), cs.data_ptr<float>(),
                        gates.data_ptr<float>(), (const bf16*)w_hh2.data_ptr(),
                        lens.data_ptr<int>(), dgates_x.data_ptr<float>(), B, L,
-                       h, 8 * h, 2 * h, /*rev0=*/false, relu);
+                       h, 8 * h, 2 * h, /*rev0=*/false, relu, (float)cell_clip);
   }
   HIP_CHECK_LAST();
   return dgates_x;
@@ -361,8 +375,7 @@ std::vector<at::Tensor> lstm_fwd(const at::Tensor& gates_x,
   auto cs = at::empty({1, B, L, h}, gates_x.options().dtype(at::kFloat));
   auto gates = at::empty({1, B, L, h4}, gates_x.options().dtype(at::kFloat));
   auto w_t = w_hh.t().contiguous().to(at::kBFloat16);  // [4h,h]
-  const size_t smem = (size_t)(4 * h * h + 16 * h) * sizeof(bf16)
-                      + (size_t)16 * 4 * h * gates_x.element_size();
+  const size_t smem = (size_t)(4 * h * h + 16 * h + 16 * 4 * h) * sizeof(bf16);
   TORCH_CHECK(smem <= 160 * 1024, "lstm fwd LDS overflow");
   const dim3 grid((B + 15) / 16, 1);
   auto stream = cur_stream(gates_x);
@@ -371,13 +384,14 @@ std::vector<at::Tensor> lstm_fwd(const at::Tensor& gates_x,
                        (const bf16*)gates_x.data_ptr(),
                        (const bf16*)w_t.data_ptr(), lens.data_ptr<int>(),
                        (bf16*)hs.data_ptr(), cs.data_ptr<float>(),
-                       gates.data_ptr<float>(), B, L, h, h4, h, reverse, relu);
+                       gates.data_ptr<float>(), B, L, h, h4, h, reverse, relu,
+                       0.f);
   } else {
     hipLaunchKernelGGL(lstm_fwd_kernel<float>, grid, dim3(256), smem, stream,
                        gates_x.data_ptr<float>(), (const bf16*)w_t.data_ptr(),
                        lens.data_ptr<int>(), hs.data_ptr<float>(),
                        cs.data_ptr<float>(), gates.data_ptr<float>(), B, L, h,
-                       h4, h, reverse, relu);
+                       h4, h, reverse, relu, 0.f);
   }
   HIP_CHECK_LAST();
   return {hs, cs.squeeze(0), gates.squeeze(0)};
@@ -400,13 +414,13 @@ std::vector<at::Tensor> lstm_bwd(const at::Tensor& dhs, const at::Tensor& hs,
                        (const bf16*)dhs_c.data_ptr(), cs.data_ptr<float>(),
                        gates.data_ptr<float>(), (const bf16*)w_b.data_ptr(),
                        lens.data_ptr<int>(), (bf16*)dgates_x.data_ptr(), B, L,
-                       h, 4 * h, h, reverse, relu);
+                       h, 4 * h, h, reverse, relu, 0.f);
   } else {
     hipLaunchKernelGGL(lstm_bwd_kernel<float>, grid, dim3(256), smem, stream,
                        dhs_c.data_ptr<float>(), cs.data_ptr<float>(),
                        gates.data_ptr<float>(), (const bf16*)w_b.data_ptr(),
                        lens.data_ptr<int>(), dgates_x.data_ptr<float>(), B, L,
-                       h, 4 * h, h, reverse, relu);
+                       h, 4 * h, h, reverse, relu, 0.f);
   }
   HIP_CHECK_LAST();
   return {dgates_x};

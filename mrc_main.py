@@ -114,6 +114,21 @@ def main(argv=None):
 
     if args.do_eval:
         rows = mrc_predict(trainer, pipe.iter_batches("test", shuffle=False))
+        # token-level ROC/PR AUC over the B/I classes (reference
+        # mrc/model.py:57-83 adds ROC-AUC and PR-AUC eval metrics)
+        try:
+            from sklearn.metrics import average_precision_score, roc_auc_score
+            import numpy as np
+            y, s = [], []
+            for row in rows:
+                m = row["text_mask"].astype(bool)
+                y.extend((row["label_ids"][m] > 0).astype(int).tolist())
+                s.extend((row["pred_ids"][m] > 0).astype(int).tolist())
+            if len(set(y)) == 2:
+                log.info("MRC token ROC-AUC %.4f PR-AUC %.4f",
+                         roc_auc_score(y, s), average_precision_score(y, s))
+        except Exception as e:  # pragma: no cover
+            log.warning("AUC metrics skipped: %s", e)
         # tag types cycle per tag query in build order
         tag_types = []
         for _ in range(len(rows) // max(1, len(pipe.tag2query)) + 1):

@@ -196,6 +196,73 @@ def run_repeat_lstm(n=50):
     print("distinct lstm dx sums:", len(grads), sorted(grads)[:4])
 
 
+def test_adam_fused_vs_foreach():
+    """One fused multi-tensor step vs the eager foreach math on identical
+    state: any element diff beyond fp32 noise = kernel bug."""
+    torch.manual_seed(5)
+    from chinesener_amd.train import optimizers as O
+    model = build_model("bert_bilstm_crf", params_small()).to("cuda")
+    batch = make_synthetic_batch(8, 128, 10, vocab_size=2000, device="cuda")
+    one_backward(model, batch)
+    import copy
+    groups = O.build_param_groups(model, 1e-3, 0.01)
+    opt = O.AdamWeightDecay(groups, lr=1e-3)
+    snap = {n: p.detach().clone() for n, p in model.named_parameters()}
+    gsnap = {n: p.grad.detach().clone() for n, p in model.named_parameters()}
+    opt.step()
+    fused = {n: p.detach().clone() for n, p in model.named_parameters()}
+    # restore and run eager
+    with torch.no_grad():
+        for n, p in model.named_parameters():
+            p.copy_(snap[n])
+            p.grad.copy_(gsnap[n])
+    opt2 = O.AdamWeightDecay(O.build_param_groups(model, 1e-3, 0.01), lr=1e-3)
+    import chinesener_amd.ops as ops_mod
+    real = ops_mod.ext_available
+    ops_mod.ext_available = lambda: False
+    try:
+        opt2.step()
+    finally:
+        ops_mod.ext_available = real
+    worst, wn = 0.0, ""
+    for n, p in model.named_parameters():
+        d = (p.detach() - fused[n]).abs().max().item()
+        if d > worst:
+            worst, wn = d, n
+    print("adam fused-vs-foreach max abs diff:", worst, wn)
+
+
+def run_act_trace(n=10):
+    """Per-step max-activation trace: which module's forward explodes."""
+    torch.manual_seed(0)
+    model = build_model("bert_bilstm_crf", params_small()).to("cuda")
+    from chinesener_amd.train.optimizers import (AdamWeightDecay, LrSchedule,
+                                                 build_param_groups)
+    opt = AdamWeightDecay(build_param_groups(model, 1e-4, 0.01), lr=1e-4)
+    sched = LrSchedule("bert", 1e-4, num_train_steps=200, warmup_ratio=0.2)
+    batch = make_synthetic_batch(8, 128, 10, vocab_size=2000, device="cuda")
+    acts = {}
+    def mk(name):
+        def hook(mod, i, o):
+            if isinstance(o, torch.Tensor):
+                acts[name] = float(o.detach().float().abs().max())
+        return hook
+    model.bert.embeddings.register_forward_hook(mk("emb"))
+    model.bert.layers[0].register_forward_hook(mk("bert0"))
+    model.bert.layers[1].register_forward_hook(mk("bert1"))
+    model.bilstm.register_forward_hook(mk("bilstm"))
+    model.logits.register_forward_hook(mk("logits"))
+    for step in range(1, n + 1):
+        opt.zero_grad(set_to_none=True)
+        loss = one_backward(model, batch)
+        gn = torch.sqrt(sum(p.grad.float().pow(2).sum()
+                            for p in model.parameters() if p.grad is not None))
+        print(f"step {step} loss {loss:.1f} gn {float(gn):.2e} acts {acts}")
+        torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
+        sched.apply(opt, step)
+        opt.step()
+
+
 if __name__ == "__main__":
-    run_repeat()
-    run_repeat_lstm()
+    test_adam_fused_vs_foreach()
+    run_act_trace()

@@ -129,3 +129,48 @@ def test_main_cli_e2e(tmp_path):
     assert os.path.exists(tmp_path / "ckpt" / "ner_people_daily_bilstm_crf")
     assert os.path.exists(tmp_path / "data" / "people_daily" /
                           "bilstm_crf_predict.pkl")
+
+
+def test_addon_parser():
+    import argparse
+    from chinesener_amd.train.addon_parser import Addon, AddonParser
+    ap = argparse.ArgumentParser()
+    AddonParser([Addon("lambda_adv", 0.05), Addon("asymmetry", None,
+                                                  action="store_true")]).append(ap)
+    args = ap.parse_args(["--lambda_adv", "0.1", "--asymmetry"])
+    assert args.lambda_adv == 0.1 and args.asymmetry is True
+
+
+def test_convert_bf16_mixed_keeps_ln_fp32():
+    import torch
+    from chinesener_amd.models import build_model
+    from chinesener_amd.models.bert import BertConfig
+    from chinesener_amd.train.precision import convert_bf16_mixed
+    cfg = BertConfig(vocab_size=100, hidden_size=32, num_hidden_layers=1,
+                     num_attention_heads=2, intermediate_size=64)
+    m = build_model("bert_bilstm_crf", {
+        "vocab_size": 100, "label_size": 5, "bert_config": cfg,
+        "rnn_params": {"hidden_units_list": [32], "cell_activation": "relu",
+                       "keep_prob_list": [1.0]}, "tag2idx": {},
+        "dropout_rate": 0.0})
+    convert_bf16_mixed(m)
+    dt = {n: p.dtype for n, p in m.named_parameters()}
+    assert dt["bert.layers.0.ln1_w"] == torch.float32
+    assert dt["crf.transitions"] == torch.float32
+    assert dt["bert.layers.0.qkv.weight"] == torch.bfloat16
+    assert dt["bilstm.w_ih_f"] == torch.bfloat16
+
+
+def test_adamw_bf16_master_cpu_math():
+    """Eager master-weight path: bf16 params track the fp32 master."""
+    import torch
+    from chinesener_amd.train.optimizers import AdamWeightDecay
+    p = torch.nn.Parameter(torch.randn(32).to(torch.bfloat16))
+    opt = AdamWeightDecay([{"params": [p]}], lr=1e-2, weight_decay=0.0)
+    for _ in range(5):
+        p.grad = torch.randn(32).to(torch.bfloat16)
+        opt.step()
+    master = opt.state[p]["master"]
+    assert master.dtype == torch.float32
+    torch.testing.assert_close(p.detach().float(),
+                               master.to(torch.bfloat16).float())
