@@ -109,3 +109,28 @@ def test_mrc_driver_smoke(tmp_path, monkeypatch):
     logf = tmp_path / "ckpt" / "ner_msra_MRC" / "train.log"
     assert logf.exists()
     assert "span-level report" in logf.read_text()
+
+
+def test_mrc_span_model():
+    import torch
+    from chinesener_amd.models.bert import BertConfig
+    from chinesener_amd.mrc.span_model import (MrcSpan, make_span_labels)
+    cfg = BertConfig(vocab_size=300, hidden_size=32, num_hidden_layers=1,
+                     num_attention_heads=2, intermediate_size=64)
+    model = MrcSpan({"vocab_size": 300, "bert_config": cfg})
+    B, L = 2, 12
+    label_ids = torch.zeros(B, L, dtype=torch.long)
+    label_ids[0, 3] = 1
+    label_ids[0, 4] = 2
+    label_ids[1, 0] = 1
+    start, end, span = make_span_labels(label_ids)
+    assert start[0, 3] == 1 and end[0, 4] == 1 and span[0, 3, 4] == 1
+    assert start[1, 0] == 1 and end[1, 0] == 1 and span[1, 0, 0] == 1
+    batch = {"token_ids": torch.randint(1, 300, (B, L)),
+             "mask": torch.ones(B, L, dtype=torch.long),
+             "text_mask": torch.ones(B, L, dtype=torch.long),
+             "start_ids": start, "end_ids": end, "span_ids": span}
+    out = model(batch, compute_pred=True)
+    assert out.loss is not None and torch.isfinite(out.loss)
+    out.loss.backward()
+    assert out.pred_ids.shape == (B, L)
