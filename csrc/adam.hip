@@ -26,27 +26,35 @@ struct ChunkMeta {
 };
 
 template <typename T>
+__device__ __forceinline__ void adamw_one(const ChunkMeta& mt, long i,
+                                          float b1, float b2, float eps,
+                                          T* p, const T* g) {
+  const float gv = to_f32(g[i]);
+  const float m = b1 * mt.m[i] + (1.f - b1) * gv;
+  const float v = b2 * mt.v[i] + (1.f - b2) * gv * gv;
+  mt.m[i] = m;
+  mt.v[i] = v;
+  const float pv = mt.master ? mt.master[i] : to_f32(p[i]);
+  const float upd = pv - mt.lr * (m * __frcp_rn(sqrtf(v) + eps) + mt.wd * pv);
+  if (mt.master) mt.master[i] = upd;
+  from_f32(upd, &p[i]);
+}
+
+template <typename T>
 __global__ void multi_tensor_adamw_kernel(ChunkMeta* metas, int n_tensors,
                                           float b1, float b2, float eps) {
+  const long stride = (long)gridDim.x * blockDim.x * 4;
   for (int ti = blockIdx.y; ti < n_tensors; ti += gridDim.y) {
     const ChunkMeta mt = metas[ti];
-    const float lr = mt.lr;
-    const float wd = mt.wd;
     T* p = reinterpret_cast<T*>(mt.p);
     const T* g = reinterpret_cast<const T*>(mt.g);
-    const long stride = (long)gridDim.x * blockDim.x;
-    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < mt.n;
-         i += stride) {
-      const float gv = to_f32(g[i]);
-      const float m = b1 * mt.m[i] + (1.f - b1) * gv;
-      const float v = b2 * mt.v[i] + (1.f - b2) * gv * gv;
-      mt.m[i] = m;
-      mt.v[i] = v;
-      const float pv = mt.master ? mt.master[i] : to_f32(p[i]);
-      const float upd = pv - lr * (m * __frcp_rn(sqrtf(v) + eps) + wd * pv);
-      if (mt.master) mt.master[i] = upd;
-      from_f32(upd, &p[i]);
+    const long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+    long i = base;
+    for (; i + 3 < mt.n; i += stride) {
+#pragma unroll
+      for (int e = 0; e < 4; ++e) adamw_one<T>(mt, i + e, b1, b2, eps, p, g);
     }
+    for (; i < mt.n; ++i) adamw_one<T>(mt, i, b1, b2, eps, p, g);
   }
 }
 
@@ -113,22 +121,51 @@ struct NormMeta {
 
 __global__ void multi_tensor_l2_kernel(NormMeta* metas, int n_tensors,
                                        float* out) {
+  // every block walks every tensor (grid-stride over elements, 4-wide
+  // vector loads); block-level LDS reduce -> ONE atomic per block
+  __shared__ float warp_acc[16];
   float acc = 0.f;
-  for (int ti = blockIdx.y; ti < n_tensors; ti += gridDim.y) {
+  const long stride = (long)gridDim.x * blockDim.x * 4;
+  for (int ti = 0; ti < n_tensors; ++ti) {
     const NormMeta mt = metas[ti];
-    const long stride = (long)gridDim.x * blockDim.x;
-    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < mt.n;
-         i += stride) {
-      const float v = mt.is_bf16
-                          ? to_f32(reinterpret_cast<const bf16*>(mt.g)[i])
-                          : reinterpret_cast<const float*>(mt.g)[i];
-      acc += v * v;
+    const long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+    if (mt.is_bf16) {
+      const bf16* g = reinterpret_cast<const bf16*>(mt.g);
+      long i = base;
+      for (; i + 3 < mt.n; i += stride) {
+        const s16x4 v4 = *reinterpret_cast<const s16x4*>(g + i);
+        const bf16* vb = reinterpret_cast<const bf16*>(&v4);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const float v = to_f32(vb[e]);
+          acc += v * v;
+        }
+      }
+      for (; i < mt.n; ++i) {  // ragged tail (rare: params are 4-aligned)
+        const float v = to_f32(g[i]);
+        acc += v * v;
+      }
+    } else {
+      const float* g = reinterpret_cast<const float*>(mt.g);
+      long i = base;
+      for (; i + 3 < mt.n; i += stride) {
+        const f32x4 v4 = *reinterpret_cast<const f32x4*>(g + i);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) acc += v4[e] * v4[e];
+      }
+      for (; i < mt.n; ++i) acc += g[i] * g[i];
     }
   }
-  // wave reduce then one atomic per wave
   for (int off = WAVE / 2; off > 0; off >>= 1)
     acc += __shfl_down(acc, off, WAVE);
-  if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(out, acc);
+  const int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) warp_acc[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.f;
+    for (int w = 0; w < (int)(blockDim.x / WAVE); ++w) s += warp_acc[w];
+    atomicAdd(out, s);
+  }
 }
 
 // Returns a 1-element fp32 tensor holding sum of squares (caller sqrts).
@@ -148,8 +185,8 @@ at::Tensor multi_tensor_sumsq(std::vector<at::Tensor> grads) {
                                  at::TensorOptions().dtype(at::kByte))
                        .to(grads[0].device(), false);
   const int block = 256;
-  const int gx = std::min<long>((total / n + block - 1) / block, 256);
-  dim3 grid(std::max(gx, 1), std::min(n, 64));
+  // ~2048 waves total saturate HBM for a read-only reduction
+  dim3 grid(512, 1);
   hipLaunchKernelGGL(multi_tensor_l2_kernel, grid, dim3(block), 0,
                      cur_stream(grads[0]), (NormMeta*)meta_blob.data_ptr(), n,
                      out.data_ptr<float>());
@@ -168,17 +205,32 @@ struct ScaleMeta {
 __global__ void multi_tensor_scale_kernel(ScaleMeta* metas, int n_tensors,
                                           const float* scale) {
   const float s = *scale;
+  const long stride = (long)gridDim.x * blockDim.x * 4;
   for (int ti = blockIdx.y; ti < n_tensors; ti += gridDim.y) {
     const ScaleMeta mt = metas[ti];
-    const long stride = (long)gridDim.x * blockDim.x;
-    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < mt.n;
-         i += stride) {
-      if (mt.is_bf16) {
-        bf16* g = reinterpret_cast<bf16*>(mt.g);
-        g[i] = __float2bfloat16(to_f32(g[i]) * s);
-      } else {
-        reinterpret_cast<float*>(mt.g)[i] *= s;
+    const long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+    if (mt.is_bf16) {
+      bf16* g = reinterpret_cast<bf16*>(mt.g);
+      long i = base;
+      for (; i + 3 < mt.n; i += stride) {
+        s16x4 v4 = *reinterpret_cast<const s16x4*>(g + i);
+        bf16* vb = reinterpret_cast<bf16*>(&v4);
+#pragma unroll
+        for (int e = 0; e < 4; ++e)
+          vb[e] = __float2bfloat16(to_f32(vb[e]) * s);
+        *reinterpret_cast<s16x4*>(g + i) = v4;
       }
+      for (; i < mt.n; ++i) g[i] = __float2bfloat16(to_f32(g[i]) * s);
+    } else {
+      float* g = reinterpret_cast<float*>(mt.g);
+      long i = base;
+      for (; i + 3 < mt.n; i += stride) {
+        f32x4 v4 = *reinterpret_cast<const f32x4*>(g + i);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) v4[e] *= s;
+        *reinterpret_cast<f32x4*>(g + i) = v4;
+      }
+      for (; i < mt.n; ++i) g[i] *= s;
     }
   }
 }
