@@ -58,6 +58,9 @@ def main():
         "vocab_size": 21128, "label_size": 10,
         "num_train_steps": 100000, "step_per_epoch": 1000,
     })
+    if use_gpu:
+        from chinesener_amd.ops.tunable import load_tuned_gemm_table
+        load_tuned_gemm_table()
     model = build_model(args.model, params).to(device)
     use_bf16 = use_gpu and params.get("dtype", "bf16") == "bf16"
     if use_bf16:

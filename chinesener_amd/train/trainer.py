@@ -31,6 +31,9 @@ class Trainer:
         self.params = params
         self.rank = rank
         self.device = device or ("cuda" if torch.cuda.is_available() else "cpu")
+        if self.device.startswith("cuda"):
+            from ..ops.tunable import load_tuned_gemm_table
+            load_tuned_gemm_table()
         self.model.to(self.device)
         # pure-bf16 weights + fp32 masters in the fused optimizer (no
         # autocast weight-cast kernels; see train/precision.py)

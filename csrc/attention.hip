@@ -206,7 +206,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 // backward: P recomputed from lse; dV = P^T dO ; dS = P*(dP - delta);
 // dK = dS^T Q ; dQ = dS K. P^T is overwritten by dS^T in place.
 // ---------------------------------------------------------------------
-__global__ __launch_bounds__(256) void attn_bwd_kernel(
+__global__ __launch_bounds__(512) void attn_bwd_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q,
     const bf16* __restrict__ k, const bf16* __restrict__ v,
     const bf16* __restrict__ o, const float* __restrict__ lse,
@@ -239,22 +239,33 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   stage_tile(k + qb, q_rs, k_s, L, Lpad, D);
   stage_tile(v + qb, q_rs, v_s, L, Lpad, D);
   stage_tile(dout + ob, o_rs, do_s, L, Lpad, D);
-  for (int r = threadIdx.x; r < Lpad; r += blockDim.x) {
+  // delta[r] = dot(dO[r], O[r]) — one wave per row, 8-wide vector loads
+  for (int r = wid; r < Lpad; r += blockDim.x / WAVE) {
     float acc = 0.f;
     if (r < L) {
-      for (int d = 0; d < D; ++d)
-        acc += to_f32(dout[ob + (long)r * o_rs + d]) *
-               to_f32(o[ob + (long)r * o_rs + d]);
-      lse_s[r] = lse[(long)bh * L + r];
-    } else {
+      for (int d = lane * 8; d < D; d += WAVE * 8) {
+        const s16x8 vdo =
+            *reinterpret_cast<const s16x8*>(dout + ob + (long)r * o_rs + d);
+        const s16x8 vo =
+            *reinterpret_cast<const s16x8*>(o + ob + (long)r * o_rs + d);
+        const bf16* pd = reinterpret_cast<const bf16*>(&vdo);
+        const bf16* po = reinterpret_cast<const bf16*>(&vo);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) acc += to_f32(pd[e]) * to_f32(po[e]);
+      }
+      for (int off = WAVE / 2; off > 0; off >>= 1)
+        acc += __shfl_down(acc, off, WAVE);
+      acc = __shfl(acc, 0, WAVE);
+      if (lane == 0) lse_s[r] = lse[(long)bh * L + r];
+    } else if (lane == 0) {
       lse_s[r] = 0.f;
     }
-    delta_s[r] = acc;
+    if (lane == 0) delta_s[r] = acc;
   }
   __syncthreads();
 
   // phase 1: recompute P -> P^T
-  for (int m0 = wid * 16; m0 < Lpad; m0 += 64) {
+  for (int m0 = wid * 16; m0 < Lpad; m0 += (blockDim.x >> 6) << 4) {
     cfrag acc[MAXNF];
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) acc[nf] = cfrag{0.f, 0.f, 0.f, 0.f};
@@ -290,7 +301,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   __syncthreads();
 
   // phase 2: dV[keys] = P^T dO
-  for (int k0 = wid * 16; k0 < L; k0 += 64) {
+  for (int k0 = wid * 16; k0 < L; k0 += (blockDim.x >> 6) << 4) {
     cfrag acc[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) acc[i] = cfrag{0.f, 0.f, 0.f, 0.f};
@@ -318,7 +329,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   __syncthreads();
 
   // phase 3: dP = dO V^T ; dS = P (dP - delta) * scale -> overwrite P^T
-  for (int m0 = wid * 16; m0 < L; m0 += 64) {
+  for (int m0 = wid * 16; m0 < L; m0 += (blockDim.x >> 6) << 4) {
     cfrag acc[MAXNF];
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) acc[nf] = cfrag{0.f, 0.f, 0.f, 0.f};
@@ -353,7 +364,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   __syncthreads();
 
   // phase 4: dK[keys] = dS^T Q
-  for (int k0 = wid * 16; k0 < L; k0 += 64) {
+  for (int k0 = wid * 16; k0 < L; k0 += (blockDim.x >> 6) << 4) {
     cfrag acc[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) acc[i] = cfrag{0.f, 0.f, 0.f, 0.f};
@@ -380,7 +391,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   }
 
   // phase 5: dQ = dS K (A strided from dS^T, B row-major K)
-  for (int m0 = wid * 16; m0 < L; m0 += 64) {
+  for (int m0 = wid * 16; m0 < L; m0 += (blockDim.x >> 6) << 4) {
     cfrag acc[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) acc[i] = cfrag{0.f, 0.f, 0.f, 0.f};
@@ -460,7 +471,7 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
                       * sizeof(bf16) + 2 * s.Lpad * sizeof(float);
   TORCH_CHECK(smem <= 160 * 1024, "attn bwd LDS overflow");
   const long bs = (long)s.H * s.L * s.D, hs = (long)s.L * s.D, rs = s.D;
-  hipLaunchKernelGGL(attn_bwd_kernel, dim3(s.B * s.H), dim3(256), smem,
+  hipLaunchKernelGGL(attn_bwd_kernel, dim3(s.B * s.H), dim3(512), smem,
                      cur_stream(q), (const bf16*)dout.data_ptr(),
                      (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
                      (const bf16*)v.data_ptr(), (const bf16*)o.data_ptr(),
@@ -511,7 +522,7 @@ std::vector<at::Tensor> attn_bwd_qkv(const at::Tensor& dout,
   const long o_bs = (long)s.L * HD, o_hs = s.D, o_rs = HD;
   const bf16* base = (const bf16*)qkv.data_ptr();
   bf16* dbase = (bf16*)dqkv.data_ptr();
-  hipLaunchKernelGGL(attn_bwd_kernel, dim3(s.B * s.H), dim3(256), smem,
+  hipLaunchKernelGGL(attn_bwd_kernel, dim3(s.B * s.H), dim3(512), smem,
                      cur_stream(qkv), (const bf16*)dout.data_ptr(), base,
                      base + HD, base + 2 * HD, (const bf16*)o.data_ptr(),
                      lse.data_ptr<float>(), lens.data_ptr<int>(), dbase,
