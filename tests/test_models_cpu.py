@@ -87,3 +87,48 @@ def test_bigram_use_bert_variant():
     batch = make_tiny_batch("bert_bilstm_crf_bigram")
     out = model(batch)
     assert torch.isfinite(out.loss)
+
+
+def test_tf_bert_name_mapping_loader():
+    """TF-namespace weights load into BertModel with fused qkv and
+    produce the mapped values (reference init_from_checkpoint parity,
+    tools/train_utils.py:91-102)."""
+    import numpy as np
+    import torch
+    from chinesener_amd.models.bert import BertConfig, BertModel
+    from chinesener_amd.models.bert_loader import load_tf_bert
+    H, I, V = 32, 64, 100
+    cfg = BertConfig(vocab_size=V, hidden_size=H, num_hidden_layers=2,
+                     num_attention_heads=2, intermediate_size=I,
+                     max_position_embeddings=40)
+    model = BertModel(cfg)
+    rng = np.random.default_rng(0)
+    tf = {"bert/embeddings/word_embeddings": rng.normal(size=(V, H)),
+          "bert/embeddings/position_embeddings": rng.normal(size=(40, H)),
+          "bert/embeddings/token_type_embeddings": rng.normal(size=(2, H)),
+          "bert/embeddings/LayerNorm/gamma": rng.normal(size=(H,)),
+          "bert/embeddings/LayerNorm/beta": rng.normal(size=(H,))}
+    for i in range(2):
+        p = f"bert/encoder/layer_{i}"
+        for name, shape in [("attention/self/query", (H, H)),
+                            ("attention/self/key", (H, H)),
+                            ("attention/self/value", (H, H)),
+                            ("attention/output/dense", (H, H)),
+                            ("intermediate/dense", (H, I)),
+                            ("output/dense", (I, H))]:
+            tf[f"{p}/{name}/kernel"] = rng.normal(size=shape)
+            tf[f"{p}/{name}/bias"] = rng.normal(size=(shape[1],))
+        for ln in ("attention/output/LayerNorm", "output/LayerNorm"):
+            tf[f"{p}/{ln}/gamma"] = rng.normal(size=(H,))
+            tf[f"{p}/{ln}/beta"] = rng.normal(size=(H,))
+    missing, unexpected = load_tf_bert(model, tf)
+    assert not unexpected
+    # qkv fused correctly: query kernel^T is the first H rows
+    q_t = torch.as_tensor(tf["bert/encoder/layer_0/attention/self/query/kernel"]).T
+    torch.testing.assert_close(model.layers[0].qkv.weight[:H].double(),
+                               q_t.double())
+    v_bias = torch.as_tensor(tf["bert/encoder/layer_0/attention/self/value/bias"])
+    torch.testing.assert_close(model.layers[0].qkv.bias[2 * H:].double(),
+                               v_bias.double())
+    emb = torch.as_tensor(tf["bert/embeddings/word_embeddings"])
+    torch.testing.assert_close(model.embeddings.word.weight.double(), emb.double())
