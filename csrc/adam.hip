@@ -43,6 +43,8 @@ __device__ __forceinline__ void adamw_one(const ChunkMeta& mt, long i,
 template <typename T>
 __global__ void multi_tensor_adamw_kernel(ChunkMeta* metas, int n_tensors,
                                           float b1, float b2, float eps) {
+  // explicit 4-wide vector I/O on every stream (g/p via 4x16b or f32x4,
+  // m/v/master via f32x4) — scalar per-lane math in registers
   const long stride = (long)gridDim.x * blockDim.x * 4;
   for (int ti = blockIdx.y; ti < n_tensors; ti += gridDim.y) {
     const ChunkMeta mt = metas[ti];
@@ -51,8 +53,53 @@ __global__ void multi_tensor_adamw_kernel(ChunkMeta* metas, int n_tensors,
     const long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
     long i = base;
     for (; i + 3 < mt.n; i += stride) {
+      float gv[4], pv[4];
+      if (sizeof(T) == 2) {
+        const s16x4 g4 = *reinterpret_cast<const s16x4*>(g + i);
+        const bf16* gb = reinterpret_cast<const bf16*>(&g4);
 #pragma unroll
-      for (int e = 0; e < 4; ++e) adamw_one<T>(mt, i + e, b1, b2, eps, p, g);
+        for (int e = 0; e < 4; ++e) gv[e] = to_f32(gb[e]);
+      } else {
+        const f32x4 g4 = *reinterpret_cast<const f32x4*>(
+            reinterpret_cast<const float*>(g) + i);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) gv[e] = g4[e];
+      }
+      f32x4 m4 = *reinterpret_cast<const f32x4*>(mt.m + i);
+      f32x4 v4 = *reinterpret_cast<const f32x4*>(mt.v + i);
+      if (mt.master) {
+        const f32x4 ma = *reinterpret_cast<const f32x4*>(mt.master + i);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) pv[e] = ma[e];
+      } else {
+#pragma unroll
+        for (int e = 0; e < 4; ++e)
+          pv[e] = sizeof(T) == 2
+                      ? to_f32(p[i + e])
+                      : reinterpret_cast<const float*>(p)[i + e];
+      }
+      f32x4 up;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        const float m = b1 * m4[e] + (1.f - b1) * gv[e];
+        const float v = b2 * v4[e] + (1.f - b2) * gv[e] * gv[e];
+        m4[e] = m;
+        v4[e] = v;
+        up[e] = pv[e] -
+                mt.lr * (m * __frcp_rn(sqrtf(v) + eps) + mt.wd * pv[e]);
+      }
+      *reinterpret_cast<f32x4*>(mt.m + i) = m4;
+      *reinterpret_cast<f32x4*>(mt.v + i) = v4;
+      if (mt.master) *reinterpret_cast<f32x4*>(mt.master + i) = up;
+      if (sizeof(T) == 2) {
+        s16x4 p4;
+        bf16* pb = reinterpret_cast<bf16*>(&p4);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) pb[e] = __float2bfloat16(up[e]);
+        *reinterpret_cast<s16x4*>(p + i) = p4;
+      } else {
+        *reinterpret_cast<f32x4*>(reinterpret_cast<float*>(p) + i) = up;
+      }
     }
     for (; i < mt.n; ++i) adamw_one<T>(mt, i, b1, b2, eps, p, g);
   }
