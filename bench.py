@@ -59,8 +59,10 @@ def main():
         "num_train_steps": 100000, "step_per_epoch": 1000,
     })
     if use_gpu:
-        from chinesener_amd.ops.tunable import load_tuned_gemm_table
-        load_tuned_gemm_table()
+        # tune unseen GEMM shapes during the untimed warmup steps, then
+        # freeze before the timed region (committed table warm-starts)
+        from chinesener_amd.ops.tunable import freeze, load_tuned_gemm_table
+        load_tuned_gemm_table(tune=True)
     model = build_model(args.model, params).to(device)
     use_bf16 = use_gpu and params.get("dtype", "bf16") == "bf16"
     if use_bf16:
@@ -107,6 +109,9 @@ def main():
 
     for i in range(args.warmup):
         train_step(batches[i % len(batches)])
+    if use_gpu:
+        freeze(os.path.join("gpurun_out", f"tunableop_tuned_r{rank}.csv")
+               if os.path.isdir("gpurun_out") and rank == 0 else None)
     if dist is not None:
         dist.barrier()
     if use_gpu:

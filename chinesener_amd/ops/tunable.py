@@ -17,20 +17,38 @@ _DEFAULT = os.path.join(os.path.dirname(os.path.dirname(
     "tunableop_gfx950.csv")
 
 
-def load_tuned_gemm_table(path: str | None = None) -> bool:
-    path = path or _DEFAULT
-    if not os.path.exists(path):
+def load_tuned_gemm_table(path: str | None = None, tune: bool = False) -> bool:
+    """Enable TunableOp GEMM selection.
+
+    tune=True keeps tuning ON: unseen shapes are tuned on first
+    encounter (do this during warmup, then call freeze() before the
+    timed/production region). The committed table warm-starts so only
+    novel shapes pay the tuning cost."""
+    if os.environ.get("CHINESENER_NO_TUNABLE") == "1":
         return False
+    path = path or _DEFAULT
     try:
         import torch
         if not torch.cuda.is_available():
             return False
         import torch.cuda.tunable as tunable
         tunable.enable(True)
-        tunable.tuning_enable(False)
-        tunable.read_file(path)
-        log.info("TunableOp GEMM table loaded from %s", path)
+        if os.path.exists(path):
+            tunable.read_file(path)
+            log.info("TunableOp GEMM table loaded from %s", path)
+        tunable.tuning_enable(bool(tune))
         return True
     except Exception as e:  # pragma: no cover
         log.warning("TunableOp load failed: %s", e)
         return False
+
+
+def freeze(dump_path: str | None = None) -> None:
+    """Stop tuning (call after warmup); optionally dump the winners."""
+    try:
+        import torch.cuda.tunable as tunable
+        tunable.tuning_enable(False)
+        if dump_path:
+            tunable.write_file(dump_path)
+    except Exception as e:  # pragma: no cover
+        log.warning("TunableOp freeze failed: %s", e)
