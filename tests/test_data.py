@@ -138,3 +138,29 @@ def test_make_synthetic_batch_softlexicon():
     assert b["softlexicon_ids"].shape == (2, 16, 40)
     s = b["softlexicon_weights"].sum(-1)
     assert torch.allclose(s, torch.ones_like(s), atol=1e-5)
+
+
+def test_text_embedding_loader(tmp_path):
+    # glove format (no header)
+    p = tmp_path / "glove.txt"
+    p.write_text("你 0.1 0.2 0.3\n好 0.4 0.5 0.6\n", encoding="utf-8")
+    from chinesener_amd.data.embeddings import (add_special_tokens,
+                                                combine_embeddings,
+                                                load_text_embeddings,
+                                                normalize_rows)
+    vocab, mat = load_text_embeddings(str(p))
+    assert vocab == ["你", "好"] and mat.shape == (2, 3)
+    # word2vec format (header)
+    p2 = tmp_path / "w2v.txt"
+    p2.write_text("2 3\n早 1 0 0\n安 0 1 0\n", encoding="utf-8")
+    v2, m2 = load_text_embeddings(str(p2))
+    assert v2 == ["早", "安"]
+    n = normalize_rows(m2)
+    assert abs(float((n[0] ** 2).sum()) - 1.0) < 1e-5
+    sv, sm = add_special_tokens(vocab, mat)
+    assert sv[:3] == ["<None>", "<PAD>", "<eos>"]
+    assert (sm[1] == 0).all()  # PAD row zero
+    cv, cm = combine_embeddings(vocab, mat, ["你", "你好"],
+                                np.array([[9, 9, 9], [7, 7, 7]], np.float32))
+    assert cv == ["你", "好", "你好"]
+    assert cm.shape == (3, 3) and (cm[2] == 7).all()
