@@ -17,6 +17,19 @@ import os
 import sys
 import time
 
+# hipBLASLt GEMM algorithm tuning (TunableOp) must be configured via env
+# BEFORE torch initializes; tuning runs lazily inside the untimed warmup
+# steps (warm-started from the committed table), so timed steps see only
+# tuned algos. CHINESENER_NO_TUNABLE=1 opts out.
+if os.environ.get("CHINESENER_NO_TUNABLE") != "1":
+    _repo = os.path.dirname(os.path.abspath(__file__))
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", "100")
+    os.environ.setdefault(
+        "PYTORCH_TUNABLEOP_FILENAME",
+        os.path.join(_repo, "profiles", "tunableop_gfx950_.csv"))
+
 import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -59,8 +72,6 @@ def main():
         "num_train_steps": 100000, "step_per_epoch": 1000,
     })
     if use_gpu:
-        # tune unseen GEMM shapes during the untimed warmup steps, then
-        # freeze before the timed region (committed table warm-starts)
         from chinesener_amd.ops.tunable import freeze, load_tuned_gemm_table
         load_tuned_gemm_table(tune=True)
     model = build_model(args.model, params).to(device)
@@ -110,8 +121,7 @@ def main():
     for i in range(args.warmup):
         train_step(batches[i % len(batches)])
     if use_gpu:
-        freeze(os.path.join("gpurun_out", f"tunableop_tuned_r{rank}.csv")
-               if os.path.isdir("gpurun_out") and rank == 0 else None)
+        freeze()
     if dist is not None:
         dist.barrier()
     if use_gpu:

@@ -26,6 +26,8 @@ def load_tuned_gemm_table(path: str | None = None, tune: bool = False) -> bool:
     novel shapes pay the tuning cost."""
     if os.environ.get("CHINESENER_NO_TUNABLE") == "1":
         return False
+    if "PYTORCH_TUNABLEOP_ENABLED" in os.environ:
+        return True  # env-configured at process start — authoritative
     path = path or _DEFAULT
     try:
         import torch
@@ -44,11 +46,12 @@ def load_tuned_gemm_table(path: str | None = None, tune: bool = False) -> bool:
 
 
 def freeze(dump_path: str | None = None) -> None:
-    """Stop tuning (call after warmup); optionally dump the winners."""
+    """Stop tuning (call after warmup). This torch has no write_file;
+    results are flushed to PYTORCH_TUNABLEOP_FILENAME at process exit."""
+    if "PYTORCH_TUNABLEOP_ENABLED" in os.environ:
+        return  # env-configured: leave the env behaviour alone
     try:
         import torch.cuda.tunable as tunable
         tunable.tuning_enable(False)
-        if dump_path:
-            tunable.write_file(dump_path)
     except Exception as e:  # pragma: no cover
         log.warning("TunableOp freeze failed: %s", e)
