@@ -1,10 +1,12 @@
 """hipBLASLt GEMM algorithm selection via PyTorch TunableOp.
 
-profiles/tunableop_gfx950.csv holds the offline-tuned winners for the
-bench/training GEMM shapes on MI355X (tuned once with
-PYTORCH_TUNABLEOP_TUNING=1; ~14% end-to-end step time on
-bert_bilstm_crf). load_tuned_gemm_table() activates them read-only —
-unknown shapes fall back to the default heuristic."""
+Measured on MI355X: fresh in-process tuning cuts bert_bilstm_crf step
+time ~10% (hipBLASLt's heuristic picks losing algos for several
+backward-GEMM shapes). Saved result tables do NOT reproduce across
+processes — the stored hipBLASLt algo indices are process-local — so
+the supported mode is: enable tuning via env before torch init and let
+the untimed warmup tune (bench.py does exactly this); this module only
+offers the opt-in helpers."""
 from __future__ import annotations
 
 import logging
