@@ -15,6 +15,14 @@ import os
 import pickle
 import sys
 
+# GEMM autotuning must be configured before torch init (see bench.py)
+if os.environ.get("CHINESENER_NO_TUNABLE") != "1":
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", "300")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                          "/tmp/chinesener_tunableop_.csv")
+
 import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
