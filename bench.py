@@ -118,10 +118,16 @@ def main():
         opt.step()
         return out.loss
 
+    t_w0 = time.perf_counter()
     for i in range(args.warmup):
         train_step(batches[i % len(batches)])
     if use_gpu:
         freeze()
+        if rank == 0 and os.environ.get("CHINESENER_BENCH_DEBUG"):
+            import torch.cuda.tunable as _tun
+            print(f"[debug] warmup {time.perf_counter() - t_w0:.1f}s, "
+                  f"tunable enabled={_tun.is_enabled()} "
+                  f"results={len(_tun.get_results())}", file=sys.stderr)
     if dist is not None:
         dist.barrier()
     if use_gpu:
