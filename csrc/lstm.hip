@@ -85,35 +85,55 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     mylen[r] = (b < B) ? lens[b] : 0;
   }
 
-  for (int step = 0; step < L; ++step) {
-    const int t = reverse ? (L - 1 - step) : step;
-    // cooperative staging of gates_x[:,t,dir-slice] for the 16 rows
-    {
-      const int nv = 16 * 4 * h / 8;
-      for (int i = threadIdx.x; i < nv; i += blockDim.x) {
-        const int row = (i * 8) / (4 * h);
-        const int col = (i * 8) % (4 * h);
-        const int b = b0 + row;
-        if (b < B) {
-          const long g = ((long)b * L + t) * gxs + gx_off + col;
-          if (sizeof(T) == 2) {
-            reinterpret_cast<s16x8*>(gx_s)[i] =
-                *reinterpret_cast<const s16x8*>(gates_x + g);
-          } else {
-            const f32x4 lo = *reinterpret_cast<const f32x4*>(gates_x + g);
-            const f32x4 hi = *reinterpret_cast<const f32x4*>(gates_x + g + 4);
-            bf16 packed[8];
+  // software pipeline: each thread owns h/32 8-element chunks of the
+  // [16,4h] gates_x tile; step t+1's chunks are loaded into registers
+  // while step t's MFMAs run, then written to LDS after the epilogue.
+  const int NC = h / 32;                 // chunks per thread (<= 4)
+  long chunk_base[4];
+  int chunk_off[4];
 #pragma unroll
-            for (int e = 0; e < 4; ++e) {
-              packed[e] = __float2bfloat16(lo[e]);
-              packed[e + 4] = __float2bfloat16(hi[e]);
-            }
-            reinterpret_cast<s16x8*>(gx_s)[i] =
-                *reinterpret_cast<const s16x8*>(packed);
-          }
+  for (int c = 0; c < 4; ++c) {
+    if (c >= NC) { chunk_base[c] = -1; continue; }
+    const int i = threadIdx.x + c * blockDim.x;
+    const int row = (i * 8) / (4 * h);
+    const int col = (i * 8) % (4 * h);
+    const int b = b0 + row;
+    chunk_base[c] = (b < B) ? ((long)b * L) * gxs + gx_off + col : -1;
+    chunk_off[c] = i;
+  }
+  s16x8 pre[4];
+  auto load_chunks = [&](int t) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      if (c >= NC || chunk_base[c] < 0) { pre[c] = s16x8{}; continue; }
+      const long g = chunk_base[c] + (long)t * gxs;
+      if (sizeof(T) == 2) {
+        pre[c] = *reinterpret_cast<const s16x8*>(gates_x + g);
+      } else {
+        const f32x4 lo = *reinterpret_cast<const f32x4*>(gates_x + g);
+        const f32x4 hi = *reinterpret_cast<const f32x4*>(gates_x + g + 4);
+        bf16 packed[8];
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          packed[e] = __float2bfloat16(lo[e]);
+          packed[e + 4] = __float2bfloat16(hi[e]);
         }
+        pre[c] = *reinterpret_cast<const s16x8*>(packed);
       }
     }
+  };
+  auto store_chunks = [&]() {
+#pragma unroll
+    for (int c = 0; c < 4; ++c)
+      if (c < NC) reinterpret_cast<s16x8*>(gx_s)[chunk_off[c]] = pre[c];
+  };
+  load_chunks(reverse ? L - 1 : 0);
+  store_chunks();
+
+  for (int step = 0; step < L; ++step) {
+    const int t = reverse ? (L - 1 - step) : step;
+    if (step + 1 < L)  // issue next step's loads early (latency overlap)
+      load_chunks(reverse ? (L - 2 - step) : (step + 1));
     // gates = h_prev @ W_hh  (+ gates_x added in the epilogue)
     cfrag acc[4][2];
 #pragma unroll
@@ -132,7 +152,7 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
               ah, lds_frag_l(wT, col0, h, kk * 32), acc[g][jj], 0, 0, 0);
         }
     }
-    __syncthreads();  // hb reads done; safe to overwrite below
+    __syncthreads();  // hb + gx_s reads of this step done below this point
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int b = b0 + lrow + r;
@@ -161,15 +181,16 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
         const long obase = ((long)b * L + t) * hss + hs_off + j;
         from_f32(valid ? h_new : 0.f, &hs[obase]);
         cs[(dbase + (long)b * L + t) * h + j] = c_reg[r][jj];
-        gates_out[gbase + 0 * h + j] = gi;
-        gates_out[gbase + 1 * h + j] = gf;
-        gates_out[gbase + 2 * h + j] = gg;
-        gates_out[gbase + 3 * h + j] = go;
+        // interleaved [.., h, 4] layout: one f32x4 store here, one f32x4
+        // load per (r,jj) in backward
+        const f32x4 g4 = {gi, gf, gg, go};
+        *reinterpret_cast<f32x4*>(gates_out + gbase + (long)j * 4) = g4;
         hb[(lrow + r) * h + j] =
             __float2bfloat16(valid ? h_new : to_f32(hb[(lrow + r) * h + j]));
       }
     }
-    __syncthreads();  // hb updated for next step
+    __syncthreads();  // hb updated; gx_s safe to overwrite for next step
+    if (step + 1 < L) store_chunks();
   }
 }
 
@@ -208,15 +229,47 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
     mylen[r] = (b < B) ? lens[b] : 0;
   }
 
+  // software pipeline: per-thread loads for step t+1 (gates f32x4 in the
+  // interleaved [..,h,4] layout, c_t, c_prev, upstream dh) are issued
+  // while step t's MFMA runs.
+  f32x4 pg[4][2];      // gates (i,f,g,o) per (r,jj)
+  float pc[4][2], pcp[4][2], pdh[4][2];
+
+  auto load_step = [&](int t) {
+    const int t_prev = reverse ? (t + 1) : (t - 1);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int b = b0 + lrow + r;
+      const bool valid = (b < B) && (t < mylen[r]);
+      const long gbase = valid ? ((dbase + (long)b * L + t) * 4 * h) : 0;
+#pragma unroll
+      for (int jj = 0; jj < 2; ++jj) {
+        const int j = wid * 32 + jj * 16 + (lane & 15);
+        if (j >= h || !valid) {
+          pg[r][jj] = f32x4{};
+          pc[r][jj] = pcp[r][jj] = pdh[r][jj] = 0.f;
+          continue;
+        }
+        const long obase = (dbase + (long)b * L + t) * h + j;
+        pg[r][jj] =
+            *reinterpret_cast<const f32x4*>(gates + gbase + (long)j * 4);
+        pc[r][jj] = cs[obase];
+        pcp[r][jj] = (t_prev >= 0 && t_prev < L && t_prev < mylen[r])
+                         ? cs[(dbase + (long)b * L + t_prev) * h + j]
+                         : 0.f;
+        pdh[r][jj] = to_f32(dhs[((long)b * L + t) * hss + hs_off + j]);
+      }
+    }
+  };
+
+  load_step(reverse ? 0 : L - 1);
   for (int step = L - 1; step >= 0; --step) {
     const int t = reverse ? (L - 1 - step) : step;  // reverse of fwd order
-    const int t_prev = reverse ? (t + 1) : (t - 1);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int b = b0 + lrow + r;
       const bool inb = b < B;
       const bool valid = inb && (t < mylen[r]);
-      const long gbase = inb ? ((dbase + (long)b * L + t) * 4 * h) : 0;
       const long xbase = inb ? (((long)b * L + t) * gxs + gx_off) : 0;
 #pragma unroll
       for (int jj = 0; jj < 2; ++jj) {
@@ -224,19 +277,14 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
         if (j >= h) continue;
         float dgi = 0.f, dgf = 0.f, dgg = 0.f, dgo = 0.f;
         if (valid) {
-          const long obase = (dbase + (long)b * L + t) * h + j;
-          const float gi = gates[gbase + 0 * h + j];
-          const float gf = gates[gbase + 1 * h + j];
-          const float gg = gates[gbase + 2 * h + j];
-          const float go = gates[gbase + 3 * h + j];
-          const float c_t = cs[obase];
-          const float c_prev =
-              (t_prev >= 0 && t_prev < L && t_prev < mylen[r])
-                  ? cs[(dbase + (long)b * L + t_prev) * h + j]
-                  : 0.f;
+          const float gi = pg[r][jj][0];
+          const float gf = pg[r][jj][1];
+          const float gg = pg[r][jj][2];
+          const float go = pg[r][jj][3];
+          const float c_t = pc[r][jj];
+          const float c_prev = pcp[r][jj];
           const float ac = act_f(c_t, relu);
-          const float dh =
-              dh_reg[r][jj] + to_f32(dhs[((long)b * L + t) * hss + hs_off + j]);
+          const float dh = dh_reg[r][jj] + pdh[r][jj];
           float dc = dc_reg[r][jj] + dh * go * dact_from_out(ac, relu);
           dgo = dh * ac * go * (1.f - go);
           // clamp boundary: no grad through a clipped cell state
@@ -261,6 +309,8 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
       }
     }
     __syncthreads();
+    if (step - 1 >= 0)  // prefetch next iteration while the MFMAs run
+      load_step(reverse ? (L - step) : (step - 1));
     // dh_prev = dgates @ W_hh^T : [16,4h] @ [4h,h]; wave w owns output
     // cols [w*32,(w+1)*32) = N-frags {2w, 2w+1}
     cfrag acc[2];
