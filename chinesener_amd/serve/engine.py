@@ -165,3 +165,16 @@ class InferenceEngine:
         dev = {k: t.to(self.device) for k, t in tensors.items()}
         out = self.model(dev, compute_pred=True)
         return out.pred_ids.cpu().numpy()
+
+
+class FastPredict:
+    """Reference FastPredict parity (tools/fast_predict.py:10-38): the
+    reference holds an Estimator.predict generator open to avoid graph
+    reload per call; here the captured hipGraph engine IS that persistent
+    predictor — this thin alias keeps the familiar API."""
+
+    def __init__(self, engine: InferenceEngine):
+        self.engine = engine
+
+    def predict(self, features):
+        return self.engine.predict(features)

@@ -18,6 +18,7 @@ from .checkpoints import CheckpointManager
 from .metrics import TagMetrics
 from .optimizers import build_optimizer, clip_gradients
 from .precision import convert_bf16_mixed, wants_pure_bf16
+from ..utils.profiling import roctx_range
 
 log = logging.getLogger("chinesener_amd")
 
@@ -69,14 +70,18 @@ class Trainer:
             self.dp.zero_grad()
         else:
             self.optimizer.zero_grad(set_to_none=True)
-        out = self._forward(batch)
-        out.loss.backward()
+        with roctx_range("forward"):
+            out = self._forward(batch)
+        with roctx_range("backward"):
+            out.loss.backward()
         if self.dp is not None:
-            self.dp.finalize_backward()      # wait bucketed all-reduces
-        clip_gradients(self.model, self.family)
-        self.step += 1
-        self.schedule.apply(self.optimizer, self.step)
-        self.optimizer.step()
+            with roctx_range("allreduce_wait"):
+                self.dp.finalize_backward()  # wait bucketed all-reduces
+        with roctx_range("clip+optimizer"):
+            clip_gradients(self.model, self.family)
+            self.step += 1
+            self.schedule.apply(self.optimizer, self.step)
+            self.optimizer.step()
         return float(out.loss.detach())
 
     def train(self, batches: Iterable[Dict[str, torch.Tensor]],
