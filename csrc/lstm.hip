@@ -229,47 +229,15 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
     mylen[r] = (b < B) ? lens[b] : 0;
   }
 
-  // software pipeline: per-thread loads for step t+1 (gates f32x4 in the
-  // interleaved [..,h,4] layout, c_t, c_prev, upstream dh) are issued
-  // while step t's MFMA runs.
-  f32x4 pg[4][2];      // gates (i,f,g,o) per (r,jj)
-  float pc[4][2], pcp[4][2], pdh[4][2];
-
-  auto load_step = [&](int t) {
-    const int t_prev = reverse ? (t + 1) : (t - 1);
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int b = b0 + lrow + r;
-      const bool valid = (b < B) && (t < mylen[r]);
-      const long gbase = valid ? ((dbase + (long)b * L + t) * 4 * h) : 0;
-#pragma unroll
-      for (int jj = 0; jj < 2; ++jj) {
-        const int j = wid * 32 + jj * 16 + (lane & 15);
-        if (j >= h || !valid) {
-          pg[r][jj] = f32x4{};
-          pc[r][jj] = pcp[r][jj] = pdh[r][jj] = 0.f;
-          continue;
-        }
-        const long obase = (dbase + (long)b * L + t) * h + j;
-        pg[r][jj] =
-            *reinterpret_cast<const f32x4*>(gates + gbase + (long)j * 4);
-        pc[r][jj] = cs[obase];
-        pcp[r][jj] = (t_prev >= 0 && t_prev < L && t_prev < mylen[r])
-                         ? cs[(dbase + (long)b * L + t_prev) * h + j]
-                         : 0.f;
-        pdh[r][jj] = to_f32(dhs[((long)b * L + t) * hss + hs_off + j]);
-      }
-    }
-  };
-
-  load_step(reverse ? 0 : L - 1);
   for (int step = L - 1; step >= 0; --step) {
     const int t = reverse ? (L - 1 - step) : step;  // reverse of fwd order
+    const int t_prev = reverse ? (t + 1) : (t - 1);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int b = b0 + lrow + r;
       const bool inb = b < B;
       const bool valid = inb && (t < mylen[r]);
+      const long gbase = inb ? ((dbase + (long)b * L + t) * 4 * h) : 0;
       const long xbase = inb ? (((long)b * L + t) * gxs + gx_off) : 0;
 #pragma unroll
       for (int jj = 0; jj < 2; ++jj) {
@@ -277,14 +245,19 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
         if (j >= h) continue;
         float dgi = 0.f, dgf = 0.f, dgg = 0.f, dgo = 0.f;
         if (valid) {
-          const float gi = pg[r][jj][0];
-          const float gf = pg[r][jj][1];
-          const float gg = pg[r][jj][2];
-          const float go = pg[r][jj][3];
-          const float c_t = pc[r][jj];
-          const float c_prev = pcp[r][jj];
+          const long obase = (dbase + (long)b * L + t) * h + j;
+          // interleaved [..,h,4] gate layout: one f32x4 load
+          const f32x4 g4 =
+              *reinterpret_cast<const f32x4*>(gates + gbase + (long)j * 4);
+          const float gi = g4[0], gf = g4[1], gg = g4[2], go = g4[3];
+          const float c_t = cs[obase];
+          const float c_prev =
+              (t_prev >= 0 && t_prev < L && t_prev < mylen[r])
+                  ? cs[(dbase + (long)b * L + t_prev) * h + j]
+                  : 0.f;
           const float ac = act_f(c_t, relu);
-          const float dh = dh_reg[r][jj] + pdh[r][jj];
+          const float dh =
+              dh_reg[r][jj] + to_f32(dhs[((long)b * L + t) * hss + hs_off + j]);
           float dc = dc_reg[r][jj] + dh * go * dact_from_out(ac, relu);
           dgo = dh * ac * go * (1.f - go);
           // clamp boundary: no grad through a clipped cell state
@@ -309,8 +282,6 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
       }
     }
     __syncthreads();
-    if (step - 1 >= 0)  // prefetch next iteration while the MFMAs run
-      load_step(reverse ? (L - step) : (step - 1));
     // dh_prev = dgates @ W_hh^T : [16,4h] @ [4h,h]; wave w owns output
     // cols [w*32,(w+1)*32) = N-frags {2w, 2w+1}
     cfrag acc[2];
