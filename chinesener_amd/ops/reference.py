@@ -33,10 +33,11 @@ def add_layernorm(x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor,
 
 # ------------------------------------------------------------- bias gelu
 def bias_gelu(x: torch.Tensor, bias: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """BERT's erf-form gelu after a bias add."""
+    """BERT's gelu after a bias add — tanh approximation, the form
+    google-research BERT / the reference's bert_base actually computes."""
     if bias is not None:
         x = x + bias
-    return F.gelu(x)
+    return F.gelu(x, approximate="tanh")
 
 
 # ------------------------------------------------------------- attention

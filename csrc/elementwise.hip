@@ -258,15 +258,25 @@ __global__ __launch_bounds__(256) void layernorm_bwd_bf16_kernel(
 }
 
 // ---------------------------------------------------------------------
-// bias + GELU (erf form, BERT): y = 0.5*(x+b)*(1+erf((x+b)/sqrt(2)))
+// bias + GELU, tanh form — the approximation google-research BERT (and
+// hence the reference's bert_base package) actually computes:
+//   y = 0.5 u (1 + tanh(0.7978845608 (u + 0.044715 u^3)))
+// tanh via exp2: tanh(a) = 1 - 2/(exp2(2a*log2e)+1) — one fast exp2.
 // ---------------------------------------------------------------------
+#define GELU_C0 0.7978845608028654f
+#define GELU_C1 0.044715f
+__device__ __forceinline__ float fast_tanh_f(float a) {
+  const float e = exp2f(a * 2.885390081777927f);  // exp(2a)
+  return 1.f - 2.f / (e + 1.f);
+}
 __device__ __forceinline__ float gelu_f(float u) {
-  return 0.5f * u * (1.f + erff(u * 0.70710678118654752f));
+  return 0.5f * u * (1.f + fast_tanh_f(GELU_C0 * u * (1.f + GELU_C1 * u * u)));
 }
 __device__ __forceinline__ float dgelu_f(float u) {
-  const float cdf = 0.5f * (1.f + erff(u * 0.70710678118654752f));
-  const float pdf = 0.3989422804014327f * __expf(-0.5f * u * u);
-  return cdf + u * pdf;
+  const float a = GELU_C0 * u * (1.f + GELU_C1 * u * u);
+  const float t = fast_tanh_f(a);
+  const float da = GELU_C0 * (1.f + 3.f * GELU_C1 * u * u);
+  return 0.5f * (1.f + t) + 0.5f * u * (1.f - t * t) * da;
 }
 
 template <typename T, bool BWD>
