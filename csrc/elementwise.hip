@@ -291,10 +291,13 @@ __global__ void bias_gelu_kernel(const T* __restrict__ x,
     s16x8 draw{};
     if (BWD) draw = reinterpret_cast<const s16x8*>(dy)[v];
     const int col0 = (int)((v * 8) % F);
+    const f32x4 b_lo = *reinterpret_cast<const f32x4*>(bias + col0);
+    const f32x4 b_hi = *reinterpret_cast<const f32x4*>(bias + col0 + 4);
     T o[8];
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
-      const float u = to_f32(reinterpret_cast<const T*>(&raw)[e]) + bias[col0 + e];
+      const float be = e < 4 ? b_lo[e] : b_hi[e - 4];
+      const float u = to_f32(reinterpret_cast<const T*>(&raw)[e]) + be;
       if (BWD) {
         const float g = to_f32(reinterpret_cast<const T*>(&draw)[e]);
         from_f32(g * dgelu_f(u), &o[e]);
@@ -477,7 +480,7 @@ static at::Tensor bias_gelu_impl(const at::Tensor& x, const at::Tensor& bias,
   auto stream = cur_stream(x);
   const int block = 256;
   if (x.scalar_type() == at::kBFloat16) {
-    const long grid = std::min<long>((n / 8 + block - 1) / block, 4096);
+    const long grid = std::min<long>((n / 8 + block - 1) / block, 32768);
     if (dy) {
       hipLaunchKernelGGL((bias_gelu_kernel<bf16, true>), dim3(grid), dim3(block),
                          0, stream, (const bf16*)x.data_ptr(),
