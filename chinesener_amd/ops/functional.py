@@ -378,11 +378,16 @@ class _SoftLexiconFn(torch.autograd.Function):
 
 
 def softlexicon_fuse(table, ids, weights):
-    """Fused gather-scale-reduce: [V,E] x [B,L,40] -> [B,L,4E]."""
+    """Fused gather-scale-reduce: [V,E] x [B,L,40] -> [B,L,4E].
+
+    Kernel computes in fp32; bf16 tables/weights (pure-bf16 mode) go
+    through differentiable casts so grads flow back in the param dtype."""
     if hip_enabled(table):
-        return _SoftLexiconFn.apply(table.contiguous(),
-                                    ids.to(torch.int32).contiguous(),
-                                    weights.contiguous())
+        out_dtype = table.dtype
+        out = _SoftLexiconFn.apply(table.float().contiguous(),
+                                   ids.to(torch.int32).contiguous(),
+                                   weights.float().contiguous())
+        return out.to(out_dtype)
     return ref.softlexicon_fuse(table, ids, weights)
 
 
