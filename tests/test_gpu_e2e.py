@@ -190,3 +190,30 @@ def test_convergence_entity_f1(tmp_path):
     f1 = rep["micro avg"]["f1"]
     print("entity micro F1 after 800 steps:", f1)
     assert f1 > 0.22, rep["micro avg"]
+
+
+def test_mrc_train_step_gpu(tmp_path):
+    """MRC path on GPU: dynamic padded batch lengths through the fused
+    attention kernels + masked-CE."""
+    _cuda()
+    torch.manual_seed(9)
+    from chinesener_amd.config import resolve_params
+    from chinesener_amd.models import build_model, model_params
+    from chinesener_amd.models.bert import BertConfig
+    from chinesener_amd.mrc.dataset import MrcDataset
+    from chinesener_amd.train.trainer import Trainer
+    cfg = BertConfig(vocab_size=21128, hidden_size=768, num_hidden_layers=2,
+                     num_attention_heads=12, intermediate_size=3072)
+    params = resolve_params(model_params("mrc_bio"), {
+        "bert_config": cfg, "vocab_size": 21128, "label_size": 3,
+        "num_train_steps": 100, "step_per_epoch": 10, "model_name": "mrc_bio"})
+    model = build_model("mrc_bio", params)
+    trainer = Trainer(model, "mrc_bio", params, str(tmp_path / "ck"))
+    ds = MrcDataset(str(tmp_path / "msra"), "msra", batch_size=8,
+                    max_seq_len=170)
+    losses = []
+    for i, batch in enumerate(ds.iter_batches("valid", shuffle=False)):
+        losses.append(trainer.train_step(batch))
+        if i >= 4:
+            break
+    assert all(torch.isfinite(torch.tensor(losses))), losses
