@@ -328,24 +328,48 @@ class _BiLstmFn(torch.autograd.Function):
                 None, None, None)
 
 
+def _pad_gates(w, h, hp):
+    """Zero-pad each i/f/g/o gate segment of a [..., 4h] tensor to hp."""
+    if h == hp:
+        return w
+    shape = list(w.shape[:-1])
+    wv = w.reshape(*shape, 4, h)
+    return F.pad(wv, (0, hp - h)).reshape(*shape, 4 * hp)
+
+
 def bilstm(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b, lens,
            activation: str = "tanh", state_dropout=None,
            cell_clip: float = 0.0):
     """BiLSTM over padded [B,L,E] -> [B,L,2h].
 
-    HIP path: hidden % 32 == 0 and hidden <= 128 (LDS-resident W_hh);
-    other sizes run the torch recurrence (slow path, logged once).
+    HIP path: hidden <= 256 (W_hh LDS-resident up to 128, L2-streamed
+    above); non-multiple-of-32 hidden sizes are zero-padded — padded
+    units stay exactly 0 through the recurrence (sigmoid(0)*act(0)
+    structure), so sliced outputs and grads are exact. Larger sizes run
+    the torch recurrence (slow path, logged once).
     cell_clip > 0 bounds the cell state (TF LSTMCell cell_clip) — the
     relu recurrence needs it to stay off the exponential-growth regime."""
     h = w_hh_f.shape[0]
-    if hip_enabled(x) and h % 32 == 0 and h <= 128:
+    hp = ((h + 31) // 32) * 32
+    if hip_enabled(x) and hp <= 256:
+        if hp != h:
+            w_ih_f = _pad_gates(w_ih_f, h, hp)
+            w_ih_b = _pad_gates(w_ih_b, h, hp)
+            b_f = _pad_gates(b_f, h, hp)
+            b_b = _pad_gates(b_b, h, hp)
+            # W_hh: pad input rows, then each gate segment
+            w_hh_f = _pad_gates(F.pad(w_hh_f, (0, 0, 0, hp - h)), h, hp)
+            w_hh_b = _pad_gates(F.pad(w_hh_b, (0, 0, 0, hp - h)), h, hp)
         # single fused x-projection GEMM for both directions
         w_ih2 = torch.cat([w_ih_f, w_ih_b], dim=1).to(x.dtype)
         b2 = torch.cat([b_f, b_b]).to(x.dtype)
         gates_x = (x @ w_ih2 + b2).contiguous()
-        return _BiLstmFn.apply(gates_x, w_hh_f, w_hh_b,
-                               lens.to(torch.int32), activation == "relu",
-                               float(cell_clip))
+        hs = _BiLstmFn.apply(gates_x, w_hh_f, w_hh_b,
+                             lens.to(torch.int32), activation == "relu",
+                             float(cell_clip))
+        if hp != h:
+            hs = torch.cat([hs[..., :h], hs[..., hp:hp + h]], dim=-1)
+        return hs
     if x.is_cuda:
         global _LSTM_FALLBACK_WARNED
         if not _LSTM_FALLBACK_WARNED:

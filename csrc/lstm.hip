@@ -4,22 +4,32 @@
 // sequential part: per step  gates = gates_x[t] + h_{t-1} @ W_hh,
 // LSTM cell update, variable-length masking.
 //
-// Design: BOTH directions run in ONE launch (blockIdx.y = direction) so
-// their sequence loops overlap on different CUs instead of serializing on
-// the stream — with B=64 each direction is only B/16 workgroups, far under
-// the 256 CUs. Inputs/outputs are strided so the python-side layout is
-// [B,L,8h] gates (fw|bw halves) and [B,L,2h] hidden (the BiLSTM concat is
-// free). Recurrent weights stay resident in LDS for the whole sequence
-// (W_hh bf16 <= 128 KiB for h <= 128); one workgroup (4 waves) owns a
-// 16-row batch tile and loops time in-kernel — no per-step launches.
-// MFMA 16x16x32 bf16 computes [16,4h] gates; wave w owns hidden slice
-// j in [w*32,(w+1)*32) so each lane locally combines its i/f/g/o gates
-// and carries the cell state in registers.
+// Design:
+//  * BOTH directions run in ONE launch (blockIdx.y = direction) so their
+//    sequence loops overlap on different CUs.
+//  * Strided I/O: python layout is [B,L,8h] gates (fw|bw halves) and
+//    [B,L,2h] hidden — the BiLSTM concat is free.
+//  * Recurrent weights: LDS-resident for the whole sequence when
+//    4h*h*2B fits (h <= 128); for h in (128, 256] the MFMA B-operands
+//    stream from global memory (L2-resident: W_hh is <= 512 KiB and
+//    re-read every timestep). Hidden sizes must be a multiple of 32 —
+//    the wrapper zero-pads (exact: padded units stay 0 through time).
+//  * One workgroup (4 waves) owns a 16-row batch tile and loops time
+//    in-kernel; MFMA 16x16x32 bf16 computes the [16,4h] gate panel;
+//    column frag f (16 cols) is owned by wave f%4, so each lane locally
+//    combines its i/f/g/o gates and carries cell state in registers.
+//  * Software pipeline: step t+1's gates_x chunks are loaded into
+//    registers while step t's MFMAs run (fwd).
 //
-// fwd stores activated gates + cell states (fp32) for backward; bwd
-// replays in reverse, producing pre-activation gate grads (dgates_x);
-// dW_hh / dW_ih / db / dx are library GEMMs in the wrapper.
+// fwd stores activated gates interleaved ([D,B,L,h,4] — one f32x4 per
+// (row, j) in bwd) + cell states; bwd replays in reverse producing
+// pre-activation gate grads; dW_hh / dW_ih / db / dx are library GEMMs
+// in the wrapper.
 #include "common.h"
+
+#define H_MAX 256
+#define NQ_MAX 4  // column frags per wave: h/16/4 <= 4 for h <= 256
+#define NC_MAX 8  // gates_x chunks per thread: 8h/256 <= 8
 
 using bfrag = mfma_bf16x8;
 using cfrag = mfma_f32x4;
@@ -39,24 +49,24 @@ __device__ __forceinline__ float dact_from_out(float y, bool relu) {
   return relu ? (y > 0.f ? 1.f : 0.f) : (1.f - y * y);
 }
 
-// dir = blockIdx.y. gates_x rows have stride gxs with this direction's
-// slice at column offset dir*4h; hs rows have stride hss, offset dir*h.
-// cs/gates_out are per-direction contiguous ([D,B,L,h] / [D,B,L,4h]).
-// Direction 1 (when D=2) scans the sequence reversed.
+// dir = blockIdx.y. gates_x rows: stride gxs, this direction's slice at
+// column offset dir*4h; hs rows: stride hss, offset dir*h. cs/gates_out
+// per-direction contiguous ([D,B,L,h] / [D,B,L,h,4] interleaved).
+// Direction 1 (when gridDim.y == 2) scans the sequence reversed.
 template <typename T>
 __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     const T* __restrict__ gates_x,    // [B,L,gxs]
     const bf16* __restrict__ w_hh_t,  // [D,4h,h] (transposed by wrapper)
     const int* __restrict__ lens, T* __restrict__ hs,  // [B,L,hss]
     float* __restrict__ cs,                            // [D,B,L,h]
-    float* __restrict__ gates_out,                     // [D,B,L,4h] activated
+    float* __restrict__ gates_out,                     // [D,B,L,h,4]
     int B, int L, int h, int gxs, int hss, bool rev0, bool relu,
     float cell_clip) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  bf16* wT = reinterpret_cast<bf16*>(smem_raw);  // [4h][h]
-  bf16* hb = wT + 4 * h * h;                     // [16][h] current h (bf16)
-  bf16* gx_s = hb + 16 * h;  // [16][4h] staged gates_x (bf16 also for fp32
-                             // inputs: halves the LDS footprint)
+  bf16* hb = reinterpret_cast<bf16*>(smem_raw);  // [16][h] current h
+  bf16* gx_s = hb + 16 * h;                      // [16][4h] staged gates_x
+  bf16* wT = gx_s + 16 * 4 * h;                  // [4h][h] iff w_lds
+  const bool w_lds = h <= 128;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int b0 = blockIdx.x * 16;
@@ -65,18 +75,24 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
   const int gx_off = dir * 4 * h;
   const int hs_off = dir * h;
   const long dbase = (long)dir * B * L;  // row offset into cs/gates_out
+  const bf16* wsrc = w_hh_t + (long)dir * 4 * h * h;
 
-  for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x)
-    reinterpret_cast<s16x8*>(wT)[i] =
-        reinterpret_cast<const s16x8*>(w_hh_t + (long)dir * 4 * h * h)[i];
+  if (w_lds) {
+    for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x)
+      reinterpret_cast<s16x8*>(wT)[i] =
+          reinterpret_cast<const s16x8*>(wsrc)[i];
+    wsrc = wT;
+  }
   for (int i = threadIdx.x; i < 16 * h / 8; i += blockDim.x)
     reinterpret_cast<s16x8*>(hb)[i] = s16x8{};
   __syncthreads();
 
   const int NKK = h / 32;
-  // lane-owned cell state: rows r=0..3 (row = (lane>>4)*4+r), cols jj=0..1
-  // (j = wid*32 + jj*16 + (lane&15))
-  float c_reg[4][2] = {};
+  const int N16 = h / 16;                  // column frags per gate
+  const int NQ = (N16 + 3) / 4;            // frags per wave
+  // lane-owned cell state: rows r=0..3 (row = (lane>>4)*4+r), col frag
+  // f = wid + 4*q, j = f*16 + (lane&15)
+  float c_reg[4][NQ_MAX] = {};
   const int lrow = ((lane >> 4) << 2);  // +r
   int mylen[4];
 #pragma unroll
@@ -85,14 +101,12 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     mylen[r] = (b < B) ? lens[b] : 0;
   }
 
-  // software pipeline: each thread owns h/32 8-element chunks of the
-  // [16,4h] gates_x tile; step t+1's chunks are loaded into registers
-  // while step t's MFMAs run, then written to LDS after the epilogue.
-  const int NC = h / 32;                 // chunks per thread (<= 4)
-  long chunk_base[4];
-  int chunk_off[4];
+  // software pipeline for gates_x staging (see header)
+  const int NC = 8 * h / 256;            // chunks per thread
+  long chunk_base[NC_MAX];
+  int chunk_off[NC_MAX];
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
+  for (int c = 0; c < NC_MAX; ++c) {
     if (c >= NC) { chunk_base[c] = -1; continue; }
     const int i = threadIdx.x + c * blockDim.x;
     const int row = (i * 8) / (4 * h);
@@ -101,11 +115,12 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     chunk_base[c] = (b < B) ? ((long)b * L) * gxs + gx_off + col : -1;
     chunk_off[c] = i;
   }
-  s16x8 pre[4];
+  s16x8 pre[NC_MAX];
   auto load_chunks = [&](int t) {
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      if (c >= NC || chunk_base[c] < 0) { pre[c] = s16x8{}; continue; }
+    for (int c = 0; c < NC_MAX; ++c) {
+      if (c >= NC) break;
+      if (chunk_base[c] < 0) { pre[c] = s16x8{}; continue; }
       const long g = chunk_base[c] + (long)t * gxs;
       if (sizeof(T) == 2) {
         pre[c] = *reinterpret_cast<const s16x8*>(gates_x + g);
@@ -124,7 +139,7 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
   };
   auto store_chunks = [&]() {
 #pragma unroll
-    for (int c = 0; c < 4; ++c)
+    for (int c = 0; c < NC_MAX; ++c)
       if (c < NC) reinterpret_cast<s16x8*>(gx_s)[chunk_off[c]] = pre[c];
   };
   load_chunks(reverse ? L - 1 : 0);
@@ -135,21 +150,22 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     if (step + 1 < L)  // issue next step's loads early (latency overlap)
       load_chunks(reverse ? (L - 2 - step) : (step + 1));
     // gates = h_prev @ W_hh  (+ gates_x added in the epilogue)
-    cfrag acc[4][2];
+    cfrag acc[4][NQ_MAX];
 #pragma unroll
     for (int g = 0; g < 4; ++g)
 #pragma unroll
-      for (int jj = 0; jj < 2; ++jj) acc[g][jj] = cfrag{0.f, 0.f, 0.f, 0.f};
+      for (int q = 0; q < NQ_MAX; ++q) acc[g][q] = cfrag{0.f, 0.f, 0.f, 0.f};
     for (int kk = 0; kk < NKK; ++kk) {
       const bfrag ah = lds_frag_l(hb, 0, h, kk * 32);
 #pragma unroll
       for (int g = 0; g < 4; ++g)
 #pragma unroll
-        for (int jj = 0; jj < 2; ++jj) {
-          if (wid * 32 + jj * 16 >= h) continue;  // h < 128: idle slices
-          const int col0 = g * h + wid * 32 + jj * 16;
-          acc[g][jj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              ah, lds_frag_l(wT, col0, h, kk * 32), acc[g][jj], 0, 0, 0);
+        for (int q = 0; q < NQ_MAX; ++q) {
+          const int f = wid + 4 * q;
+          if (q >= NQ || f >= N16) continue;
+          const int col0 = g * h + f * 16;
+          acc[g][q] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ah, lds_frag_l(wsrc, col0, h, kk * 32), acc[g][q], 0, 0, 0);
         }
     }
     __syncthreads();  // hb + gx_s reads of this step done below this point
@@ -160,29 +176,28 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
       const bool valid = t < mylen[r];
       const long gbase = (dbase + (long)b * L + t) * 4 * h;
 #pragma unroll
-      for (int jj = 0; jj < 2; ++jj) {
-        const int j = wid * 32 + jj * 16 + (lane & 15);
-        if (j >= h) continue;
+      for (int q = 0; q < NQ_MAX; ++q) {
+        const int f = wid + 4 * q;
+        if (q >= NQ || f >= N16) continue;
+        const int j = f * 16 + (lane & 15);
         const int lr4h = (lrow + r) * 4 * h;
-        float gi = acc[0][jj][r] + to_f32(gx_s[lr4h + 0 * h + j]);
-        float gf = acc[1][jj][r] + to_f32(gx_s[lr4h + 1 * h + j]);
-        float gg = acc[2][jj][r] + to_f32(gx_s[lr4h + 2 * h + j]);
-        float go = acc[3][jj][r] + to_f32(gx_s[lr4h + 3 * h + j]);
+        float gi = acc[0][q][r] + to_f32(gx_s[lr4h + 0 * h + j]);
+        float gf = acc[1][q][r] + to_f32(gx_s[lr4h + 1 * h + j]);
+        float gg = acc[2][q][r] + to_f32(gx_s[lr4h + 2 * h + j]);
+        float go = acc[3][q][r] + to_f32(gx_s[lr4h + 3 * h + j]);
         gi = 1.f / (1.f + __expf(-gi));
         gf = 1.f / (1.f + __expf(-gf));
         go = 1.f / (1.f + __expf(-go));
         gg = act_f(gg, relu);
-        float c_new = gf * c_reg[r][jj] + gi * gg;
+        float c_new = gf * c_reg[r][q] + gi * gg;
         // TF LSTMCell cell_clip: bounds the (relu) recurrence
         if (cell_clip > 0.f)
           c_new = fminf(fmaxf(c_new, -cell_clip), cell_clip);
         const float h_new = go * act_f(c_new, relu);
-        if (valid) c_reg[r][jj] = c_new;
+        if (valid) c_reg[r][q] = c_new;
         const long obase = ((long)b * L + t) * hss + hs_off + j;
         from_f32(valid ? h_new : 0.f, &hs[obase]);
-        cs[(dbase + (long)b * L + t) * h + j] = c_reg[r][jj];
-        // interleaved [.., h, 4] layout: one f32x4 store here, one f32x4
-        // load per (r,jj) in backward
+        cs[(dbase + (long)b * L + t) * h + j] = c_reg[r][q];
         const f32x4 g4 = {gi, gf, gg, go};
         *reinterpret_cast<f32x4*>(gates_out + gbase + (long)j * 4) = g4;
         hb[(lrow + r) * h + j] =
@@ -198,14 +213,15 @@ template <typename T>
 __global__ __launch_bounds__(256) void lstm_bwd_kernel(
     const T* __restrict__ dhs,        // [B,L,hss] upstream grad (strided)
     const float* __restrict__ cs,     // [D,B,L,h] carried cell states
-    const float* __restrict__ gates,  // [D,B,L,4h] activated
+    const float* __restrict__ gates,  // [D,B,L,h,4] activated, interleaved
     const bf16* __restrict__ w_hh,    // [D,h,4h] (original layout)
     const int* __restrict__ lens, T* __restrict__ dgates_x,  // [B,L,gxs]
     int B, int L, int h, int gxs, int hss, bool rev0, bool relu,
     float cell_clip) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  bf16* w_s = reinterpret_cast<bf16*>(smem_raw);  // [h][4h]
-  bf16* dg_s = w_s + 4 * h * h;                   // [16][4h]
+  bf16* dg_s = reinterpret_cast<bf16*>(smem_raw);  // [16][4h]
+  bf16* w_s = dg_s + 16 * 4 * h;                   // [h][4h] iff w_lds
+  const bool w_lds = h <= 128;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int b0 = blockIdx.x * 16;
@@ -214,14 +230,20 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
   const int gx_off = dir * 4 * h;
   const int hs_off = dir * h;
   const long dbase = (long)dir * B * L;
-  for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x)
-    reinterpret_cast<s16x8*>(w_s)[i] =
-        reinterpret_cast<const s16x8*>(w_hh + (long)dir * 4 * h * h)[i];
+  const bf16* wsrc = w_hh + (long)dir * 4 * h * h;
+  if (w_lds) {
+    for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x)
+      reinterpret_cast<s16x8*>(w_s)[i] =
+          reinterpret_cast<const s16x8*>(wsrc)[i];
+    wsrc = w_s;
+  }
   __syncthreads();
 
+  const int N16 = h / 16;
+  const int NQ = (N16 + 3) / 4;
   const int lrow = ((lane >> 4) << 2);
-  float dc_reg[4][2] = {};
-  float dh_reg[4][2] = {};
+  float dc_reg[4][NQ_MAX] = {};
+  float dh_reg[4][NQ_MAX] = {};
   int mylen[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -240,13 +262,14 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
       const long gbase = inb ? ((dbase + (long)b * L + t) * 4 * h) : 0;
       const long xbase = inb ? (((long)b * L + t) * gxs + gx_off) : 0;
 #pragma unroll
-      for (int jj = 0; jj < 2; ++jj) {
-        const int j = wid * 32 + jj * 16 + (lane & 15);
-        if (j >= h) continue;
+      for (int q = 0; q < NQ_MAX; ++q) {
+        const int f = wid + 4 * q;
+        if (q >= NQ || f >= N16) continue;
+        const int j = f * 16 + (lane & 15);
         float dgi = 0.f, dgf = 0.f, dgg = 0.f, dgo = 0.f;
         if (valid) {
           const long obase = (dbase + (long)b * L + t) * h + j;
-          // interleaved [..,h,4] gate layout: one f32x4 load
+          // interleaved layout: one f32x4 load
           const f32x4 g4 =
               *reinterpret_cast<const f32x4*>(gates + gbase + (long)j * 4);
           const float gi = g4[0], gf = g4[1], gg = g4[2], go = g4[3];
@@ -257,15 +280,15 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
                   : 0.f;
           const float ac = act_f(c_t, relu);
           const float dh =
-              dh_reg[r][jj] + to_f32(dhs[((long)b * L + t) * hss + hs_off + j]);
-          float dc = dc_reg[r][jj] + dh * go * dact_from_out(ac, relu);
+              dh_reg[r][q] + to_f32(dhs[((long)b * L + t) * hss + hs_off + j]);
+          float dc = dc_reg[r][q] + dh * go * dact_from_out(ac, relu);
           dgo = dh * ac * go * (1.f - go);
           // clamp boundary: no grad through a clipped cell state
           if (cell_clip > 0.f && fabsf(c_t) >= cell_clip) dc = 0.f;
           dgi = dc * gg * gi * (1.f - gi);
           dgf = dc * c_prev * gf * (1.f - gf);
           dgg = dc * gi * dact_from_out(gg, relu);
-          dc_reg[r][jj] = dc * gf;  // carry to previous step
+          dc_reg[r][q] = dc * gf;  // carry to previous step
         }
         // write pre-activation gate grads (global + LDS for the MFMA)
         if (inb) {
@@ -282,27 +305,26 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
       }
     }
     __syncthreads();
-    // dh_prev = dgates @ W_hh^T : [16,4h] @ [4h,h]; wave w owns output
-    // cols [w*32,(w+1)*32) = N-frags {2w, 2w+1}
-    cfrag acc[2];
+    // dh_prev = dgates @ W_hh^T : [16,4h] @ [4h,h]; col frag f = wid+4q
+    cfrag acc[NQ_MAX];
 #pragma unroll
-    for (int jj = 0; jj < 2; ++jj) acc[jj] = cfrag{0.f, 0.f, 0.f, 0.f};
+    for (int q = 0; q < NQ_MAX; ++q) acc[q] = cfrag{0.f, 0.f, 0.f, 0.f};
     for (int kk = 0; kk < 4 * h / 32; ++kk) {
       const bfrag adg = lds_frag_l(dg_s, 0, 4 * h, kk * 32);
 #pragma unroll
-      for (int jj = 0; jj < 2; ++jj) {
-        const int col0 = wid * 32 + jj * 16;
-        if (col0 >= h) continue;
-        acc[jj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            adg, lds_frag_l(w_s, col0, 4 * h, kk * 32), acc[jj], 0, 0, 0);
+      for (int q = 0; q < NQ_MAX; ++q) {
+        const int f = wid + 4 * q;
+        if (q >= NQ || f >= N16) continue;
+        acc[q] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            adg, lds_frag_l(wsrc, f * 16, 4 * h, kk * 32), acc[q], 0, 0, 0);
       }
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const bool valid = t < mylen[r];
 #pragma unroll
-      for (int jj = 0; jj < 2; ++jj)
-        if (valid) dh_reg[r][jj] = acc[jj][r];
+      for (int q = 0; q < NQ_MAX; ++q)
+        if (valid) dh_reg[r][q] = acc[q][r];
       // invalid step: h passed through unchanged -> dh carries unchanged
     }
     __syncthreads();  // dg_s reads done before next step overwrites
@@ -310,9 +332,16 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
 }
 
 // ===================================================================== host
+static size_t lstm_fwd_smem(int h, bool w_lds) {
+  return ((size_t)16 * h + (size_t)16 * 4 * h + (w_lds ? (size_t)4 * h * h : 0))
+         * sizeof(bf16);
+}
+static size_t lstm_bwd_smem(int h, bool w_lds) {
+  return ((size_t)16 * 4 * h + (w_lds ? (size_t)4 * h * h : 0)) * sizeof(bf16);
+}
 
 // Bidirectional fused path: gates_x [B,L,8h] (fw|bw), w_hh_t2 [2,4h,h].
-// Returns hs [B,L,2h] (concat free), cs [2,B,L,h], gates [2,B,L,4h].
+// Returns hs [B,L,2h] (concat free), cs [2,B,L,h], gates [2,B,L,h,4].
 std::vector<at::Tensor> bilstm_fwd_l(const at::Tensor& gates_x,
                                      const at::Tensor& w_hh_t2,
                                      const at::Tensor& lens, bool relu,
@@ -320,8 +349,8 @@ std::vector<at::Tensor> bilstm_fwd_l(const at::Tensor& gates_x,
   CHECK_CUDA_CONTIG(gates_x);
   const int B = gates_x.size(0), L = gates_x.size(1);
   const int h = gates_x.size(2) / 8;
-  TORCH_CHECK(h % 32 == 0 && h <= 128,
-              "bilstm kernel: hidden must be a multiple of 32 and <= 128, got ",
+  TORCH_CHECK(h % 32 == 0 && h <= H_MAX,
+              "bilstm kernel: hidden must be a multiple of 32 and <= 256, got ",
               h);
   TORCH_CHECK(w_hh_t2.size(0) == 2 && w_hh_t2.scalar_type() == at::kBFloat16,
               "w_hh_t2 must be bf16 [2,4h,h]");
@@ -329,7 +358,7 @@ std::vector<at::Tensor> bilstm_fwd_l(const at::Tensor& gates_x,
   auto cs = at::empty({2, B, L, h}, gates_x.options().dtype(at::kFloat));
   auto gates = at::empty({2, B, L, 4 * h},
                          gates_x.options().dtype(at::kFloat));
-  const size_t smem = (size_t)(4 * h * h + 16 * h + 16 * 4 * h) * sizeof(bf16);
+  const size_t smem = lstm_fwd_smem(h, h <= 128);
   TORCH_CHECK(smem <= 160 * 1024, "lstm fwd LDS overflow");
   const dim3 grid((B + 15) / 16, 2);
   auto stream = cur_stream(gates_x);
@@ -361,7 +390,7 @@ at::Tensor bilstm_bwd_l(const at::Tensor& dhs, const at::Tensor& cs,
               "w_hh2 must be bf16 [2,h,4h]");
   auto dhs_c = dhs.contiguous();
   auto dgates_x = at::empty({B, L, 8 * h}, dhs.options());
-  const size_t smem = (size_t)(4 * h * h + 16 * 4 * h) * sizeof(bf16);
+  const size_t smem = lstm_bwd_smem(h, h <= 128);
   TORCH_CHECK(smem <= 160 * 1024, "lstm bwd LDS overflow");
   const dim3 grid((B + 15) / 16, 2);
   auto stream = cur_stream(dhs);
@@ -391,12 +420,12 @@ std::vector<at::Tensor> lstm_fwd(const at::Tensor& gates_x,
   CHECK_CUDA_CONTIG(gates_x);
   const int B = gates_x.size(0), L = gates_x.size(1);
   const int h4 = gates_x.size(2), h = h4 / 4;
-  TORCH_CHECK(h % 32 == 0 && h <= 128, "lstm kernel: bad hidden ", h);
+  TORCH_CHECK(h % 32 == 0 && h <= H_MAX, "lstm kernel: bad hidden ", h);
   auto hs = at::empty({B, L, h}, gates_x.options());
   auto cs = at::empty({1, B, L, h}, gates_x.options().dtype(at::kFloat));
   auto gates = at::empty({1, B, L, h4}, gates_x.options().dtype(at::kFloat));
   auto w_t = w_hh.t().contiguous().to(at::kBFloat16);  // [4h,h]
-  const size_t smem = (size_t)(4 * h * h + 16 * h + 16 * 4 * h) * sizeof(bf16);
+  const size_t smem = lstm_fwd_smem(h, h <= 128);
   TORCH_CHECK(smem <= 160 * 1024, "lstm fwd LDS overflow");
   const dim3 grid((B + 15) / 16, 1);
   auto stream = cur_stream(gates_x);
@@ -425,7 +454,7 @@ std::vector<at::Tensor> lstm_bwd(const at::Tensor& dhs, const at::Tensor& hs,
   const int B = dhs.size(0), L = dhs.size(1), h = dhs.size(2);
   auto dgates_x = at::empty({B, L, 4 * h}, dhs.options());
   auto w_b = w_hh.contiguous().to(at::kBFloat16);  // [h,4h]
-  const size_t smem = (size_t)(4 * h * h + 16 * 4 * h) * sizeof(bf16);
+  const size_t smem = lstm_bwd_smem(h, h <= 128);
   TORCH_CHECK(smem <= 160 * 1024, "lstm bwd LDS overflow");
   const dim3 grid((B + 15) / 16, 1);
   auto stream = cur_stream(dhs);

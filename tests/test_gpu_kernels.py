@@ -365,3 +365,39 @@ def test_mfma_fragment_layout():
     b = torch.randn(32, 16, device="cuda")
     c = ops.get_ext().mfma_probe(a.contiguous(), b.t().contiguous())
     torch.testing.assert_close(c, a @ b, atol=0.2, rtol=0.05)
+
+
+@pytest.mark.parametrize("h", [200, 160, 256])
+def test_bilstm_large_hidden_vs_reference(h):
+    """h > 128 streams W_hh from L2; h=200 exercises the zero-padding
+    path (the softlexicon models' BiLSTM(200))."""
+    _cuda()
+    torch.manual_seed(20)
+    from chinesener_amd.ops import functional as fn
+    B, L, E = 4, 12, 16
+    x32 = (torch.randn(B, L, E, device="cuda") * 0.5)
+    ws32 = [(torch.randn(E, 4 * h, device="cuda") * 0.1).requires_grad_()
+            for _ in range(2)]
+    whs32 = [(torch.randn(h, 4 * h, device="cuda") * 0.1).requires_grad_()
+             for _ in range(2)]
+    bs32 = [(torch.randn(4 * h, device="cuda") * 0.1).requires_grad_()
+            for _ in range(2)]
+    lens = torch.tensor([12, 9, 3, 12], device="cuda")
+
+    xr = x32.detach().clone().requires_grad_()
+    out_ref = ref.bilstm_forward(xr, ws32[0], whs32[0], bs32[0],
+                                 ws32[1], whs32[1], bs32[1], lens, "tanh")
+    (out_ref ** 2).sum().backward()
+    ref_grads = [t.grad.clone() for t in [xr] + ws32 + whs32 + bs32]
+    for t in ws32 + whs32 + bs32:
+        t.grad = None
+
+    x2 = x32.detach().to(torch.bfloat16).requires_grad_()
+    out = fn.bilstm(x2, ws32[0], whs32[0], bs32[0],
+                    ws32[1], whs32[1], bs32[1], lens, "tanh")
+    assert out.shape == (B, L, 2 * h)
+    torch.testing.assert_close(out.float(), out_ref, atol=0.1, rtol=0.1)
+    (out.float() ** 2).sum().backward()
+    got = [x2.grad.float()] + [t.grad for t in ws32 + whs32 + bs32]
+    for g, r in zip(got, ref_grads):
+        torch.testing.assert_close(g, r, atol=0.5, rtol=0.2)
