@@ -77,12 +77,15 @@ class BertLayer(nn.Module):
 
     def forward(self, x, mask, lens):
         B, L, H = x.shape
-        qkv = self.qkv(x).reshape(B, L, 3, self.n_heads, self.head_dim)
+        qkv = ops.linear(x, self.qkv.weight, self.qkv.bias) \
+            .reshape(B, L, 3, self.n_heads, self.head_dim)
         ctx = ops.attention_qkv(qkv, mask=mask, lens=lens).reshape(B, L, H)
-        a = self.dropout(self.attn_out(ctx))
+        a = self.dropout(ops.linear(ctx, self.attn_out.weight,
+                                    self.attn_out.bias))
         x = ops.add_layernorm(a, x, self.ln1_w, self.ln1_b, self.eps)
         f = ops.bias_gelu(self.ffn_in(x), self.ffn_in_bias)
-        f = self.dropout(self.ffn_out(f))
+        f = self.dropout(ops.linear(f, self.ffn_out.weight,
+                                    self.ffn_out.bias))
         return ops.add_layernorm(f, x, self.ln2_w, self.ln2_b, self.eps)
 
 

@@ -20,6 +20,7 @@ std::vector<at::Tensor> masked_ce_fwd(const at::Tensor&, const at::Tensor&,
 at::Tensor masked_ce_bwd(const at::Tensor&, const at::Tensor&,
                          const at::Tensor&, const at::Tensor&,
                          const at::Tensor&);
+at::Tensor colsum(const at::Tensor&);
 // crf.hip
 std::vector<at::Tensor> crf_fwd(const at::Tensor&, const at::Tensor&,
                                 const at::Tensor&, const at::Tensor&);
@@ -82,6 +83,7 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("bias_gelu_bwd", &bias_gelu_bwd);
   m.def("masked_ce_fwd", &masked_ce_fwd);
   m.def("masked_ce_bwd", &masked_ce_bwd);
+  m.def("colsum", &colsum);
   m.def("crf_fwd", &crf_fwd);
   m.def("crf_viterbi", &crf_viterbi);
   m.def("softlexicon_fwd", &softlexicon_fwd);

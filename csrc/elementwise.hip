@@ -550,3 +550,57 @@ at::Tensor masked_ce_bwd(const at::Tensor& dloss, const at::Tensor& probs,
   HIP_CHECK_LAST();
   return dlogits;
 }
+
+// ---------------------------------------------------------------------
+// column sum for bias gradients: dy [M,N] -> db [N] (fp32 accumulation,
+// coalesced row-chunk blocks + one atomic per (block, column)).
+// torch's generic reduce runs ~4.5x off memory bound for this shape.
+// ---------------------------------------------------------------------
+template <typename T>
+__global__ void colsum_kernel(const T* __restrict__ dy, float* __restrict__ ws,
+                              long M, int N, int rows_per_chunk) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= N) return;
+  const long r0 = (long)blockIdx.y * rows_per_chunk;
+  const long r1 = min(M, r0 + rows_per_chunk);
+  float acc = 0.f;
+  for (long r = r0; r < r1; ++r) acc += to_f32(dy[r * N + c]);
+  atomicAdd(&ws[c], acc);
+}
+
+template <typename T>
+__global__ void colsum_cast_kernel(const float* __restrict__ ws,
+                                   T* __restrict__ out, int N) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c < N) from_f32(ws[c], &out[c]);
+}
+
+at::Tensor colsum(const at::Tensor& dy) {
+  CHECK_CUDA_CONTIG(dy);
+  const long M = dy.size(0);
+  const int N = dy.size(1);
+  auto ws = at::zeros({N}, dy.options().dtype(at::kFloat));
+  auto out = at::empty({N}, dy.options());
+  const int block = 256;
+  const int rows_per_chunk = 64;
+  const dim3 grid((N + block - 1) / block,
+                  (unsigned)((M + rows_per_chunk - 1) / rows_per_chunk));
+  auto stream = cur_stream(dy);
+  if (dy.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(colsum_kernel<bf16>, grid, dim3(block), 0, stream,
+                       (const bf16*)dy.data_ptr(), ws.data_ptr<float>(), M, N,
+                       rows_per_chunk);
+    hipLaunchKernelGGL(colsum_cast_kernel<bf16>, dim3((N + 255) / 256),
+                       dim3(256), 0, stream, ws.data_ptr<float>(),
+                       (bf16*)out.data_ptr(), N);
+  } else {
+    hipLaunchKernelGGL(colsum_kernel<float>, grid, dim3(block), 0, stream,
+                       dy.data_ptr<float>(), ws.data_ptr<float>(), M, N,
+                       rows_per_chunk);
+    hipLaunchKernelGGL(colsum_cast_kernel<float>, dim3((N + 255) / 256),
+                       dim3(256), 0, stream, ws.data_ptr<float>(),
+                       out.data_ptr<float>(), N);
+  }
+  HIP_CHECK_LAST();
+  return out;
+}

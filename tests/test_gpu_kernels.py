@@ -401,3 +401,41 @@ def test_bilstm_large_hidden_vs_reference(h):
     got = [x2.grad.float()] + [t.grad for t in ws32 + whs32 + bs32]
     for g, r in zip(got, ref_grads):
         torch.testing.assert_close(g, r, atol=0.5, rtol=0.2)
+
+
+def test_fused_linear_matches_torch():
+    """ops.linear: GemmAndBias forward + custom colsum dbias."""
+    _cuda()
+    torch.manual_seed(21)
+    import torch.nn.functional as F
+    from chinesener_amd import ops as O
+    x = torch.randn(4, 10, 64, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(96, 64, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    b = torch.randn(96, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = O.linear(x, w, b)
+    g = torch.randn_like(y)
+    y.backward(g)
+    got = (x.grad.clone(), w.grad.clone(), b.grad.clone())
+    x2 = x.detach().clone().requires_grad_()
+    w2 = w.detach().clone().requires_grad_()
+    b2 = b.detach().clone().requires_grad_()
+    F.linear(x2, w2, b2).backward(g)
+    torch.testing.assert_close(got[0], x2.grad, atol=1e-2, rtol=1e-2)
+    torch.testing.assert_close(got[1], w2.grad, atol=1e-1, rtol=5e-2)
+    torch.testing.assert_close(got[2].float(), b2.grad.float(),
+                               atol=2e-1, rtol=2e-2)
+
+
+def test_colsum_kernel_exact():
+    _cuda()
+    torch.manual_seed(22)
+    from chinesener_amd import ops as O
+    dy = torch.randn(1000, 130, device="cuda")
+    torch.testing.assert_close(O.get_ext().colsum(dy), dy.sum(0),
+                               atol=1e-3, rtol=1e-4)
+    dyb = dy.to(torch.bfloat16)
+    torch.testing.assert_close(O.get_ext().colsum(dyb).float(),
+                               dyb.float().sum(0), atol=1.0, rtol=2e-2)
