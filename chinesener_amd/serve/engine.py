@@ -53,6 +53,12 @@ class InferenceEngine:
         self.model, self.params = load_exported(name, export_root,
                                                 device=self.device,
                                                 version=version)
+        self.bf16 = (self.device.startswith("cuda")
+                     and self.params.get("dtype", "bf16") == "bf16")
+        if self.bf16:
+            # pure-bf16 inference (fused bf16 attention/LN/GELU kernels)
+            from ..train.precision import convert_bf16_mixed
+            convert_bf16_mixed(self.model)
         self.model.eval()
         self.max_seq_len = max_seq_len or self.params.get("max_seq_len", 150)
         self.batch_sizes = sorted(batch_sizes)
@@ -84,7 +90,12 @@ class InferenceEngine:
             feats["bichar_ids"] = torch.zeros(batch, L, dtype=torch.long)
         if "mtl" in name or "adv" in name:
             feats["task_ids"] = torch.ones(batch, L, dtype=torch.long)
-        return {k: v.to(self.device) for k, v in feats.items()}
+        out = {}
+        for k, v in feats.items():
+            if v.is_floating_point() and self.bf16:
+                v = v.to(torch.bfloat16)
+            out[k] = v.to(self.device)
+        return out
 
     @torch.no_grad()
     def _capture(self, batch: int) -> _CapturedGraph:
@@ -142,7 +153,7 @@ class InferenceEngine:
             if k in _INT_KEYS or a.dtype.kind in "iu":
                 t = t.long()
             else:
-                t = t.float()
+                t = t.to(torch.bfloat16) if self.bf16 else t.float()
             tensors[k] = t
 
         bucket = self._bucket(batch) if self.use_graph else None
