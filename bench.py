@@ -26,7 +26,7 @@ import time
 if os.environ.get("CHINESENER_NO_TUNABLE") != "1":
     os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
     os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
-    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", "300")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", "100")
     os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
                           "/tmp/chinesener_tunableop_.csv")
 
