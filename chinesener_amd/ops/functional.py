@@ -94,8 +94,11 @@ class _LinearFn(torch.autograd.Function):
 
 def linear(x, weight, bias=None):
     """Linear with fused-epilogue forward (GemmAndBias) and custom dbias.
-    Falls back to F.linear off-GPU or without bias."""
-    if bias is not None and hip_enabled(x):
+
+    Used on the uniform-dtype pure-bf16 path; under autocast (mixed
+    param/activation dtypes) or off-GPU it falls back to F.linear."""
+    if (bias is not None and hip_enabled(x) and x.dtype == weight.dtype
+            and not torch.is_autocast_enabled()):
         return _LinearFn.apply(x, weight, bias.to(x.dtype))
     return F.linear(x, weight, bias)
 
