@@ -50,8 +50,10 @@ class SingleEval:
         if topn:
             print(f"\n--- top {topn} mismatched samples ---")
             shown = 0
-            for t, p in zip(self.y_true, self.y_pred):
+            for row, t, p in zip(self.rows, self.y_true, self.y_pred):
                 if t != p and shown < topn:
+                    if "raw" in row:
+                        print("text:", row["raw"])
                     print("true:", " ".join(t))
                     print("pred:", " ".join(p))
                     shown += 1

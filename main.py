@@ -148,6 +148,11 @@ def main(argv=None):
         # dump test predictions per task (reference main.py:52-55, :105-112)
         if len(datas) == 1:
             rows = trainer.predict(pipe.iter_batches("test", shuffle=False))
+            # attach raw sentences so evaluation.py --topn can show text
+            # (the reference pkl carries tokens, main.py:52-55)
+            raws = pipe.raw_sentences("test")
+            for row, raw in zip(rows, raws):
+                row["raw"] = raw
             trainer.dump_predictions(rows, os.path.join(args.data_dir, datas[0]),
                                      name)
         else:
