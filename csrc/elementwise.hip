@@ -168,10 +168,10 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
 }
 
 
-// fast bf16 backward: 8 rows/block (2 per wave), vectorized loads,
+// fast bf16 backward: 32 rows/block (8 per wave), vectorized loads,
 // dw/db accumulated in registers then combined via LDS; H <= 1024.
-// 2 rows/wave (not 8): the per-row wave-reduce chain is latency-bound,
-// so more blocks with shorter serial row loops win.
+// (2 rows/wave was tried: the 4x extra global dw/db atomics cost more
+// than the shorter serial row loop saves.)
 __global__ __launch_bounds__(256) void layernorm_bwd_bf16_kernel(
     const bf16* __restrict__ dy, const bf16* __restrict__ x,
     const float* __restrict__ w, const float* __restrict__ mean,
@@ -196,8 +196,8 @@ __global__ __launch_bounds__(256) void layernorm_bwd_bf16_kernel(
       for (int e = 0; e < 8; ++e) wv[c][e] = w[i * 8 + e];
     }
   }
-  for (int rr = 0; rr < 2; ++rr) {
-    const long row = (long)blockIdx.x * 8 + wid * 2 + rr;
+  for (int rr = 0; rr < 8; ++rr) {
+    const long row = (long)blockIdx.x * 32 + wid * 8 + rr;
     if (row >= N) break;
     const bf16* dyr = dy + row * H;
     const bf16* xr = x + row * H;
@@ -451,7 +451,7 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto stream = cur_stream(x);
   if (x.scalar_type() == at::kBFloat16) {
     TORCH_CHECK(H % 8 == 0 && H <= 1024, "layernorm bwd bf16: bad H ", H);
-    hipLaunchKernelGGL(layernorm_bwd_bf16_kernel, dim3((N + 7) / 8),
+    hipLaunchKernelGGL(layernorm_bwd_bf16_kernel, dim3((N + 31) / 32),
                        dim3(256), smem, stream,
                        (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
                        wf.data_ptr<float>(), mean.data_ptr<float>(),
