@@ -46,9 +46,12 @@ __device__ __forceinline__ cfrag mfma16(bfrag a, bfrag b, cfrag c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
-// stage a [rows x D] global tile (row stride rs) into LDS [rpad x D]
+// stage a [rows x D] global tile (row stride rs) into LDS [rpad x ld]
+// (ld > D pads the row stride so MFMA operand reads are bank-conflict
+// free: ld*2B/4B/4 odd <=> ld % 16 == 8 for the strides used here)
 __device__ __forceinline__ void stage_tile(const bf16* g, long rs, bf16* s,
-                                           int rows, int rpad, int D) {
+                                           int rows, int rpad, int D,
+                                           int ld) {
   const int nv = rpad * D / 8;
   for (int i = threadIdx.x; i < nv; i += blockDim.x) {
     const int r = (i * 8) / D;
@@ -56,13 +59,14 @@ __device__ __forceinline__ void stage_tile(const bf16* g, long rs, bf16* s,
     s16x8 val{};
     if (r < rows)
       val = *reinterpret_cast<const s16x8*>(g + (long)r * rs + c);
-    reinterpret_cast<s16x8*>(s)[i] = val;
+    *reinterpret_cast<s16x8*>(s + (long)r * ld + c) = val;
   }
 }
 
-// stage transposed: global [rows x D] (row stride rs) -> LDS [D x rpad]
+// stage transposed: global [rows x D] (row stride rs) -> LDS [D x ld]
 __device__ __forceinline__ void stage_tile_T(const bf16* g, long rs, bf16* s,
-                                             int rows, int rpad, int D) {
+                                             int rows, int rpad, int D,
+                                             int ld) {
   for (int i = threadIdx.x; i < rpad * D / 8; i += blockDim.x) {
     const int r = (i * 8) / D;
     const int c0 = (i * 8) % D;
@@ -71,7 +75,7 @@ __device__ __forceinline__ void stage_tile_T(const bf16* g, long rs, bf16* s,
       val = *reinterpret_cast<const s16x8*>(g + (long)r * rs + c0);
 #pragma unroll
     for (int e = 0; e < 8; ++e)
-      s[(long)(c0 + e) * rpad + r] = reinterpret_cast<const bf16*>(&val)[e];
+      s[(long)(c0 + e) * ld + r] = reinterpret_cast<const bf16*>(&val)[e];
   }
 }
 
@@ -83,12 +87,12 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const bf16* __restrict__ v, const int* __restrict__ lens,
     bf16* __restrict__ out, float* __restrict__ lse, int B, int H, int L,
     int D, int Lpad, float scale, long q_bs, long q_hs, long q_rs, long o_bs,
-    long o_hs, long o_rs) {
+    long o_hs, long o_rs, int ds, int ls) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  bf16* q_s = reinterpret_cast<bf16*>(smem_raw);    // [Lpad][D]
-  bf16* k_s = q_s + Lpad * D;                       // [Lpad][D]
-  bf16* vt_s = k_s + Lpad * D;                      // [D][Lpad]
-  bf16* p_s = vt_s + D * Lpad;                      // [4 waves][16][Lpad]
+  bf16* q_s = reinterpret_cast<bf16*>(smem_raw);    // [Lpad][ds]
+  bf16* k_s = q_s + Lpad * ds;                      // [Lpad][ds]
+  bf16* vt_s = k_s + Lpad * ds;                     // [D][ls]
+  bf16* p_s = vt_s + D * ls;                        // [4 waves][16][ls]
 
   const int bh = blockIdx.x;
   const int b = bh / H;
@@ -99,15 +103,15 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
 
-  stage_tile(q + qb, q_rs, q_s, L, Lpad, D);
-  stage_tile(k + qb, q_rs, k_s, L, Lpad, D);
-  stage_tile_T(v + qb, q_rs, vt_s, L, Lpad, D);
+  stage_tile(q + qb, q_rs, q_s, L, Lpad, D, ds);
+  stage_tile(k + qb, q_rs, k_s, L, Lpad, D, ds);
+  stage_tile_T(v + qb, q_rs, vt_s, L, Lpad, D, ls);
   __syncthreads();
 
   const int NF = Lpad / 16;
   const int NKK = D / 32;
   const int NFD = D / 16;
-  bf16* pw = p_s + wid * 16 * Lpad;
+  bf16* pw = p_s + wid * 16 * ls;
 
   for (int m0 = wid * 16; m0 < L; m0 += 4 * 16) {
     cfrag acc[MAXNF];
@@ -116,14 +120,14 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     bfrag aq[2];
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
-      if (kk < NKK) aq[kk] = lds_frag(q_s, m0, D, kk * 32);
+      if (kk < NKK) aq[kk] = lds_frag(q_s, m0, ds, kk * 32);
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) {
       if (nf < NF) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
           if (kk < NKK)
-            acc[nf] = mfma16(aq[kk], lds_frag(k_s, nf * 16, D, kk * 32),
+            acc[nf] = mfma16(aq[kk], lds_frag(k_s, nf * 16, ds, kk * 32),
                              acc[nf]);
       }
     }
@@ -164,7 +168,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int lr = ((lane >> 4) << 2) + r;
-          pw[lr * Lpad + nf * 16 + (lane & 15)] = __float2bfloat16(acc[nf][r]);
+          pw[lr * ls + nf * 16 + (lane & 15)] = __float2bfloat16(acc[nf][r]);
         }
       }
     }
@@ -180,11 +184,11 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
     for (int i = 0; i < 4; ++i) oacc[i] = cfrag{0.f, 0.f, 0.f, 0.f};
     for (int kk = 0; kk < Lpad / 32; ++kk) {
-      const bfrag ap = lds_frag(pw, 0, Lpad, kk * 32);
+      const bfrag ap = lds_frag(pw, 0, ls, kk * 32);
 #pragma unroll
       for (int nd = 0; nd < 4; ++nd)
         if (nd < NFD)
-          oacc[nd] = mfma16(ap, lds_frag(vt_s, nd * 16, Lpad, kk * 32),
+          oacc[nd] = mfma16(ap, lds_frag(vt_s, nd * 16, ls, kk * 32),
                             oacc[nd]);
     }
 #pragma unroll
@@ -213,14 +217,14 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
     const int* __restrict__ lens, bf16* __restrict__ dq,
     bf16* __restrict__ dk, bf16* __restrict__ dv, int B, int H, int L, int D,
     int Lpad, float scale, long q_bs, long q_hs, long q_rs, long o_bs,
-    long o_hs, long o_rs) {
+    long o_hs, long o_rs, int ds, int ls) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  bf16* q_s = reinterpret_cast<bf16*>(smem_raw);   // [Lpad][D]
-  bf16* k_s = q_s + Lpad * D;                      // [Lpad][D]
-  bf16* v_s = k_s + Lpad * D;                      // [Lpad][D] row major
-  bf16* do_s = v_s + Lpad * D;                     // [Lpad][D]
-  bf16* pt_s = do_s + Lpad * D;                    // [Lpad][Lpad] P^T / dS^T
-  float* delta_s = reinterpret_cast<float*>(pt_s + (long)Lpad * Lpad);
+  bf16* q_s = reinterpret_cast<bf16*>(smem_raw);   // [Lpad][ds]
+  bf16* k_s = q_s + Lpad * ds;                     // [Lpad][ds]
+  bf16* v_s = k_s + Lpad * ds;                     // [Lpad][ds] row major
+  bf16* do_s = v_s + Lpad * ds;                    // [Lpad][ds]
+  bf16* pt_s = do_s + Lpad * ds;                   // [Lpad][ls] P^T / dS^T
+  float* delta_s = reinterpret_cast<float*>(pt_s + (long)Lpad * ls);
   float* lse_s = delta_s + Lpad;
 
   const int bh = blockIdx.x;
@@ -235,10 +239,10 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
   const int NKK = D / 32;
   const int NFD = D / 16;
 
-  stage_tile(q + qb, q_rs, q_s, L, Lpad, D);
-  stage_tile(k + qb, q_rs, k_s, L, Lpad, D);
-  stage_tile(v + qb, q_rs, v_s, L, Lpad, D);
-  stage_tile(dout + ob, o_rs, do_s, L, Lpad, D);
+  stage_tile(q + qb, q_rs, q_s, L, Lpad, D, ds);
+  stage_tile(k + qb, q_rs, k_s, L, Lpad, D, ds);
+  stage_tile(v + qb, q_rs, v_s, L, Lpad, D, ds);
+  stage_tile(dout + ob, o_rs, do_s, L, Lpad, D, ds);
   // delta[r] = dot(dO[r], O[r]) — one wave per row, 8-wide vector loads
   for (int r = wid; r < Lpad; r += blockDim.x / WAVE) {
     float acc = 0.f;
@@ -272,14 +276,14 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
     bfrag aq[2];
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
-      if (kk < NKK) aq[kk] = lds_frag(q_s, m0, D, kk * 32);
+      if (kk < NKK) aq[kk] = lds_frag(q_s, m0, ds, kk * 32);
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) {
       if (nf < NF) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
           if (kk < NKK)
-            acc[nf] = mfma16(aq[kk], lds_frag(k_s, nf * 16, D, kk * 32),
+            acc[nf] = mfma16(aq[kk], lds_frag(k_s, nf * 16, ds, kk * 32),
                              acc[nf]);
       }
     }
@@ -293,7 +297,7 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
           float p = 0.f;
           if (row < L && col < len)
             p = __expf(acc[nf][r] * scale - lse_s[row]);
-          pt_s[(long)col * Lpad + row] = __float2bfloat16(p);
+          pt_s[(long)col * ls + row] = __float2bfloat16(p);
         }
       }
     }
@@ -306,11 +310,11 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
 #pragma unroll
     for (int i = 0; i < 4; ++i) acc[i] = cfrag{0.f, 0.f, 0.f, 0.f};
     for (int kk = 0; kk < Lpad / 32; ++kk) {
-      const bfrag ap = lds_frag(pt_s, k0, Lpad, kk * 32);
+      const bfrag ap = lds_frag(pt_s, k0, ls, kk * 32);
 #pragma unroll
       for (int nd = 0; nd < 4; ++nd)
         if (nd < NFD)
-          acc[nd] = mfma16(ap, lds_fragB_rowmajor(do_s, kk * 32, D, nd * 16),
+          acc[nd] = mfma16(ap, lds_fragB_rowmajor(do_s, kk * 32, ds, nd * 16),
                            acc[nd]);
     }
 #pragma unroll
@@ -336,14 +340,14 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
     bfrag ado[2];
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
-      if (kk < NKK) ado[kk] = lds_frag(do_s, m0, D, kk * 32);
+      if (kk < NKK) ado[kk] = lds_frag(do_s, m0, ds, kk * 32);
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) {
       if (nf < NF) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
           if (kk < NKK)
-            acc[nf] = mfma16(ado[kk], lds_frag(v_s, nf * 16, D, kk * 32),
+            acc[nf] = mfma16(ado[kk], lds_frag(v_s, nf * 16, ds, kk * 32),
                              acc[nf]);
       }
     }
@@ -354,9 +358,9 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
         for (int r = 0; r < 4; ++r) {
           const int row = m0 + ((lane >> 4) << 2) + r;
           const int col = nf * 16 + (lane & 15);
-          const float p = to_f32(pt_s[(long)col * Lpad + row]);
-          const float ds = p * (acc[nf][r] - delta_s[row]) * scale;
-          pt_s[(long)col * Lpad + row] = __float2bfloat16(ds);
+          const float p = to_f32(pt_s[(long)col * ls + row]);
+          const float dsv = p * (acc[nf][r] - delta_s[row]) * scale;
+          pt_s[(long)col * ls + row] = __float2bfloat16(dsv);
         }
       }
     }
@@ -369,11 +373,11 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
 #pragma unroll
     for (int i = 0; i < 4; ++i) acc[i] = cfrag{0.f, 0.f, 0.f, 0.f};
     for (int kk = 0; kk < Lpad / 32; ++kk) {
-      const bfrag ads = lds_frag(pt_s, k0, Lpad, kk * 32);
+      const bfrag ads = lds_frag(pt_s, k0, ls, kk * 32);
 #pragma unroll
       for (int nd = 0; nd < 4; ++nd)
         if (nd < NFD)
-          acc[nd] = mfma16(ads, lds_fragB_rowmajor(q_s, kk * 32, D, nd * 16),
+          acc[nd] = mfma16(ads, lds_fragB_rowmajor(q_s, kk * 32, ds, nd * 16),
                            acc[nd]);
     }
 #pragma unroll
@@ -400,12 +404,12 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
 #pragma unroll
       for (int e = 0; e < 8; ++e)
         reinterpret_cast<bf16*>(&ads)[e] =
-            pt_s[(long)(kk * 32 + ((lane >> 4) << 3) + e) * Lpad + m0 +
+            pt_s[(long)(kk * 32 + ((lane >> 4) << 3) + e) * ls + m0 +
                  (lane & 15)];
 #pragma unroll
       for (int nd = 0; nd < 4; ++nd)
         if (nd < NFD)
-          acc[nd] = mfma16(ads, lds_fragB_rowmajor(k_s, kk * 32, D, nd * 16),
+          acc[nd] = mfma16(ads, lds_fragB_rowmajor(k_s, kk * 32, ds, nd * 16),
                            acc[nd]);
     }
 #pragma unroll
@@ -437,6 +441,27 @@ static AttnShape attn_shape(int B, int H, int L, int D) {
   return {B, H, L, D, Lpad};
 }
 
+
+// bank-conflict-free LDS strides: pad so stride_bytes/16 is odd
+// (ld % 16 == 8 for bf16); fall back to unpadded when over 160 KiB.
+static void attn_strides(int D, int Lpad, bool bwd, int& ds, int& ls,
+                         size_t& smem) {
+  ds = D + 8;
+  ls = Lpad + 8;
+  auto sz = [&](int ds_, int ls_) -> size_t {
+    if (bwd)
+      return (size_t)(4 * Lpad * ds_ + (long)Lpad * ls_) * sizeof(bf16)
+             + 2 * Lpad * sizeof(float);
+    return (size_t)(2 * Lpad * ds_ + D * ls_ + 4 * 16 * ls_) * sizeof(bf16);
+  };
+  smem = sz(ds, ls);
+  if (smem > 160 * 1024) {  // drop padding (rare: bwd at Lpad > ~168)
+    ds = D;
+    ls = Lpad;
+    smem = sz(ds, ls);
+  }
+}
+
 // old layout API: q,k,v,out all [B,H,L,D]
 std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
                                  const at::Tensor& v, const at::Tensor& lens,
@@ -446,7 +471,9 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
   auto s = attn_shape(q.size(0), q.size(1), q.size(2), q.size(3));
   auto out = at::empty_like(q);
   auto lse = at::empty({s.B, s.H, s.L}, q.options().dtype(at::kFloat));
-  const size_t smem = (size_t)(3 * s.Lpad * s.D + 4 * 16 * s.Lpad) * sizeof(bf16);
+  int ds, ls;
+  size_t smem;
+  attn_strides(s.D, s.Lpad, false, ds, ls, smem);
   TORCH_CHECK(smem <= 160 * 1024, "attn fwd LDS overflow");
   const long bs = (long)s.H * s.L * s.D, hs = (long)s.L * s.D, rs = s.D;
   hipLaunchKernelGGL(attn_fwd_kernel, dim3(s.B * s.H), dim3(256), smem,
@@ -454,7 +481,7 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
                      (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
                      lens.data_ptr<int>(), (bf16*)out.data_ptr(),
                      lse.data_ptr<float>(), s.B, s.H, s.L, s.D, s.Lpad,
-                     (float)scale, bs, hs, rs, bs, hs, rs);
+                     (float)scale, bs, hs, rs, bs, hs, rs, ds, ls);
   HIP_CHECK_LAST();
   return {out, lse};
 }
@@ -467,8 +494,9 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
   auto dq = at::empty_like(q);
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
-  const size_t smem = (size_t)(4 * s.Lpad * s.D + (long)s.Lpad * s.Lpad)
-                      * sizeof(bf16) + 2 * s.Lpad * sizeof(float);
+  int ds, ls;
+  size_t smem;
+  attn_strides(s.D, s.Lpad, true, ds, ls, smem);
   TORCH_CHECK(smem <= 160 * 1024, "attn bwd LDS overflow");
   const long bs = (long)s.H * s.L * s.D, hs = (long)s.L * s.D, rs = s.D;
   hipLaunchKernelGGL(attn_bwd_kernel, dim3(s.B * s.H), dim3(512), smem,
@@ -478,7 +506,7 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
                      lse.data_ptr<float>(), lens.data_ptr<int>(),
                      (bf16*)dq.data_ptr(), (bf16*)dk.data_ptr(),
                      (bf16*)dv.data_ptr(), s.B, s.H, s.L, s.D, s.Lpad,
-                     (float)scale, bs, hs, rs, bs, hs, rs);
+                     (float)scale, bs, hs, rs, bs, hs, rs, ds, ls);
   HIP_CHECK_LAST();
   return {dq, dk, dv};
 }
@@ -492,7 +520,9 @@ std::vector<at::Tensor> attn_fwd_qkv(const at::Tensor& qkv,
   auto s = attn_shape(qkv.size(0), qkv.size(3), qkv.size(1), qkv.size(4));
   auto out = at::empty({s.B, s.L, s.H, s.D}, qkv.options());
   auto lse = at::empty({s.B, s.H, s.L}, qkv.options().dtype(at::kFloat));
-  const size_t smem = (size_t)(3 * s.Lpad * s.D + 4 * 16 * s.Lpad) * sizeof(bf16);
+  int ds, ls;
+  size_t smem;
+  attn_strides(s.D, s.Lpad, false, ds, ls, smem);
   TORCH_CHECK(smem <= 160 * 1024, "attn fwd LDS overflow");
   const long HD = (long)s.H * s.D;
   const long q_bs = (long)s.L * 3 * HD, q_hs = s.D, q_rs = 3 * HD;
@@ -502,7 +532,7 @@ std::vector<at::Tensor> attn_fwd_qkv(const at::Tensor& qkv,
                      cur_stream(qkv), base, base + HD, base + 2 * HD,
                      lens.data_ptr<int>(), (bf16*)out.data_ptr(),
                      lse.data_ptr<float>(), s.B, s.H, s.L, s.D, s.Lpad,
-                     (float)scale, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs);
+                     (float)scale, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs, ds, ls);
   HIP_CHECK_LAST();
   return {out, lse};
 }
@@ -514,8 +544,9 @@ std::vector<at::Tensor> attn_bwd_qkv(const at::Tensor& dout,
   CHECK_CUDA_CONTIG(dout);
   auto s = attn_shape(qkv.size(0), qkv.size(3), qkv.size(1), qkv.size(4));
   auto dqkv = at::empty_like(qkv);
-  const size_t smem = (size_t)(4 * s.Lpad * s.D + (long)s.Lpad * s.Lpad)
-                      * sizeof(bf16) + 2 * s.Lpad * sizeof(float);
+  int ds, ls;
+  size_t smem;
+  attn_strides(s.D, s.Lpad, true, ds, ls, smem);
   TORCH_CHECK(smem <= 160 * 1024, "attn bwd LDS overflow");
   const long HD = (long)s.H * s.D;
   const long q_bs = (long)s.L * 3 * HD, q_hs = s.D, q_rs = 3 * HD;
@@ -527,7 +558,7 @@ std::vector<at::Tensor> attn_bwd_qkv(const at::Tensor& dout,
                      base + HD, base + 2 * HD, (const bf16*)o.data_ptr(),
                      lse.data_ptr<float>(), lens.data_ptr<int>(), dbase,
                      dbase + HD, dbase + 2 * HD, s.B, s.H, s.L, s.D, s.Lpad,
-                     (float)scale, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs);
+                     (float)scale, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs, ds, ls);
   HIP_CHECK_LAST();
   return {dqkv};
 }
