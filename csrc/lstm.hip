@@ -63,9 +63,12 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     int B, int L, int h, int gxs, int hss, bool rev0, bool relu,
     float cell_clip) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  bf16* hb = reinterpret_cast<bf16*>(smem_raw);  // [16][h] current h
-  bf16* gx_s = hb + 16 * h;                      // [16][4h] staged gates_x
-  bf16* wT = gx_s + 16 * 4 * h;                  // [4h][h] iff w_lds
+  // padded strides (odd 16B groups) keep the MFMA operand reads LDS
+  // bank-conflict free; gx_s reads are per-lane scalars (no padding)
+  const int hs_ld = h + 8;
+  bf16* hb = reinterpret_cast<bf16*>(smem_raw);  // [16][hs_ld] current h
+  bf16* gx_s = hb + 16 * hs_ld;                  // [16][4h] staged gates_x
+  bf16* wT = gx_s + 16 * 4 * h;                  // [4h][hs_ld] iff w_lds
   const bool w_lds = h <= 128;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -77,13 +80,18 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
   const long dbase = (long)dir * B * L;  // row offset into cs/gates_out
   const bf16* wsrc = w_hh_t + (long)dir * 4 * h * h;
 
+  int w_ld = h;  // stride of the MFMA B-operand source
   if (w_lds) {
-    for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x)
-      reinterpret_cast<s16x8*>(wT)[i] =
+    for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x) {
+      const int r = (i * 8) / h;
+      const int c = (i * 8) % h;
+      *reinterpret_cast<s16x8*>(wT + (long)r * hs_ld + c) =
           reinterpret_cast<const s16x8*>(wsrc)[i];
+    }
     wsrc = wT;
+    w_ld = hs_ld;
   }
-  for (int i = threadIdx.x; i < 16 * h / 8; i += blockDim.x)
+  for (int i = threadIdx.x; i < 16 * hs_ld / 8; i += blockDim.x)
     reinterpret_cast<s16x8*>(hb)[i] = s16x8{};
   __syncthreads();
 
@@ -156,7 +164,7 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
 #pragma unroll
       for (int q = 0; q < NQ_MAX; ++q) acc[g][q] = cfrag{0.f, 0.f, 0.f, 0.f};
     for (int kk = 0; kk < NKK; ++kk) {
-      const bfrag ah = lds_frag_l(hb, 0, h, kk * 32);
+      const bfrag ah = lds_frag_l(hb, 0, hs_ld, kk * 32);
 #pragma unroll
       for (int g = 0; g < 4; ++g)
 #pragma unroll
@@ -165,7 +173,7 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
           if (q >= NQ || f >= N16) continue;
           const int col0 = g * h + f * 16;
           acc[g][q] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              ah, lds_frag_l(wsrc, col0, h, kk * 32), acc[g][q], 0, 0, 0);
+              ah, lds_frag_l(wsrc, col0, w_ld, kk * 32), acc[g][q], 0, 0, 0);
         }
     }
     __syncthreads();  // hb + gx_s reads of this step done below this point
@@ -200,8 +208,8 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
         cs[(dbase + (long)b * L + t) * h + j] = c_reg[r][q];
         const f32x4 g4 = {gi, gf, gg, go};
         *reinterpret_cast<f32x4*>(gates_out + gbase + (long)j * 4) = g4;
-        hb[(lrow + r) * h + j] =
-            __float2bfloat16(valid ? h_new : to_f32(hb[(lrow + r) * h + j]));
+        hb[(lrow + r) * hs_ld + j] =
+            __float2bfloat16(valid ? h_new : to_f32(hb[(lrow + r) * hs_ld + j]));
       }
     }
     __syncthreads();  // hb updated; gx_s safe to overwrite for next step
@@ -219,8 +227,9 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
     int B, int L, int h, int gxs, int hss, bool rev0, bool relu,
     float cell_clip) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  bf16* dg_s = reinterpret_cast<bf16*>(smem_raw);  // [16][4h]
-  bf16* w_s = dg_s + 16 * 4 * h;                   // [h][4h] iff w_lds
+  const int g_ld = 4 * h + 8;  // padded (bank-conflict-free frag reads)
+  bf16* dg_s = reinterpret_cast<bf16*>(smem_raw);  // [16][g_ld]
+  bf16* w_s = dg_s + 16 * g_ld;                    // [h][g_ld] iff w_lds
   const bool w_lds = h <= 128;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -231,11 +240,16 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
   const int hs_off = dir * h;
   const long dbase = (long)dir * B * L;
   const bf16* wsrc = w_hh + (long)dir * 4 * h * h;
+  int w_ld = 4 * h;
   if (w_lds) {
-    for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x)
-      reinterpret_cast<s16x8*>(w_s)[i] =
+    for (int i = threadIdx.x; i < 4 * h * h / 8; i += blockDim.x) {
+      const int r = (i * 8) / (4 * h);
+      const int c = (i * 8) % (4 * h);
+      *reinterpret_cast<s16x8*>(w_s + (long)r * g_ld + c) =
           reinterpret_cast<const s16x8*>(wsrc)[i];
+    }
     wsrc = w_s;
+    w_ld = g_ld;
   }
   __syncthreads();
 
@@ -298,10 +312,10 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
           from_f32(dgo, &dgates_x[xbase + 3 * h + j]);
         }
         const int lr = lrow + r;
-        dg_s[lr * 4 * h + 0 * h + j] = __float2bfloat16(dgi);
-        dg_s[lr * 4 * h + 1 * h + j] = __float2bfloat16(dgf);
-        dg_s[lr * 4 * h + 2 * h + j] = __float2bfloat16(dgg);
-        dg_s[lr * 4 * h + 3 * h + j] = __float2bfloat16(dgo);
+        dg_s[lr * g_ld + 0 * h + j] = __float2bfloat16(dgi);
+        dg_s[lr * g_ld + 1 * h + j] = __float2bfloat16(dgf);
+        dg_s[lr * g_ld + 2 * h + j] = __float2bfloat16(dgg);
+        dg_s[lr * g_ld + 3 * h + j] = __float2bfloat16(dgo);
       }
     }
     __syncthreads();
@@ -310,13 +324,13 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
 #pragma unroll
     for (int q = 0; q < NQ_MAX; ++q) acc[q] = cfrag{0.f, 0.f, 0.f, 0.f};
     for (int kk = 0; kk < 4 * h / 32; ++kk) {
-      const bfrag adg = lds_frag_l(dg_s, 0, 4 * h, kk * 32);
+      const bfrag adg = lds_frag_l(dg_s, 0, g_ld, kk * 32);
 #pragma unroll
       for (int q = 0; q < NQ_MAX; ++q) {
         const int f = wid + 4 * q;
         if (q >= NQ || f >= N16) continue;
         acc[q] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            adg, lds_frag_l(wsrc, f * 16, 4 * h, kk * 32), acc[q], 0, 0, 0);
+            adg, lds_frag_l(wsrc, f * 16, w_ld, kk * 32), acc[q], 0, 0, 0);
       }
     }
 #pragma unroll
@@ -333,11 +347,12 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
 
 // ===================================================================== host
 static size_t lstm_fwd_smem(int h, bool w_lds) {
-  return ((size_t)16 * h + (size_t)16 * 4 * h + (w_lds ? (size_t)4 * h * h : 0))
-         * sizeof(bf16);
+  return ((size_t)16 * (h + 8) + (size_t)16 * 4 * h +
+          (w_lds ? (size_t)4 * h * (h + 8) : 0)) * sizeof(bf16);
 }
 static size_t lstm_bwd_smem(int h, bool w_lds) {
-  return ((size_t)16 * 4 * h + (w_lds ? (size_t)4 * h * h : 0)) * sizeof(bf16);
+  return ((size_t)16 * (4 * h + 8) + (w_lds ? (size_t)h * (4 * h + 8) : 0))
+         * sizeof(bf16);
 }
 
 // Bidirectional fused path: gates_x [B,L,8h] (fw|bw), w_hh_t2 [2,4h,h].
