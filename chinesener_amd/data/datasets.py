@@ -69,6 +69,11 @@ DATASETS: Dict[str, DatasetSpec] = {
     "msr": DatasetSpec("msr", {t: i for i, t in enumerate(
         [PAD_TAG, "B", "M", "E", "S", CLS_TAG, SEP_TAG])}, 150,
         [], 86918, 4000, 3985, scheme="bies"),
+    # virtual dataset: train split replaced by the augmentation dump
+    # (reference data/people_daily_augment/preprocess.py:8-20)
+    "people_daily_augment": DatasetSpec(
+        "people_daily_augment", _bio_tagset(["LOC", "PER", "ORG"]), 150,
+        ["LOC", "PER", "ORG"], 20865, 2318, 4636),
 }
 
 
@@ -147,6 +152,19 @@ def load_data(name: str, data_dir: str, split: str) -> Tuple[Sentences, Tags]:
     """Dispatch on dataset layout; falls back to synthetic when files absent."""
     spec = get_spec(name)
     try:
+        if name == "people_daily_augment":
+            # train comes from train_augment.pkl (augment.py dump);
+            # other splits fall through to the base corpus
+            base_dir = data_dir.replace("_augment", "")
+            if split == "train":
+                import pickle
+                with open(os.path.join(data_dir, "train_augment.pkl"),
+                          "rb") as f:
+                    blob = pickle.load(f)
+                if isinstance(blob, dict):
+                    return blob["sentences"], blob["tags"]
+                return blob      # (sentences, tags) tuple form
+            return load_conll(base_dir, f"example.{split}")
         if name == "msra":
             return load_sentence_tag_dirs(data_dir, split)
         if name == "cluener":

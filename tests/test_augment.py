@@ -94,3 +94,21 @@ def test_mlm_head_shapes():
     # tied embedding: grads flow to the word embedding through the head
     logits.sum().backward()
     assert bert.embeddings.word.weight.grad is not None
+
+
+def test_people_daily_augment_dataset(tmp_path):
+    """Virtual dataset: train split comes from train_augment.pkl
+    (reference data/people_daily_augment/preprocess.py:8-20)."""
+    from chinesener_amd.data.datasets import load_data
+    d = build_entity_dict(["people_daily"], str(tmp_path))
+    aug_dir = tmp_path / "people_daily_augment"
+    path = augment("people_daily", str(tmp_path), entity_dict=d,
+                   methods=["entity_replace"])
+    # move the dump into the virtual dataset dir
+    aug_dir.mkdir(exist_ok=True)
+    os.replace(path, aug_dir / "train_augment.pkl")
+    s, t = load_data("people_daily_augment", str(aug_dir), "train")
+    assert len(s) > 0 and len(s) == len(t)
+    # valid split falls back (synthetic here)
+    s2, t2 = load_data("people_daily_augment", str(aug_dir), "valid")
+    assert len(s2) > 0
