@@ -70,6 +70,13 @@ class AdamWeightDecay(torch.optim.Optimizer):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay, lr_scale=1.0)
         super().__init__(params, defaults)
+        # device-side base LR (hipGraph-captured steps: the schedule
+        # writes this scalar outside the graph, the fused kernel reads it)
+        self.lr_dev: Optional[torch.Tensor] = None
+
+    def enable_graph_lr(self, device) -> torch.Tensor:
+        self.lr_dev = torch.ones(1, dtype=torch.float32, device=device)
+        return self.lr_dev
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -99,7 +106,11 @@ class AdamWeightDecay(torch.optim.Optimizer):
                         s["master"] = p.detach().float().clone()
             ms = [s["m"] for s in states]
             vs = [s["v"] for s in states]
-            lr = group["lr"] * group.get("lr_scale", 1.0)
+            if self.lr_dev is not None:
+                # graph mode: meta.lr = per-group scale; base from lr_dev
+                lr = group.get("lr_scale", 1.0)
+            else:
+                lr = group["lr"] * group.get("lr_scale", 1.0)
             b1, b2 = group["betas"]
             eps, wd = group["eps"], group["weight_decay"]
             if params[0].is_cuda and ops.ext_available():
@@ -139,7 +150,7 @@ class AdamWeightDecay(torch.optim.Optimizer):
             cat = [sum((f[i] for f in bins), []) for i in range(7)]
             ops.get_ext().multi_tensor_adamw(cat[0], cat[1], cat[2], cat[3],
                                              cat[4], cat[5], cat[6],
-                                             b1, b2, eps)
+                                             b1, b2, eps, self.lr_dev)
         return loss
 
 
@@ -178,6 +189,9 @@ class LrSchedule:
         lr = self.lr_at(step)
         for g in optimizer.param_groups:
             g["lr"] = lr
+        lr_dev = getattr(optimizer, "lr_dev", None)
+        if lr_dev is not None:   # graph mode: one tiny device write
+            lr_dev.fill_(lr)
         return lr
 
 

@@ -28,21 +28,28 @@ struct ChunkMeta {
 template <typename T>
 __device__ __forceinline__ void adamw_one(const ChunkMeta& mt, long i,
                                           float b1, float b2, float eps,
-                                          T* p, const T* g) {
+                                          T* p, const T* g, float mul) {
   const float gv = to_f32(g[i]);
   const float m = b1 * mt.m[i] + (1.f - b1) * gv;
   const float v = b2 * mt.v[i] + (1.f - b2) * gv * gv;
   mt.m[i] = m;
   mt.v[i] = v;
   const float pv = mt.master ? mt.master[i] : to_f32(p[i]);
-  const float upd = pv - mt.lr * (m * __frcp_rn(sqrtf(v) + eps) + mt.wd * pv);
+  const float upd =
+      pv - mul * mt.lr * (m * __frcp_rn(sqrtf(v) + eps) + mt.wd * pv);
   if (mt.master) mt.master[i] = upd;
   from_f32(upd, &p[i]);
 }
 
+// lr_mul: optional device scalar multiplied into every tensor's lr —
+// lets a hipGraph-captured step keep a live LR schedule (the schedule
+// writes the device scalar outside the graph; meta.lr holds the
+// per-group scale).
 template <typename T>
 __global__ void multi_tensor_adamw_kernel(ChunkMeta* metas, int n_tensors,
-                                          float b1, float b2, float eps) {
+                                          float b1, float b2, float eps,
+                                          const float* lr_mul) {
+  const float mul = lr_mul ? *lr_mul : 1.f;
   // explicit 4-wide vector I/O on every stream (g/p via 4x16b or f32x4,
   // m/v/master via f32x4) — scalar per-lane math in registers
   const long stride = (long)gridDim.x * blockDim.x * 4;
@@ -85,8 +92,8 @@ __global__ void multi_tensor_adamw_kernel(ChunkMeta* metas, int n_tensors,
         const float v = b2 * v4[e] + (1.f - b2) * gv[e] * gv[e];
         m4[e] = m;
         v4[e] = v;
-        up[e] = pv[e] -
-                mt.lr * (m * __frcp_rn(sqrtf(v) + eps) + mt.wd * pv[e]);
+        up[e] = pv[e] - mul * mt.lr *
+                    (m * __frcp_rn(sqrtf(v) + eps) + mt.wd * pv[e]);
       }
       *reinterpret_cast<f32x4*>(mt.m + i) = m4;
       *reinterpret_cast<f32x4*>(mt.v + i) = v4;
@@ -101,13 +108,13 @@ __global__ void multi_tensor_adamw_kernel(ChunkMeta* metas, int n_tensors,
         *reinterpret_cast<f32x4*>(reinterpret_cast<float*>(p) + i) = up;
       }
     }
-    for (; i < mt.n; ++i) adamw_one<T>(mt, i, b1, b2, eps, p, g);
+    for (; i < mt.n; ++i) adamw_one<T>(mt, i, b1, b2, eps, p, g, mul);
   }
 }
 
 static void launch_adamw(std::vector<ChunkMeta>& metas, long total,
                          bool is_bf16, const at::Tensor& ref, double b1,
-                         double b2, double eps) {
+                         double b2, double eps, const float* lr_mul) {
   const int n = metas.size();
   auto meta_blob = at::from_blob(metas.data(), {(long)(n * sizeof(ChunkMeta))},
                                  at::TensorOptions().dtype(at::kByte))
@@ -119,11 +126,11 @@ static void launch_adamw(std::vector<ChunkMeta>& metas, long total,
   if (is_bf16) {
     hipLaunchKernelGGL(multi_tensor_adamw_kernel<bf16>, grid, dim3(block), 0,
                        stream, (ChunkMeta*)meta_blob.data_ptr(), n, (float)b1,
-                       (float)b2, (float)eps);
+                       (float)b2, (float)eps, lr_mul);
   } else {
     hipLaunchKernelGGL(multi_tensor_adamw_kernel<float>, grid, dim3(block), 0,
                        stream, (ChunkMeta*)meta_blob.data_ptr(), n, (float)b1,
-                       (float)b2, (float)eps);
+                       (float)b2, (float)eps, lr_mul);
   }
   HIP_CHECK_LAST();
 }
@@ -134,7 +141,8 @@ void multi_tensor_adamw(std::vector<at::Tensor> params,
                         std::vector<at::Tensor> masters,
                         std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
                         std::vector<double> lrs, std::vector<double> wds,
-                        double b1, double b2, double eps) {
+                        double b1, double b2, double eps,
+                        c10::optional<at::Tensor> lr_mul) {
   const int n = params.size();
   TORCH_CHECK(n > 0 && n <= MAX_TENSORS, "multi_tensor_adamw: bad tensor count");
   const bool has_master = !masters.empty();
@@ -156,7 +164,13 @@ void multi_tensor_adamw(std::vector<at::Tensor> params,
                 params[i].numel(), (float)lrs[i], (float)wds[i]};
     total += metas[i].n;
   }
-  launch_adamw(metas, total, is_bf16, params[0], b1, b2, eps);
+  const float* mul = nullptr;
+  if (lr_mul.has_value()) {
+    TORCH_CHECK(lr_mul->scalar_type() == at::kFloat && lr_mul->numel() == 1,
+                "lr_mul must be a 1-element fp32 device tensor");
+    mul = lr_mul->data_ptr<float>();
+  }
+  launch_adamw(metas, total, is_bf16, params[0], b1, b2, eps, mul);
 }
 
 // ------------------------------------------------------- global L2 norm

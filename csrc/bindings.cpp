@@ -35,7 +35,8 @@ std::vector<at::Tensor> softlexicon_bwd(const at::Tensor&, const at::Tensor&,
 void multi_tensor_adamw(std::vector<at::Tensor>, std::vector<at::Tensor>,
                         std::vector<at::Tensor>, std::vector<at::Tensor>,
                         std::vector<at::Tensor>, std::vector<double>,
-                        std::vector<double>, double, double, double);
+                        std::vector<double>, double, double, double,
+                        c10::optional<at::Tensor>);
 at::Tensor multi_tensor_sumsq(std::vector<at::Tensor>);
 void multi_tensor_scale(std::vector<at::Tensor>, const at::Tensor&);
 // attention.hip

@@ -288,7 +288,7 @@ def test_multi_tensor_adamw_matches_cpu_math():
     lr, b1, b2, eps, wd = 1e-3, 0.9, 0.999, 1e-6, 0.01
     ops.get_ext().multi_tensor_adamw(params, grads, [], ms, vs,
                                      [lr] * len(params), [wd] * len(params),
-                                     b1, b2, eps)
+                                     b1, b2, eps, None)
     for p, g, m, v in zip(ref_p, grads, ref_m, ref_v):
         m.mul_(b1).add_(g, alpha=1 - b1)
         v.mul_(b2).addcmul_(g, g, value=1 - b2)
@@ -312,7 +312,8 @@ def test_multi_tensor_adamw_bf16_master():
     lr, b1, b2, eps, wd = 1e-3, 0.9, 0.999, 1e-6, 0.01
     for _ in range(3):
         ops.get_ext().multi_tensor_adamw(params, grads, masters, ms, vs,
-                                         [lr] * 2, [wd] * 2, b1, b2, eps)
+                                         [lr] * 2, [wd] * 2, b1, b2, eps,
+                                         None)
     rm = [torch.zeros_like(m) for m in ref_master]
     rv = [torch.zeros_like(m) for m in ref_master]
     for _ in range(3):
