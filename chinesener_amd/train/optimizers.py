@@ -76,6 +76,7 @@ class AdamWeightDecay(torch.optim.Optimizer):
 
     def enable_graph_lr(self, device) -> torch.Tensor:
         self.lr_dev = torch.ones(1, dtype=torch.float32, device=device)
+        self._meta_cache = {}
         return self.lr_dev
 
     @torch.no_grad()
@@ -148,9 +149,27 @@ class AdamWeightDecay(torch.optim.Optimizer):
             b1, b2 = self.param_groups[0]["betas"]
             eps = self.param_groups[0]["eps"]
             cat = [sum((f[i] for f in bins), []) for i in range(7)]
-            ops.get_ext().multi_tensor_adamw(cat[0], cat[1], cat[2], cat[3],
-                                             cat[4], cat[5], cat[6],
-                                             b1, b2, eps, self.lr_dev)
+            ext = ops.get_ext()
+            if self.lr_dev is None:
+                ext.multi_tensor_adamw(cat[0], cat[1], cat[2], cat[3],
+                                       cat[4], cat[5], cat[6],
+                                       b1, b2, eps, None)
+            else:
+                # graph mode: cache the device meta blob (no H2D per step
+                # — hipGraph capture forbids host transfers); meta lrs are
+                # the static per-group scales, lr_dev carries the schedule
+                sig = tuple(t.data_ptr() for t in cat[0] + cat[1])
+                cache = self._meta_cache.get(dtype)
+                if cache is None or cache[0] != sig:
+                    blob, info = ext.adamw_build_meta(
+                        cat[0], cat[1], cat[2], cat[3], cat[4], cat[5],
+                        cat[6])
+                    total, n, isbf = info.tolist()
+                    cache = (sig, blob, total, n, bool(isbf))
+                    self._meta_cache[dtype] = cache
+                ext.multi_tensor_adamw_run(cache[1], cache[2], cache[3],
+                                           cache[4], b1, b2, eps,
+                                           self.lr_dev)
         return loss
 
 
@@ -212,10 +231,19 @@ def clip_gradients(model: torch.nn.Module, family: str,
         grads = [p.grad for p in ps]
         if grads[0].is_cuda and ops.ext_available():
             ext = ops.get_ext()
-            sumsq = ext.multi_tensor_sumsq(grads)
+            # cached device meta blob: capture-safe (no H2D in the step);
+            # pointer signature invalidates when grads are reallocated
+            sig = tuple(g.data_ptr() for g in grads)
+            cm = getattr(model, "_clip_meta", None)
+            if cm is None or cm[0] != sig:
+                blob, info = ext.norm_build_meta(grads)
+                cm = (sig, blob, int(info[0]))
+                model._clip_meta = cm
+            out = torch.zeros(1, dtype=torch.float32, device=grads[0].device)
+            sumsq = ext.multi_tensor_sumsq_run(cm[1], cm[2], out)
             norm = sumsq.sqrt()
             coef = (max_norm / (norm + 1e-6)).clamp(max=1.0)
-            ext.multi_tensor_scale(grads, coef)
+            ext.multi_tensor_scale_run(cm[1], cm[2], coef)
             return sumsq
         return torch.nn.utils.clip_grad_norm_(model.parameters(), max_norm)
     torch.nn.utils.clip_grad_value_(model.parameters(), 5.0)

@@ -135,6 +135,57 @@ static void launch_adamw(std::vector<ChunkMeta>& metas, long total,
   HIP_CHECK_LAST();
 }
 
+// Build the device-side ChunkMeta blob once (H2D copy here) so a
+// hipGraph-captured step can run the kernel without any host transfer.
+// Returns {blob, [total, n, is_bf16]}.
+std::vector<at::Tensor> adamw_build_meta(
+    std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+    std::vector<at::Tensor> masters, std::vector<at::Tensor> ms,
+    std::vector<at::Tensor> vs, std::vector<double> lrs,
+    std::vector<double> wds) {
+  const int n = params.size();
+  TORCH_CHECK(n > 0 && n <= MAX_TENSORS, "adamw_build_meta: bad count");
+  const bool has_master = !masters.empty();
+  const bool is_bf16 = params[0].scalar_type() == at::kBFloat16;
+  std::vector<ChunkMeta> metas(n);
+  long total = 0;
+  for (int i = 0; i < n; ++i) {
+    metas[i] = {params[i].data_ptr(), grads[i].data_ptr(),
+                has_master ? masters[i].data_ptr<float>() : nullptr,
+                ms[i].data_ptr<float>(), vs[i].data_ptr<float>(),
+                params[i].numel(), (float)lrs[i], (float)wds[i]};
+    total += metas[i].n;
+  }
+  auto blob = at::from_blob(metas.data(), {(long)(n * sizeof(ChunkMeta))},
+                            at::TensorOptions().dtype(at::kByte))
+                  .to(params[0].device(), false);
+  auto info = at::tensor({total, (long)n, (long)(is_bf16 ? 1 : 0)},
+                         at::TensorOptions().dtype(at::kLong));
+  return {blob, info};
+}
+
+// Capture-safe launch from a prebuilt blob (no host transfers).
+void multi_tensor_adamw_run(const at::Tensor& blob, long total, long n,
+                            bool is_bf16, double b1, double b2, double eps,
+                            c10::optional<at::Tensor> lr_mul) {
+  const int block = 256;
+  const int gx = std::min<long>((total / n + block - 1) / block, 512);
+  dim3 grid(std::max(gx, 1), std::min<long>(n, 64));
+  auto stream = cur_stream(blob);
+  const float* mul = nullptr;
+  if (lr_mul.has_value()) mul = lr_mul->data_ptr<float>();
+  if (is_bf16) {
+    hipLaunchKernelGGL(multi_tensor_adamw_kernel<bf16>, grid, dim3(block), 0,
+                       stream, (ChunkMeta*)blob.data_ptr(), (int)n, (float)b1,
+                       (float)b2, (float)eps, mul);
+  } else {
+    hipLaunchKernelGGL(multi_tensor_adamw_kernel<float>, grid, dim3(block), 0,
+                       stream, (ChunkMeta*)blob.data_ptr(), (int)n, (float)b1,
+                       (float)b2, (float)eps, mul);
+  }
+  HIP_CHECK_LAST();
+}
+
 // fp32 path (masters empty) or bf16 path (masters[i] fp32, same numel).
 void multi_tensor_adamw(std::vector<at::Tensor> params,
                         std::vector<at::Tensor> grads,
@@ -229,6 +280,30 @@ __global__ void multi_tensor_l2_kernel(NormMeta* metas, int n_tensors,
   }
 }
 
+std::vector<at::Tensor> norm_build_meta(std::vector<at::Tensor> grads) {
+  const int n = grads.size();
+  TORCH_CHECK(n > 0 && n <= MAX_TENSORS, "norm_build_meta: bad count");
+  std::vector<NormMeta> metas(n);
+  for (int i = 0; i < n; ++i)
+    metas[i] = {grads[i].data_ptr(), grads[i].numel(),
+                grads[i].scalar_type() == at::kBFloat16 ? 1 : 0};
+  auto blob = at::from_blob(metas.data(), {(long)(n * sizeof(NormMeta))},
+                            at::TensorOptions().dtype(at::kByte))
+                  .to(grads[0].device(), false);
+  auto info = at::tensor({(long)n}, at::TensorOptions().dtype(at::kLong));
+  return {blob, info};
+}
+
+at::Tensor multi_tensor_sumsq_run(const at::Tensor& blob, long n,
+                                  const at::Tensor& out) {
+  hipLaunchKernelGGL(multi_tensor_l2_kernel, dim3(512, 1), dim3(256), 0,
+                     cur_stream(blob), (NormMeta*)blob.data_ptr(), (int)n,
+                     out.data_ptr<float>());
+  HIP_CHECK_LAST();
+  return out;
+}
+
+
 // Returns a 1-element fp32 tensor holding sum of squares (caller sqrts).
 at::Tensor multi_tensor_sumsq(std::vector<at::Tensor> grads) {
   const int n = grads.size();
@@ -315,6 +390,15 @@ void multi_tensor_scale(std::vector<at::Tensor> grads,
   dim3 grid(std::max(gx, 1), std::min(n, 64));
   hipLaunchKernelGGL(multi_tensor_scale_kernel, grid, dim3(block), 0,
                      cur_stream(grads[0]), (ScaleMeta*)meta_blob.data_ptr(), n,
+                     scale.data_ptr<float>());
+  HIP_CHECK_LAST();
+}
+
+void multi_tensor_scale_run(const at::Tensor& blob, long n,
+                            const at::Tensor& scale) {
+  // NormMeta and ScaleMeta share layout {ptr, n, is_bf16}
+  hipLaunchKernelGGL(multi_tensor_scale_kernel, dim3(256, 64), dim3(256), 0,
+                     cur_stream(blob), (ScaleMeta*)blob.data_ptr(), (int)n,
                      scale.data_ptr<float>());
   HIP_CHECK_LAST();
 }

@@ -39,6 +39,15 @@ void multi_tensor_adamw(std::vector<at::Tensor>, std::vector<at::Tensor>,
                         c10::optional<at::Tensor>);
 at::Tensor multi_tensor_sumsq(std::vector<at::Tensor>);
 void multi_tensor_scale(std::vector<at::Tensor>, const at::Tensor&);
+std::vector<at::Tensor> adamw_build_meta(
+    std::vector<at::Tensor>, std::vector<at::Tensor>, std::vector<at::Tensor>,
+    std::vector<at::Tensor>, std::vector<at::Tensor>, std::vector<double>,
+    std::vector<double>);
+void multi_tensor_adamw_run(const at::Tensor&, long, long, bool, double,
+                            double, double, c10::optional<at::Tensor>);
+std::vector<at::Tensor> norm_build_meta(std::vector<at::Tensor>);
+at::Tensor multi_tensor_sumsq_run(const at::Tensor&, long, const at::Tensor&);
+void multi_tensor_scale_run(const at::Tensor&, long, const at::Tensor&);
 // attention.hip
 std::vector<at::Tensor> attn_fwd(const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&, double);
@@ -92,6 +101,11 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("multi_tensor_adamw", &multi_tensor_adamw);
   m.def("multi_tensor_sumsq", &multi_tensor_sumsq);
   m.def("multi_tensor_scale", &multi_tensor_scale);
+  m.def("adamw_build_meta", &adamw_build_meta);
+  m.def("multi_tensor_adamw_run", &multi_tensor_adamw_run);
+  m.def("norm_build_meta", &norm_build_meta);
+  m.def("multi_tensor_sumsq_run", &multi_tensor_sumsq_run);
+  m.def("multi_tensor_scale_run", &multi_tensor_scale_run);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("attn_fwd_qkv", &attn_fwd_qkv);
