@@ -129,63 +129,25 @@ def main():
                   f"tunable enabled={_tun.is_enabled()} "
                   f"results={len(_tun.get_results())}", file=sys.stderr)
 
-    # hipGraph-capture the whole training step (forward, backward, DP
-    # all-reduce waits, clip, fused optimizer): removes the ~700
-    # launch gaps/step. The LR schedule stays live through the
-    # device-side lr multiplier (optimizer.enable_graph_lr). Falls back
-    # to eager if capture fails (e.g. RCCL build without graph support).
-    graph_step = None
-    if use_gpu and os.environ.get("CHINESENER_NO_STEPGRAPH") != "1" \
-            and hasattr(opt, "enable_graph_lr"):
-        try:
-            opt.enable_graph_lr(device)
-            schedule.apply(opt, max(step_num, 1))
-            static = {k: v.clone() for k, v in batches[0].items()}
-
-            def body():
-                opt.zero_grad(set_to_none=False)
-                if dp is not None:
-                    for b_ in dp.buckets:
-                        b_.flat.zero_()
-                        b_.ready = 0
-                out = model(cast(static))
-                out.loss.backward()
-                if dp is not None:
-                    dp.finalize_backward()
-                clip_gradients(model, family)
-                opt.step()
-                return out.loss
-
-            side = torch.cuda.Stream()
-            side.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(side):
-                for _ in range(2):
-                    body()
-            torch.cuda.current_stream().wait_stream(side)
-            graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
-                static_loss = body()
-
-            def graph_step_fn(batch):
-                nonlocal step_num
-                for k, v in batch.items():
-                    static[k].copy_(v, non_blocking=True)
-                step_num += 1
-                schedule.apply(opt, step_num)   # device lr write
-                graph.replay()
-                return static_loss
-
-            graph_step = graph_step_fn
+    # hipGraph-capture the whole training step (see train/graph_step.py)
+    graphed = None
+    if use_gpu and os.environ.get("CHINESENER_NO_STEPGRAPH") != "1":
+        from chinesener_amd.train.graph_step import GraphedTrainStep
+        from chinesener_amd.train.optimizers import clip_gradients as _clip
+        g = GraphedTrainStep(model, opt, schedule,
+                             lambda m: _clip(m, family), dp=dp, cast=cast)
+        if g.try_capture(batches[0]):
+            graphed = g
             if rank == 0 and os.environ.get("CHINESENER_BENCH_DEBUG"):
                 print("[debug] step captured in hipGraph", file=sys.stderr)
-        except Exception as e:
-            opt.lr_dev = None
-            graph_step = None
-            if rank == 0:
-                print(f"[bench] hipGraph step capture failed ({e}); "
-                      f"running eager", file=sys.stderr)
-    if graph_step is not None:
-        train_step = graph_step
+
+    def graph_train_step(batch):
+        nonlocal step_num
+        step_num += 1
+        return graphed.replay(batch, step_num)
+
+    if graphed is not None:
+        train_step = graph_train_step
     if dist is not None:
         dist.barrier()
     if use_gpu:

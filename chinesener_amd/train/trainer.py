@@ -16,6 +16,7 @@ import torch
 from ..models import ModelOutput, NerModel, optimizer_family
 from .checkpoints import CheckpointManager
 from .metrics import TagMetrics
+from .graph_step import GraphedTrainStep
 from .optimizers import build_optimizer, clip_gradients
 from .precision import convert_bf16_mixed, wants_pure_bf16
 from ..utils.profiling import roctx_range
@@ -49,6 +50,13 @@ class Trainer:
         self.use_bf16 = False  # autocast replaced by pure-bf16 weights
         self.step = self.ckpt.restore(model, self.optimizer,
                                       map_location=self.device)
+        # hipGraph-captured step: engaged lazily once batch shapes repeat
+        # (CHINESENER_NO_STEPGRAPH=1 disables)
+        self._graph: GraphedTrainStep | None = None
+        self._last_sig = None
+        self.use_step_graph = (self.device.startswith("cuda")
+                               and os.environ.get("CHINESENER_NO_STEPGRAPH")
+                               != "1")
 
     # ------------------------------------------------------------- train
     def _cast(self, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
@@ -64,8 +72,26 @@ class Trainer:
         return self.model(batch)
 
     def train_step(self, batch: Dict[str, torch.Tensor]) -> float:
-        batch = self._cast({k: v.to(self.device, non_blocking=True)
-                            for k, v in batch.items()})
+        batch = {k: v.to(self.device, non_blocking=True)
+                 for k, v in batch.items()}
+        if self.use_step_graph:
+            if self._graph is not None and self._graph.matches(batch):
+                self.step += 1
+                loss = self._graph.replay(batch, self.step)
+                return float(loss.detach())
+            sig = tuple(sorted((k, tuple(v.shape)) for k, v in batch.items()))
+            if self._graph is None and sig == self._last_sig:
+                g = GraphedTrainStep(
+                    self.model, self.optimizer, self.schedule,
+                    lambda m: clip_gradients(m, self.family),
+                    dp=self.dp, cast=self._cast)
+                if g.try_capture(batch):
+                    self._graph = g
+                    self.step += 1
+                    return float(g.replay(batch, self.step).detach())
+                self.use_step_graph = False
+            self._last_sig = sig
+        batch = self._cast(batch)
         if self.dp is not None:
             self.dp.zero_grad()
         else:
