@@ -132,14 +132,14 @@ def main():
     # hipGraph-capture the whole training step (see train/graph_step.py)
     graphed = None
     if use_gpu and os.environ.get("CHINESENER_NO_STEPGRAPH") != "1":
+        import logging
+        logging.basicConfig(level=logging.INFO, stream=sys.stderr)
         from chinesener_amd.train.graph_step import GraphedTrainStep
         from chinesener_amd.train.optimizers import clip_gradients as _clip
         g = GraphedTrainStep(model, opt, schedule,
                              lambda m: _clip(m, family), dp=dp, cast=cast)
         if g.try_capture(batches[0]):
             graphed = g
-            if rank == 0 and os.environ.get("CHINESENER_BENCH_DEBUG"):
-                print("[debug] step captured in hipGraph", file=sys.stderr)
 
     def graph_train_step(batch):
         nonlocal step_num
