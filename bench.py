@@ -138,7 +138,18 @@ def main():
         from chinesener_amd.train.optimizers import clip_gradients as _clip
         g = GraphedTrainStep(model, opt, schedule,
                              lambda m: _clip(m, family), dp=dp, cast=cast)
-        if g.try_capture(batches[0]):
+        ok = g.try_capture(batches[0])
+        if dist is not None:
+            # every rank must agree graph-vs-eager or the collective
+            # schedules diverge and the job deadlocks
+            flag = torch.tensor([1 if ok else 0], device=device)
+            dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+            if not bool(flag.item()):
+                if ok:
+                    g.graph = None
+                    opt.lr_dev = None
+                ok = False
+        if ok:
             graphed = g
 
     def graph_train_step(batch):
