@@ -73,6 +73,7 @@ class BertLayer(nn.Module):
         self.ln2_w = nn.Parameter(torch.ones(H))
         self.ln2_b = nn.Parameter(torch.zeros(H))
         self.dropout = nn.Dropout(cfg.hidden_dropout_prob)
+        self.p_drop = cfg.hidden_dropout_prob
         self.eps = cfg.layer_norm_eps
 
     def forward(self, x, mask, lens):
@@ -80,13 +81,13 @@ class BertLayer(nn.Module):
         qkv = ops.linear(x, self.qkv.weight, self.qkv.bias) \
             .reshape(B, L, 3, self.n_heads, self.head_dim)
         ctx = ops.attention_qkv(qkv, mask=mask, lens=lens).reshape(B, L, H)
-        a = self.dropout(ops.linear(ctx, self.attn_out.weight,
-                                    self.attn_out.bias))
-        x = ops.add_layernorm(a, x, self.ln1_w, self.ln1_b, self.eps)
+        a = ops.linear(ctx, self.attn_out.weight, self.attn_out.bias)
+        x = ops.dropout_add_layernorm(a, x, self.ln1_w, self.ln1_b, self.eps,
+                                      self.p_drop, self.training)
         f = ops.bias_gelu(self.ffn_in(x), self.ffn_in_bias)
-        f = self.dropout(ops.linear(f, self.ffn_out.weight,
-                                    self.ffn_out.bias))
-        return ops.add_layernorm(f, x, self.ln2_w, self.ln2_b, self.eps)
+        f = ops.linear(f, self.ffn_out.weight, self.ffn_out.bias)
+        return ops.dropout_add_layernorm(f, x, self.ln2_w, self.ln2_b,
+                                         self.eps, self.p_drop, self.training)
 
 
 class BertModel(nn.Module):

@@ -58,6 +58,57 @@ class _AddLayerNormFn(torch.autograd.Function):
         return dx, dx, dw, db, None
 
 
+_rng_counter = None
+
+
+def _rng_counter_for(device):
+    global _rng_counter
+    if _rng_counter is None or _rng_counter.device != device:
+        seed = torch.initial_seed() & 0x7FFFFFFF
+        _rng_counter = torch.tensor([seed], dtype=torch.int64, device=device)
+    return _rng_counter
+
+
+class _DropoutAddLNFn(torch.autograd.Function):
+    """Fused dropout(x) + residual + LayerNorm (BERT post-LN); counter-
+    based RNG bumps a device scalar so hipGraph replays draw fresh
+    masks."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, bias, eps, keep):
+        y, s, mask, mean, rstd = get_ext().dropout_add_ln_fwd(
+            x, res, weight, bias, eps, keep,
+            _rng_counter_for(x.device))
+        ctx.save_for_backward(s, weight, mask, mean, rstd)
+        ctx.keep = keep
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        s, weight, mask, mean, rstd = ctx.saved_tensors
+        dsum, dw, db = get_ext().layernorm_bwd(dy.contiguous(), s, weight,
+                                               mean, rstd)
+        dx = get_ext().mask_scale(dsum, mask, 1.0 / ctx.keep)
+        return dx, dsum, dw, db, None, None
+
+
+def dropout_add_layernorm(x, residual, weight, bias, eps: float = 1e-12,
+                          p: float = 0.1, training: bool = True):
+    """LayerNorm(dropout(x) + residual) in one kernel (plus the shared LN
+    backward) — replaces the separate dropout fwd + mask-mul bwd per
+    encoder sublayer."""
+    if training and p > 0 and hip_enabled(x) and x.dtype == torch.bfloat16:
+        shape = x.shape
+        y = _DropoutAddLNFn.apply(
+            x.reshape(-1, shape[-1]).contiguous(),
+            residual.to(x.dtype).reshape(-1, shape[-1]).contiguous(),
+            weight, bias, eps, 1.0 - p)
+        return y.reshape(shape)
+    if training and p > 0:
+        x = F.dropout(x, p, training)
+    return add_layernorm(x, residual, weight, bias, eps)
+
+
 def add_layernorm(x, residual, weight, bias, eps: float = 1e-12):
     """LayerNorm(x + residual) — the BERT post-LN residual pattern."""
     if hip_enabled(x):

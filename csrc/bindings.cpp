@@ -21,6 +21,12 @@ at::Tensor masked_ce_bwd(const at::Tensor&, const at::Tensor&,
                          const at::Tensor&, const at::Tensor&,
                          const at::Tensor&);
 at::Tensor colsum(const at::Tensor&);
+std::vector<at::Tensor> dropout_add_ln_fwd(const at::Tensor&,
+                                           const at::Tensor&,
+                                           const at::Tensor&,
+                                           const at::Tensor&, double, double,
+                                           const at::Tensor&);
+at::Tensor mask_scale(const at::Tensor&, const at::Tensor&, double);
 // crf.hip
 std::vector<at::Tensor> crf_fwd(const at::Tensor&, const at::Tensor&,
                                 const at::Tensor&, const at::Tensor&);
@@ -94,6 +100,8 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("masked_ce_fwd", &masked_ce_fwd);
   m.def("masked_ce_bwd", &masked_ce_bwd);
   m.def("colsum", &colsum);
+  m.def("dropout_add_ln_fwd", &dropout_add_ln_fwd);
+  m.def("mask_scale", &mask_scale);
   m.def("crf_fwd", &crf_fwd);
   m.def("crf_viterbi", &crf_viterbi);
   m.def("softlexicon_fwd", &softlexicon_fwd);

@@ -606,3 +606,162 @@ at::Tensor colsum(const at::Tensor& dy) {
   HIP_CHECK_LAST();
   return out;
 }
+
+// ---------------------------------------------------------------------
+// fused dropout + residual-add + LayerNorm (the BERT post-LN pattern:
+// LN(dropout(sublayer_out) + x)). Counter-based RNG (splitmix64 hash of
+// (step_counter, element)) — the device counter is bumped by a 1-thread
+// kernel after each call, so hipGraph replays draw fresh masks.
+// ---------------------------------------------------------------------
+__device__ __forceinline__ float hash_uniform(unsigned long long ctr,
+                                              unsigned long long idx) {
+  unsigned long long z = ctr * 0x9E3779B97F4A7C15ull ^ idx;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  z = z ^ (z >> 31);
+  return (float)(z >> 40) * (1.f / 16777216.f);  // top 24 bits -> [0,1)
+}
+
+__global__ void dropout_add_ln_fwd_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ res,
+    const float* __restrict__ w, const float* __restrict__ b,
+    bf16* __restrict__ y, bf16* __restrict__ sum_out,
+    unsigned char* __restrict__ mask_out, float* __restrict__ mean,
+    float* __restrict__ rstd, const unsigned long long* __restrict__ ctr,
+    long N, int H, float eps, float keep) {
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long row = (long)blockIdx.x * (blockDim.x / WAVE) + wid;
+  if (row >= N) return;
+  const unsigned long long c = *ctr;
+  const float inv_keep = 1.f / keep;
+  const bf16* xr = x + row * H;
+  const bf16* rr = res + row * H;
+  float acc = 0.f, acc2 = 0.f;
+  const int HV = H / 8;
+  constexpr int MAXIT = 6;  // H <= 3072
+  float vals[MAXIT][8];
+  unsigned char mks[MAXIT][8];
+#pragma unroll
+  for (int it = 0; it < MAXIT; ++it) {
+    const int i = lane + it * WAVE;
+    if (i < HV) {
+      const s16x8 rx = reinterpret_cast<const s16x8*>(xr)[i];
+      const s16x8 rres = reinterpret_cast<const s16x8*>(rr)[i];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const long idx = row * H + i * 8 + e;
+        const unsigned char m = hash_uniform(c, (unsigned long long)idx) < keep;
+        mks[it][e] = m;
+        const float xv =
+            to_f32(reinterpret_cast<const bf16*>(&rx)[e]) * m * inv_keep +
+            to_f32(reinterpret_cast<const bf16*>(&rres)[e]);
+        vals[it][e] = xv;
+        acc += xv;
+        acc2 += xv * xv;
+      }
+    }
+  }
+  acc = wave_reduce_sum(acc);
+  acc2 = wave_reduce_sum(acc2);
+  const float mu = acc / H;
+  const float var = fmaxf(acc2 / H - mu * mu, 0.f);
+  const float rs = rsqrtf(var + eps);
+  if (lane == 0) {
+    mean[row] = mu;
+    rstd[row] = rs;
+  }
+  bf16* yr = y + row * H;
+  bf16* sr = sum_out + row * H;
+  unsigned char* mr = mask_out + row * H;
+#pragma unroll
+  for (int it = 0; it < MAXIT; ++it) {
+    const int i = lane + it * WAVE;
+    if (i < HV) {
+      bf16 ov[8], sv[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float s = vals[it][e];
+        sv[e] = __float2bfloat16(s);
+        ov[e] = __float2bfloat16((s - mu) * rs * w[i * 8 + e] + b[i * 8 + e]);
+        mr[i * 8 + e] = mks[it][e];
+      }
+      reinterpret_cast<s16x8*>(yr)[i] = *reinterpret_cast<s16x8*>(ov);
+      reinterpret_cast<s16x8*>(sr)[i] = *reinterpret_cast<s16x8*>(sv);
+    }
+  }
+}
+
+__global__ void bump_counter_kernel(unsigned long long* ctr) { ++(*ctr); }
+
+// dx_drop = dsum * mask / keep (the dropout half of the fused bwd; the
+// LN half reuses layernorm_bwd on the stored sum, giving dsum = dres)
+__global__ void mask_scale_kernel(const bf16* __restrict__ dsum,
+                                  const unsigned char* __restrict__ mask,
+                                  bf16* __restrict__ dx, long n, float scale) {
+  const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (i0 + 7 >= n) {
+    for (long i = i0; i < n; ++i)
+      dx[i] = __float2bfloat16(to_f32(dsum[i]) * mask[i] * scale);
+    return;
+  }
+  const s16x8 d = *reinterpret_cast<const s16x8*>(dsum + i0);
+  bf16 o[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e)
+    o[e] = __float2bfloat16(to_f32(reinterpret_cast<const bf16*>(&d)[e]) *
+                            mask[i0 + e] * scale);
+  *reinterpret_cast<s16x8*>(dx + i0) = *reinterpret_cast<const s16x8*>(o);
+}
+
+std::vector<at::Tensor> dropout_add_ln_fwd(const at::Tensor& x,
+                                           const at::Tensor& res,
+                                           const at::Tensor& w,
+                                           const at::Tensor& b, double eps,
+                                           double keep,
+                                           const at::Tensor& counter) {
+  CHECK_CUDA_CONTIG(x);
+  CHECK_CUDA_CONTIG(res);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "dropout_add_ln: bf16 only");
+  const long N = x.size(0);
+  const int H = x.size(1);
+  TORCH_CHECK(H % 8 == 0 && H <= 3072, "dropout_add_ln: bad H ", H);
+  auto y = at::empty_like(x);
+  auto sum = at::empty_like(x);
+  auto mask = at::empty({N, (long)H}, x.options().dtype(at::kByte));
+  auto mean = at::empty({N}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({N}, x.options().dtype(at::kFloat));
+  auto wf = w.to(at::kFloat).contiguous();
+  auto bf = b.to(at::kFloat).contiguous();
+  const int rows_per_blk = 4;
+  auto stream = cur_stream(x);
+  hipLaunchKernelGGL(dropout_add_ln_fwd_kernel,
+                     dim3((N + rows_per_blk - 1) / rows_per_blk),
+                     dim3(rows_per_blk * WAVE), 0, stream,
+                     (const bf16*)x.data_ptr(), (const bf16*)res.data_ptr(),
+                     wf.data_ptr<float>(), bf.data_ptr<float>(),
+                     (bf16*)y.data_ptr(), (bf16*)sum.data_ptr(),
+                     mask.data_ptr<unsigned char>(), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(),
+                     (const unsigned long long*)counter.data_ptr(), N, H,
+                     (float)eps, (float)keep);
+  hipLaunchKernelGGL(bump_counter_kernel, dim3(1), dim3(1), 0, stream,
+                     (unsigned long long*)counter.data_ptr());
+  HIP_CHECK_LAST();
+  return {y, sum, mask, mean, rstd};
+}
+
+at::Tensor mask_scale(const at::Tensor& dsum, const at::Tensor& mask,
+                      double scale) {
+  CHECK_CUDA_CONTIG(dsum);
+  auto dx = at::empty_like(dsum);
+  const long n = dsum.numel();
+  const int block = 256;
+  const long grid = std::min<long>((n / 8 + block - 1) / block, 32768);
+  hipLaunchKernelGGL(mask_scale_kernel, dim3(grid), dim3(block), 0,
+                     cur_stream(dsum), (const bf16*)dsum.data_ptr(),
+                     mask.data_ptr<unsigned char>(), (bf16*)dx.data_ptr(), n,
+                     (float)scale);
+  HIP_CHECK_LAST();
+  return dx;
+}
