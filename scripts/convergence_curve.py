@@ -41,9 +41,23 @@ def main():
     gen = batches()
     curve = []
     t0 = time.time()
-    for step in range(1, 1201):
+    for step in range(1, int(os.environ.get("CONV_STEPS", "1200")) + 1):
         loss = trainer.train_step(next(gen))
-        if step % 200 == 0:
+        if step % int(os.environ.get("CONV_EVAL_EVERY", "200")) == 0:
+            if os.environ.get("CONV_NO_EVAL") == "1":
+                print(f"step {step} loss {loss:.2f}", flush=True)
+                continue
+            if os.environ.get("CONV_EVAL_ONE") == "1":
+                with torch.no_grad():
+                    trainer.model.eval()
+                    b = next(pipe.iter_batches("valid", shuffle=False))
+                    dev = trainer._cast({k: v.to(trainer.device)
+                                         for k, v in b.items()})
+                    out = trainer.model(dev, compute_pred=True)
+                    trainer.model.train()
+                print(f"step {step} loss {loss:.2f} evalloss "
+                      f"{float(out.loss):.2f}", flush=True)
+                continue
             rows = trainer.predict(pipe.iter_batches("valid", shuffle=False))
             idx2tag = pipe.params["idx2tag"]
             proc = [process_prediction(r, idx2tag) for r in rows]
