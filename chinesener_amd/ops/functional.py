@@ -7,6 +7,7 @@ K10 LayerNorm, K11 bias-GELU, K13 masked CE, K14 dice).
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional, Tuple
 
 import torch
@@ -97,7 +98,8 @@ def dropout_add_layernorm(x, residual, weight, bias, eps: float = 1e-12,
     """LayerNorm(dropout(x) + residual) in one kernel (plus the shared LN
     backward) — replaces the separate dropout fwd + mask-mul bwd per
     encoder sublayer."""
-    if training and p > 0 and hip_enabled(x) and x.dtype == torch.bfloat16:
+    if (training and p > 0 and hip_enabled(x) and x.dtype == torch.bfloat16
+            and os.environ.get("CHINESENER_NO_FUSED_DROPOUT") != "1"):
         shape = x.shape
         y = _DropoutAddLNFn.apply(
             x.reshape(-1, shape[-1]).contiguous(),
