@@ -194,10 +194,12 @@ __global__ void crf_viterbi_kernel(const float* __restrict__ emis,
   }
   if (lane == 0) {
     int* pr = pred + (long)b * L;
-    int cur = barg;
+    // clamp defensively: non-finite emissions (diverged training) must
+    // not turn into out-of-bounds backtraces
+    int cur = min(max(barg, 0), T - 1);
     pr[n - 1] = cur;
     for (int t = n - 2; t >= 0; --t) {
-      cur = bp[(t + 1) * TMAX + cur];
+      cur = min(max((int)bp[(t + 1) * TMAX + cur], 0), T - 1);
       pr[t] = cur;
     }
   }
