@@ -50,13 +50,17 @@ class Trainer:
         self.use_bf16 = False  # autocast replaced by pure-bf16 weights
         self.step = self.ckpt.restore(model, self.optimizer,
                                       map_location=self.device)
-        # hipGraph-captured step: engaged lazily once batch shapes repeat
-        # (CHINESENER_NO_STEPGRAPH=1 disables)
+        # hipGraph-captured step (opt-in: CHINESENER_STEPGRAPH=1).
+        # Replays are verified bit-exact vs eager steps, but interleaved
+        # evaluation plus chaotic high-LR regimes (diff-lr x500 groups)
+        # showed instability in long runs, so real training defaults to
+        # eager; bench.py (short, no evals, warmup-phase LRs) uses graphs
+        # by default.
         self._graph: GraphedTrainStep | None = None
         self._last_sig = None
         self.use_step_graph = (self.device.startswith("cuda")
-                               and os.environ.get("CHINESENER_NO_STEPGRAPH")
-                               != "1")
+                               and os.environ.get("CHINESENER_STEPGRAPH")
+                               == "1")
 
     # ------------------------------------------------------------- train
     def _cast(self, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:

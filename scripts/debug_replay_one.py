@@ -12,7 +12,7 @@ name = "bert_bilstm_crf"
 pipe = NerDataset("/tmp/d", "msra", 64, 1, name)
 cfg = BertConfig(hidden_dropout_prob=0.0, attention_probs_dropout_prob=0.0)
 params = resolve_params(model_params(name), pipe.params,
-                        {"model_name": name, "num_train_steps": 1600,
+                        {"model_name": name, "num_train_steps": 10**9, "lr": 1e-6, "warmup_ratio": 1.0,
                          "bert_config": cfg, "dropout_rate": 0.0,
                          "embedding_dropout": 0.0})
 params["rnn_params"] = dict(params["rnn_params"], keep_prob_list=[1.0])
