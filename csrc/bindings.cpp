@@ -27,6 +27,7 @@ std::vector<at::Tensor> dropout_add_ln_fwd(const at::Tensor&,
                                            const at::Tensor&, double, double,
                                            const at::Tensor&);
 at::Tensor mask_scale(const at::Tensor&, const at::Tensor&, double);
+at::Tensor wgrad(const at::Tensor&, const at::Tensor&, long);
 // crf.hip
 std::vector<at::Tensor> crf_fwd(const at::Tensor&, const at::Tensor&,
                                 const at::Tensor&, const at::Tensor&);
@@ -102,6 +103,7 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("colsum", &colsum);
   m.def("dropout_add_ln_fwd", &dropout_add_ln_fwd);
   m.def("mask_scale", &mask_scale);
+  m.def("wgrad", &wgrad);
   m.def("crf_fwd", &crf_fwd);
   m.def("crf_viterbi", &crf_viterbi);
   m.def("softlexicon_fwd", &softlexicon_fwd);

@@ -17,7 +17,8 @@ OUT_DIR = os.path.join(REPO, "chinesener_amd", "ops")
 BUILD = os.path.join(REPO, "build")
 
 HIP_SOURCES = ["elementwise.hip", "crf.hip", "softlexicon.hip", "adam.hip",
-               "attention.hip", "tener.hip", "lstm.hip", "probe.hip"]
+               "attention.hip", "tener.hip", "lstm.hip", "probe.hip",
+               "wgrad.hip"]
 CPP_SOURCES = ["bindings.cpp"]
 
 
