@@ -1,18 +1,19 @@
-// Weight-gradient GEMM: dW[N, K_in] = dY^T @ X, with dY [T, N] and
-// X [T, K_in] (T = tokens = contraction dim, e.g. 8192 for bs64xL128).
+// Weight-gradient GEMM (EXPERIMENTAL, not on the hot path): dW[N, K_in]
+// = dY^T @ X with dY [T, N], X [T, K_in] (T = tokens = contraction dim).
 //
-// hipBLASLt's heuristic runs these K-bound shapes at ~410 TF/s with a
-// pathological split-K choice; this kernel owns the shape family
-// directly: 128x128 output tiles, K split across workgroups (grid.z),
-// both operands transposed-staged through LDS with bank-conflict-free
-// strides (ld % 16 == 8), MFMA 16x16x32 bf16, fp32 atomic accumulation
-// into a workspace that the Python wrapper casts to the weight dtype.
+// Status: exact (unit-tested vs fp32 matmul) but ~185 TF/s vs
+// hipBLASLt's ~440 TF/s on the BERT wgrad shapes, so autograd keeps the
+// library GEMM. The contraction dim T is the row index of both global
+// operands, so MFMA K-contiguous fragments require a transpose
+// somewhere: v1 transposed during staging (scalar LDS stores dominated,
+// 110 TF/s); v2 (this code) stages [K][N]-oriented tiles with fully
+// vectorized copies and pays 8 scalar LDS reads per fragment instead
+// (185 TF/s). Beating Tensile here needs the assembly-grade
+// buffer_load + ds_write-swizzle transpose pipeline — future work;
+// kept as a measured baseline for it.
 //
-// Tile anatomy per workgroup (4 waves):
-//   A = dY^T tile [128 m][64 k]   (m = output rows = dY columns)
-//   B = X^T  tile [128 n][64 k]   (n = output cols = X columns)
-//   each wave owns a 32x128 slab of C: m-frags {wave*2, wave*2+1},
-//   all 8 n-frags -> 16 accumulators (64 VGPRs).
+// Tile anatomy per workgroup (4 waves): 128x128 output tile, K split
+// across grid.z with fp32 atomic accumulation into a workspace.
 #include "common.h"
 
 using bfrag = mfma_bf16x8;
@@ -20,23 +21,28 @@ using cfrag = mfma_f32x4;
 
 #define WG_TILE 128
 #define K_CHUNK 64
-#define LDS_LD (K_CHUNK + 8)  // 72 elems: 16B groups land on odd banks
+#define LDS_N (WG_TILE + 8)  // row stride of the [K_CHUNK][128] tiles
 
-__device__ __forceinline__ bfrag wg_frag(const bf16* base, int i0, int k0) {
+// fragment from a [K][N]-oriented LDS tile: element (row i0+(l&15), k)
+// = tile[k][i0 + (l&15)] — 8 scalar LDS reads per fragment, but staging
+// stays fully vectorized (no transpose).
+__device__ __forceinline__ bfrag wg_fragT(const bf16* tile, int i0, int k0) {
   const int l = threadIdx.x & (WAVE - 1);
-  return *reinterpret_cast<const bfrag*>(base + (long)(i0 + (l & 15)) * LDS_LD +
-                                         k0 + ((l >> 4) << 3));
+  const int col = i0 + (l & 15);
+  const int kb = k0 + ((l >> 4) << 3);
+  bfrag f;
+#pragma unroll
+  for (int e = 0; e < 8; ++e)
+    reinterpret_cast<bf16*>(&f)[e] = tile[(long)(kb + e) * LDS_N + col];
+  return f;
 }
 
-// stage a [rows=128][K_CHUNK] transposed tile: global src [T, width] with
-// row stride `stride`, reading rows k0..k0+K_CHUNK (clamped to kend) and
-// columns c0..c0+128 -> LDS [128][LDS_LD] as tile[c][k].
-__device__ __forceinline__ void stage_T(const bf16* src, long stride, int c0,
-                                        int width, long k0, long kend,
-                                        bf16* lds) {
-  // each thread loads 8 consecutive columns of one k-row, then scatters
-  // into LDS transposed (8 scalar LDS writes)
-  const int per_row = WG_TILE / 8;            // 16 vector loads per k-row
+// stage [K_CHUNK][128] slab of a [T, width] tensor (rows k0.., cols c0..)
+// into LDS row-major — vectorized loads AND stores.
+__device__ __forceinline__ void stage_nt(const bf16* src, long stride, int c0,
+                                         int width, long k0, long kend,
+                                         bf16* lds) {
+  const int per_row = WG_TILE / 8;            // 16 vector ops per k-row
   for (int i = threadIdx.x; i < K_CHUNK * per_row; i += blockDim.x) {
     const int kk = i / per_row;
     const int cc = (i % per_row) * 8;
@@ -44,9 +50,7 @@ __device__ __forceinline__ void stage_T(const bf16* src, long stride, int c0,
     s16x8 v{};
     if (k < kend && c0 + cc + 7 < width)
       v = *reinterpret_cast<const s16x8*>(src + k * stride + c0 + cc);
-#pragma unroll
-    for (int e = 0; e < 8; ++e)
-      lds[(long)(cc + e) * LDS_LD + kk] = reinterpret_cast<const bf16*>(&v)[e];
+    *reinterpret_cast<s16x8*>(lds + (long)kk * LDS_N + cc) = v;
   }
 }
 
@@ -56,8 +60,8 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
     float* __restrict__ ws,       // [N, Kin] zeroed fp32 workspace
     long T, int N, int Kin, int tiles_n, int ksplit) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  bf16* a_s = reinterpret_cast<bf16*>(smem_raw);       // [128][LDS_LD] x2
-  bf16* b_s = a_s + 2 * WG_TILE * LDS_LD;              // [128][LDS_LD] x2
+  bf16* a_s = reinterpret_cast<bf16*>(smem_raw);       // [K_CHUNK][LDS_N] x2
+  bf16* b_s = a_s + 2 * K_CHUNK * LDS_N;               // [K_CHUNK][LDS_N] x2
   const int tile = blockIdx.x;
   const int tm = tile / tiles_n;
   const int tn = tile - tm * tiles_n;
@@ -76,31 +80,33 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
 #pragma unroll
     for (int j = 0; j < 8; ++j) acc[i][j] = cfrag{0.f, 0.f, 0.f, 0.f};
 
-  // double-buffered K loop
   int buf = 0;
-  stage_T(dy, N, m0, N, kbeg, kend, a_s);
-  stage_T(x, Kin, n0, Kin, kbeg, kend, b_s);
+  stage_nt(dy, N, m0, N, kbeg, kend, a_s);
+  stage_nt(x, Kin, n0, Kin, kbeg, kend, b_s);
   __syncthreads();
   for (long k0 = kbeg; k0 < kend; k0 += K_CHUNK) {
     const int cur = buf;
     if (k0 + K_CHUNK < kend) {
       const int nxt = 1 - buf;
-      stage_T(dy, N, m0, N, k0 + K_CHUNK, kend, a_s + nxt * WG_TILE * LDS_LD);
-      stage_T(x, Kin, n0, Kin, k0 + K_CHUNK, kend,
-              b_s + nxt * WG_TILE * LDS_LD);
+      stage_nt(dy, N, m0, N, k0 + K_CHUNK, kend, a_s + nxt * K_CHUNK * LDS_N);
+      stage_nt(x, Kin, n0, Kin, k0 + K_CHUNK, kend,
+               b_s + nxt * K_CHUNK * LDS_N);
       buf = nxt;
     }
-    const bf16* a = a_s + cur * WG_TILE * LDS_LD;
-    const bf16* b = b_s + cur * WG_TILE * LDS_LD;
+    const bf16* a = a_s + cur * K_CHUNK * LDS_N;
+    const bf16* b = b_s + cur * K_CHUNK * LDS_N;
 #pragma unroll
     for (int kk = 0; kk < K_CHUNK / 32; ++kk) {
+      bfrag bfr[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) bfr[j] = wg_fragT(b, j * 16, kk * 32);
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
-        const bfrag af = wg_frag(a, (wid * 2 + i) * 16, kk * 32);
+        const bfrag af = wg_fragT(a, (wid * 2 + i) * 16, kk * 32);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af, wg_frag(b, j * 16, kk * 32), acc[i][j], 0, 0, 0);
+              af, bfr[j], acc[i][j], 0, 0, 0);
       }
     }
     __syncthreads();
@@ -146,7 +152,7 @@ at::Tensor wgrad(const at::Tensor& dy, const at::Tensor& x, long ksplit) {
     ksplit = std::max<long>(1, want / std::max(1, tiles_m * tiles_n));
     ksplit = std::min<long>(ksplit, (T + K_CHUNK - 1) / K_CHUNK);
   }
-  const size_t smem = (size_t)4 * WG_TILE * LDS_LD * sizeof(bf16);
+  const size_t smem = (size_t)4 * K_CHUNK * LDS_N * sizeof(bf16);
   dim3 grid(tiles_m * tiles_n, 1, (unsigned)ksplit);
   hipLaunchKernelGGL(wgrad_kernel, grid, dim3(256), smem, cur_stream(dy),
                      (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),

@@ -494,3 +494,17 @@ def test_dropout_add_ln_matches_ln_when_kept():
     ref_y = O.add_layernorm(x, res, w, b, 1e-12)
     assert m.all()
     torch.testing.assert_close(y.float(), ref_y.float(), atol=3e-2, rtol=3e-2)
+
+
+def test_wgrad_kernel_exact():
+    """Experimental MFMA wgrad (csrc/wgrad.hip): exact vs fp32 matmul
+    (not on the hot path — see the file header for perf status)."""
+    _cuda()
+    torch.manual_seed(33)
+    ext = ops.get_ext()
+    for T, N, K in [(512, 128, 256), (1000, 256, 128)]:
+        dy = torch.randn(T, N, device="cuda", dtype=torch.bfloat16)
+        x = torch.randn(T, K, device="cuda", dtype=torch.bfloat16)
+        ref = dy.float().T @ x.float()
+        got = ext.wgrad(dy, x, 0)
+        torch.testing.assert_close(got, ref, atol=2.0, rtol=2e-2)
