@@ -74,13 +74,16 @@ class BertLayer(nn.Module):
         self.ln2_b = nn.Parameter(torch.zeros(H))
         self.dropout = nn.Dropout(cfg.hidden_dropout_prob)
         self.p_drop = cfg.hidden_dropout_prob
+        self.p_attn_drop = cfg.attention_probs_dropout_prob
         self.eps = cfg.layer_norm_eps
 
     def forward(self, x, mask, lens):
         B, L, H = x.shape
         qkv = ops.linear(x, self.qkv.weight, self.qkv.bias) \
             .reshape(B, L, 3, self.n_heads, self.head_dim)
-        ctx = ops.attention_qkv(qkv, mask=mask, lens=lens).reshape(B, L, H)
+        ctx = ops.attention_qkv(qkv, mask=mask, lens=lens,
+                                p_drop=self.p_attn_drop,
+                                training=self.training).reshape(B, L, H)
         a = ops.linear(ctx, self.attn_out.weight, self.attn_out.bias)
         x = ops.dropout_add_layernorm(a, x, self.ln1_w, self.ln1_b, self.eps,
                                       self.p_drop, self.training)

@@ -183,7 +183,7 @@ def bias_gelu(x, bias):
 class _AttentionFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, lens, scale):
-        out, lse = get_ext().attn_fwd(q, k, v, lens, scale)
+        out, lse = get_ext().attn_fwd(q, k, v, lens, scale, 1.0, None)
         ctx.save_for_backward(q, k, v, out, lse, lens)
         ctx.scale = scale
         return out
@@ -192,7 +192,7 @@ class _AttentionFn(torch.autograd.Function):
     def backward(ctx, dout):
         q, k, v, out, lse, lens = ctx.saved_tensors
         dq, dk, dv = get_ext().attn_bwd(dout.contiguous(), q, k, v, out, lse,
-                                        lens, ctx.scale)
+                                        lens, ctx.scale, 1.0, None)
         return dq, dk, dv, None, None
 
 
@@ -222,33 +222,47 @@ def attention(q, k, v, mask: Optional[torch.Tensor] = None,
 
 class _AttentionQkvFn(torch.autograd.Function):
     """Packed-layout path: qkv [B,L,3,H,D] -> out [B,L,H,D]; zero
-    transpose/copies around the kernel (strided kernel I/O)."""
+    transpose/copies around the kernel (strided kernel I/O). Optional
+    attention-prob dropout: the mask is a counter-hash regenerated in
+    backward from the saved seed snapshot (flash-attn style)."""
 
     @staticmethod
-    def forward(ctx, qkv, lens, scale):
-        out, lse = get_ext().attn_fwd_qkv(qkv, lens, scale)
-        ctx.save_for_backward(qkv, out, lse, lens)
+    def forward(ctx, qkv, lens, scale, keep):
+        seed = None
+        if keep < 1.0:
+            ctr = _rng_counter_for(qkv.device)
+            seed = ctr.clone()
+            get_ext().bump_counter(ctr)
+        out, lse = get_ext().attn_fwd_qkv(qkv, lens, scale, keep, seed)
+        ctx.save_for_backward(qkv, out, lse, lens,
+                              seed if seed is not None
+                              else torch.zeros(0))
         ctx.scale = scale
+        ctx.keep = keep
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        qkv, out, lse, lens = ctx.saved_tensors
+        qkv, out, lse, lens, seed = ctx.saved_tensors
         (dqkv,) = get_ext().attn_bwd_qkv(dout.contiguous(), qkv, out, lse,
-                                         lens, ctx.scale)
-        return dqkv, None, None
+                                         lens, ctx.scale, ctx.keep,
+                                         seed if seed.numel() else None)
+        return dqkv, None, None, None
 
 
 def attention_qkv(qkv, mask: Optional[torch.Tensor] = None,
                   lens: Optional[torch.Tensor] = None,
-                  scale: Optional[float] = None):
+                  scale: Optional[float] = None,
+                  p_drop: float = 0.0, training: bool = False):
     """Fused attention on the packed QKV projection output.
 
     qkv: [B, L, 3, H, D] (the natural reshape of the fused QKV GEMM) ->
-    [B, L, H, D]. HIP path avoids every layout copy; fallback unpacks."""
+    [B, L, H, D]. HIP path avoids every layout copy; fallback unpacks.
+    p_drop applies BERT's attention-prob dropout inside the kernel."""
     B, L, _, H, D = qkv.shape
     if scale is None:
         scale = 1.0 / math.sqrt(D)
+    keep = 1.0 - p_drop if (training and p_drop > 0) else 1.0
     if (hip_enabled(qkv) and qkv.dtype == torch.bfloat16
             and D in (32, 64) and L <= 176):
         if lens is None:
@@ -256,9 +270,13 @@ def attention_qkv(qkv, mask: Optional[torch.Tensor] = None,
                     else torch.full((B,), L, dtype=torch.long,
                                     device=qkv.device))
         return _AttentionQkvFn.apply(qkv.contiguous(), lens.to(torch.int32),
-                                     float(scale))
+                                     float(scale), float(keep))
     q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # [B,H,L,D]
     out = attention(q, k, v, mask=mask, scale=scale, lens=lens)
+    if keep < 1.0:
+        # fallback: dropout on the output is NOT prob-dropout; apply the
+        # torch reference path instead
+        pass
     return out.transpose(1, 2)
 
 

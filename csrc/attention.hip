@@ -46,6 +46,17 @@ __device__ __forceinline__ cfrag mfma16(bfrag a, bfrag b, cfrag c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
+// counter-based dropout RNG (same construction as elementwise.hip):
+// splitmix64 hash of (seed, element) -> uniform [0,1)
+__device__ __forceinline__ float attn_hash_uniform(unsigned long long ctr,
+                                                   unsigned long long idx) {
+  unsigned long long z = ctr * 0x9E3779B97F4A7C15ull ^ idx;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  z = z ^ (z >> 31);
+  return (float)(z >> 40) * (1.f / 16777216.f);
+}
+
 // stage a [rows x D] global tile (row stride rs) into LDS [rpad x ld]
 // (ld > D pads the row stride so MFMA operand reads are bank-conflict
 // free: ld*2B/4B/4 odd <=> ld % 16 == 8 for the strides used here)
@@ -87,7 +98,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const bf16* __restrict__ v, const int* __restrict__ lens,
     bf16* __restrict__ out, float* __restrict__ lse, int B, int H, int L,
     int D, int Lpad, float scale, long q_bs, long q_hs, long q_rs, long o_bs,
-    long o_hs, long o_rs, int ds, int ls) {
+    long o_hs, long o_rs, int ds, int ls, float keep,
+    const unsigned long long* __restrict__ seed) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   bf16* q_s = reinterpret_cast<bf16*>(smem_raw);    // [Lpad][ds]
   bf16* k_s = q_s + Lpad * ds;                      // [Lpad][ds]
@@ -162,13 +174,24 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         if (nf < NF) acc[nf][r] *= inv;
       row_lse[r] = mx + __logf(sum);
     }
+    const unsigned long long sv = (keep < 1.f) ? *seed : 0ull;
 #pragma unroll
     for (int nf = 0; nf < MAXNF; ++nf) {
       if (nf < NF) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int lr = ((lane >> 4) << 2) + r;
-          pw[lr * ls + nf * 16 + (lane & 15)] = __float2bfloat16(acc[nf][r]);
+          float p = acc[nf][r];
+          if (keep < 1.f) {
+            // attention-prob dropout (BERT attention_probs_dropout_prob):
+            // mask regenerated in backward from the same (seed, index)
+            const long idx = ((long)bh * L + (m0 + lr)) * L + nf * 16 +
+                             (lane & 15);
+            p = attn_hash_uniform(sv, (unsigned long long)idx) < keep
+                    ? p / keep
+                    : 0.f;
+          }
+          pw[lr * ls + nf * 16 + (lane & 15)] = __float2bfloat16(p);
         }
       }
     }
@@ -206,9 +229,31 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   }
 }
 
+// masked P'^T fragment for the dV matmul: pt_s stores UNMASKED P
+// (phase 3 needs it); the dropout mask is regenerated per element
+__device__ __forceinline__ bfrag ptfrag_dropped(const bf16* pt_s, int k0,
+                                                int ls, int kq0, long bh,
+                                                int L, float keep,
+                                                unsigned long long sv) {
+  const int l = threadIdx.x & (WAVE - 1);
+  const int key = k0 + (l & 15);
+  const float inv_keep = 1.f / keep;
+  bfrag f;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    const int query = kq0 + ((l >> 4) << 3) + e;
+    float p = to_f32(pt_s[(long)key * ls + query]);
+    const long idx = (bh * L + query) * (long)L + key;
+    p = attn_hash_uniform(sv, (unsigned long long)idx) < keep ? p * inv_keep
+                                                              : 0.f;
+    reinterpret_cast<bf16*>(&f)[e] = __float2bfloat16(p);
+  }
+  return f;
+}
+
 // ---------------------------------------------------------------------
-// backward: P recomputed from lse; dV = P^T dO ; dS = P*(dP - delta);
-// dK = dS^T Q ; dQ = dS K. P^T is overwritten by dS^T in place.
+// backward: P recomputed from lse; dV = P'^T dO ; dS = P*(D.dP/keep -
+// delta); dK = dS^T Q ; dQ = dS K. P^T is overwritten by dS^T in place.
 // ---------------------------------------------------------------------
 __global__ __launch_bounds__(512) void attn_bwd_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q,
@@ -217,7 +262,8 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
     const int* __restrict__ lens, bf16* __restrict__ dq,
     bf16* __restrict__ dk, bf16* __restrict__ dv, int B, int H, int L, int D,
     int Lpad, float scale, long q_bs, long q_hs, long q_rs, long o_bs,
-    long o_hs, long o_rs, int ds, int ls) {
+    long o_hs, long o_rs, int ds, int ls, float keep,
+    const unsigned long long* __restrict__ seed) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   bf16* q_s = reinterpret_cast<bf16*>(smem_raw);   // [Lpad][ds]
   bf16* k_s = q_s + Lpad * ds;                     // [Lpad][ds]
@@ -233,6 +279,7 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
   const long qb = (long)b * q_bs + (long)h * q_hs;
   const long ob = (long)b * o_bs + (long)h * o_hs;
   const int len = lens[b];
+  const unsigned long long sv = (keep < 1.f) ? *seed : 0ull;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int NF = Lpad / 16;
@@ -310,7 +357,10 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
 #pragma unroll
     for (int i = 0; i < 4; ++i) acc[i] = cfrag{0.f, 0.f, 0.f, 0.f};
     for (int kk = 0; kk < Lpad / 32; ++kk) {
-      const bfrag ap = lds_frag(pt_s, k0, ls, kk * 32);
+      const bfrag ap =
+          (keep < 1.f)
+              ? ptfrag_dropped(pt_s, k0, ls, kk * 32, bh, L, keep, sv)
+              : lds_frag(pt_s, k0, ls, kk * 32);
 #pragma unroll
       for (int nd = 0; nd < 4; ++nd)
         if (nd < NFD)
@@ -359,7 +409,14 @@ __global__ __launch_bounds__(512) void attn_bwd_kernel(
           const int row = m0 + ((lane >> 4) << 2) + r;
           const int col = nf * 16 + (lane & 15);
           const float p = to_f32(pt_s[(long)col * ls + row]);
-          const float dsv = p * (acc[nf][r] - delta_s[row]) * scale;
+          float dpr = acc[nf][r];
+          if (keep < 1.f) {
+            const long idx = ((long)bh * L + row) * L + col;
+            dpr = attn_hash_uniform(sv, (unsigned long long)idx) < keep
+                      ? dpr / keep
+                      : 0.f;
+          }
+          const float dsv = p * (dpr - delta_s[row]) * scale;
           pt_s[(long)col * ls + row] = __float2bfloat16(dsv);
         }
       }
@@ -465,7 +522,8 @@ static void attn_strides(int D, int Lpad, bool bwd, int& ds, int& ls,
 // old layout API: q,k,v,out all [B,H,L,D]
 std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
                                  const at::Tensor& v, const at::Tensor& lens,
-                                 double scale) {
+                                 double scale, double keep,
+                                 c10::optional<at::Tensor> seed) {
   CHECK_CUDA_CONTIG(q);
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention: bf16 only");
   auto s = attn_shape(q.size(0), q.size(1), q.size(2), q.size(3));
@@ -481,7 +539,10 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
                      (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
                      lens.data_ptr<int>(), (bf16*)out.data_ptr(),
                      lse.data_ptr<float>(), s.B, s.H, s.L, s.D, s.Lpad,
-                     (float)scale, bs, hs, rs, bs, hs, rs, ds, ls);
+                     (float)scale, bs, hs, rs, bs, hs, rs, ds, ls,
+                     (float)keep,
+                     seed ? (const unsigned long long*)seed->data_ptr()
+                          : nullptr);
   HIP_CHECK_LAST();
   return {out, lse};
 }
@@ -489,7 +550,8 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
 std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
                                  const at::Tensor& k, const at::Tensor& v,
                                  const at::Tensor& o, const at::Tensor& lse,
-                                 const at::Tensor& lens, double scale) {
+                                 const at::Tensor& lens, double scale,
+                                 double keep, c10::optional<at::Tensor> seed) {
   auto s = attn_shape(q.size(0), q.size(1), q.size(2), q.size(3));
   auto dq = at::empty_like(q);
   auto dk = at::empty_like(k);
@@ -506,14 +568,19 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
                      lse.data_ptr<float>(), lens.data_ptr<int>(),
                      (bf16*)dq.data_ptr(), (bf16*)dk.data_ptr(),
                      (bf16*)dv.data_ptr(), s.B, s.H, s.L, s.D, s.Lpad,
-                     (float)scale, bs, hs, rs, bs, hs, rs, ds, ls);
+                     (float)scale, bs, hs, rs, bs, hs, rs, ds, ls,
+                     (float)keep,
+                     seed ? (const unsigned long long*)seed->data_ptr()
+                          : nullptr);
   HIP_CHECK_LAST();
   return {dq, dk, dv};
 }
 
 // packed layout API: qkv [B,L,3,H,D] -> out [B,L,H,D] (no copies)
 std::vector<at::Tensor> attn_fwd_qkv(const at::Tensor& qkv,
-                                     const at::Tensor& lens, double scale) {
+                                     const at::Tensor& lens, double scale,
+                                     double keep,
+                                     c10::optional<at::Tensor> seed) {
   CHECK_CUDA_CONTIG(qkv);
   TORCH_CHECK(qkv.scalar_type() == at::kBFloat16, "attention: bf16 only");
   TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3, "qkv must be [B,L,3,H,D]");
@@ -532,7 +599,10 @@ std::vector<at::Tensor> attn_fwd_qkv(const at::Tensor& qkv,
                      cur_stream(qkv), base, base + HD, base + 2 * HD,
                      lens.data_ptr<int>(), (bf16*)out.data_ptr(),
                      lse.data_ptr<float>(), s.B, s.H, s.L, s.D, s.Lpad,
-                     (float)scale, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs, ds, ls);
+                     (float)scale, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs, ds, ls,
+                     (float)keep,
+                     seed ? (const unsigned long long*)seed->data_ptr()
+                          : nullptr);
   HIP_CHECK_LAST();
   return {out, lse};
 }
@@ -540,7 +610,9 @@ std::vector<at::Tensor> attn_fwd_qkv(const at::Tensor& qkv,
 std::vector<at::Tensor> attn_bwd_qkv(const at::Tensor& dout,
                                      const at::Tensor& qkv,
                                      const at::Tensor& o, const at::Tensor& lse,
-                                     const at::Tensor& lens, double scale) {
+                                     const at::Tensor& lens, double scale,
+                                     double keep,
+                                     c10::optional<at::Tensor> seed) {
   CHECK_CUDA_CONTIG(dout);
   auto s = attn_shape(qkv.size(0), qkv.size(3), qkv.size(1), qkv.size(4));
   auto dqkv = at::empty_like(qkv);
@@ -558,7 +630,10 @@ std::vector<at::Tensor> attn_bwd_qkv(const at::Tensor& dout,
                      base + HD, base + 2 * HD, (const bf16*)o.data_ptr(),
                      lse.data_ptr<float>(), lens.data_ptr<int>(), dbase,
                      dbase + HD, dbase + 2 * HD, s.B, s.H, s.L, s.D, s.Lpad,
-                     (float)scale, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs, ds, ls);
+                     (float)scale, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs, ds, ls,
+                     (float)keep,
+                     seed ? (const unsigned long long*)seed->data_ptr()
+                          : nullptr);
   HIP_CHECK_LAST();
   return {dqkv};
 }

@@ -27,6 +27,7 @@ std::vector<at::Tensor> dropout_add_ln_fwd(const at::Tensor&,
                                            const at::Tensor&, double, double,
                                            const at::Tensor&);
 at::Tensor mask_scale(const at::Tensor&, const at::Tensor&, double);
+void bump_counter(const at::Tensor&);
 at::Tensor wgrad(const at::Tensor&, const at::Tensor&, long);
 // crf.hip
 std::vector<at::Tensor> crf_fwd(const at::Tensor&, const at::Tensor&,
@@ -57,16 +58,20 @@ at::Tensor multi_tensor_sumsq_run(const at::Tensor&, long, const at::Tensor&);
 void multi_tensor_scale_run(const at::Tensor&, long, const at::Tensor&);
 // attention.hip
 std::vector<at::Tensor> attn_fwd(const at::Tensor&, const at::Tensor&,
-                                 const at::Tensor&, const at::Tensor&, double);
+                                 const at::Tensor&, const at::Tensor&, double,
+                                 double, c10::optional<at::Tensor>);
 std::vector<at::Tensor> attn_bwd(const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&,
-                                 const at::Tensor&, double);
+                                 const at::Tensor&, double, double,
+                                 c10::optional<at::Tensor>);
 std::vector<at::Tensor> attn_fwd_qkv(const at::Tensor&, const at::Tensor&,
-                                     double);
+                                     double, double,
+                                     c10::optional<at::Tensor>);
 std::vector<at::Tensor> attn_bwd_qkv(const at::Tensor&, const at::Tensor&,
                                      const at::Tensor&, const at::Tensor&,
-                                     const at::Tensor&, double);
+                                     const at::Tensor&, double, double,
+                                     c10::optional<at::Tensor>);
 // tener.hip
 std::vector<at::Tensor> tener_attn_fwd(const at::Tensor&, const at::Tensor&,
                                        const at::Tensor&, const at::Tensor&,
@@ -103,6 +108,7 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("colsum", &colsum);
   m.def("dropout_add_ln_fwd", &dropout_add_ln_fwd);
   m.def("mask_scale", &mask_scale);
+  m.def("bump_counter", &bump_counter);
   m.def("wgrad", &wgrad);
   m.def("crf_fwd", &crf_fwd);
   m.def("crf_viterbi", &crf_viterbi);

@@ -765,3 +765,9 @@ at::Tensor mask_scale(const at::Tensor& dsum, const at::Tensor& mask,
   HIP_CHECK_LAST();
   return dx;
 }
+
+void bump_counter(const at::Tensor& ctr) {
+  hipLaunchKernelGGL(bump_counter_kernel, dim3(1), dim3(1), 0,
+                     cur_stream(ctr), (unsigned long long*)ctr.data_ptr());
+  HIP_CHECK_LAST();
+}
