@@ -272,11 +272,11 @@ def attention_qkv(qkv, mask: Optional[torch.Tensor] = None,
         return _AttentionQkvFn.apply(qkv.contiguous(), lens.to(torch.int32),
                                      float(scale), float(keep))
     q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # [B,H,L,D]
-    out = attention(q, k, v, mask=mask, scale=scale, lens=lens)
     if keep < 1.0:
-        # fallback: dropout on the output is NOT prob-dropout; apply the
-        # torch reference path instead
-        pass
+        out = ref.attention(q, k, v, mask, scale, p_drop=p_drop,
+                            training=training)
+    else:
+        out = attention(q, k, v, mask=mask, scale=scale, lens=lens)
     return out.transpose(1, 2)
 
 

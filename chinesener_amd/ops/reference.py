@@ -43,7 +43,8 @@ def bias_gelu(x: torch.Tensor, bias: Optional[torch.Tensor] = None) -> torch.Ten
 # ------------------------------------------------------------- attention
 def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
               mask: Optional[torch.Tensor] = None,
-              scale: Optional[float] = None) -> torch.Tensor:
+              scale: Optional[float] = None,
+              p_drop: float = 0.0, training: bool = False) -> torch.Tensor:
     """Scaled dot-product attention. q,k,v: [B,H,L,D]; mask: [B,L] (1=keep).
 
     Matches BERT attention and the reference's
@@ -58,6 +59,8 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         key_mask = mask[:, None, None, :].to(scores.dtype)
         scores = scores + (1.0 - key_mask) * NEG_INF
     probs = torch.softmax(scores.float(), dim=-1).to(q.dtype)
+    if training and p_drop > 0:
+        probs = F.dropout(probs, p_drop, training)
     return torch.matmul(probs, v)
 
 
