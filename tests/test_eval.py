@@ -115,3 +115,32 @@ def test_evaluation_cli_roundtrip(tmp_path):
     rep = ev.gen_report()
     assert rep["micro avg"]["f1"] == 1.0
     assert rep["PER"]["support"] == 4
+
+
+def test_multieval_comparison_table(tmp_path):
+    """MultiEval: pandas table sorted by weighted-avg F1 (reference
+    evaluation.py:97-111)."""
+    import pickle
+    from chinesener_amd.data.datasets import get_spec
+    spec = get_spec("msra")
+    t = spec.tag2idx
+    data_dir = tmp_path / "msra"
+    data_dir.mkdir()
+    gold = [t["[CLS]"], t["B-PER"], t["I-PER"], t["O"], t["[SEP]"]]
+    good = {"pred_ids": __import__("numpy").array(gold),
+            "label_ids": __import__("numpy").array(gold),
+            "mask": __import__("numpy").ones(5, dtype=int)}
+    bad_ids = list(gold)
+    bad_ids[1] = t["O"]
+    bad = {"pred_ids": __import__("numpy").array(bad_ids),
+           "label_ids": __import__("numpy").array(gold),
+           "mask": __import__("numpy").ones(5, dtype=int)}
+    with open(data_dir / "good_predict.pkl", "wb") as f:
+        pickle.dump([good] * 3, f)
+    with open(data_dir / "bad_predict.pkl", "wb") as f:
+        pickle.dump([bad] * 3, f)
+    import evaluation
+    table = evaluation.MultiEval(["bad", "good"], "msra",
+                                 str(tmp_path)).gen_report()
+    assert list(table["model"]) == ["good", "bad"]   # sorted by F1 desc
+    assert table.iloc[0]["f1"] == 1.0
