@@ -23,7 +23,7 @@ from typing import Dict, List, Optional, Sequence, Tuple
 import numpy as np
 
 from ..eval.entity_eval import extract_spans
-from .datasets import get_spec, load_data
+from .datasets import load_data
 
 Chunk = Tuple[str, Optional[str]]  # (surface, entity type or None)
 

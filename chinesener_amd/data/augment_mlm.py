@@ -5,7 +5,7 @@ prediction — entity spans and their tags are preserved."""
 from __future__ import annotations
 
 import random
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import List, Optional, Sequence, Tuple
 
 import torch
 

@@ -3,7 +3,7 @@ decode (reference tools/predict_utils.py:6-60 and tools/infer_utils.py:76-118
 behaviour, re-implemented)."""
 from __future__ import annotations
 
-from typing import Dict, Iterable, List, Optional, Sequence
+from typing import Dict, List, Optional, Sequence
 
 SPECIAL_TAGS = ("[PAD]", "[CLS]", "[SEP]")
 

@@ -5,7 +5,6 @@ mrc/dataset.py:27-203; Tag2Query :12-16; padded_batch train / batch(1)
 predict)."""
 from __future__ import annotations
 
-import os
 from typing import Dict, Iterator, List, Optional, Sequence
 
 import numpy as np

@@ -4,7 +4,7 @@ O(L^2) span-match head over candidate (start, end) pairs, with a
 weighted 3-part loss and span extraction."""
 from __future__ import annotations
 
-from typing import Dict, List, Tuple
+from typing import Dict, Tuple
 
 import torch
 import torch.nn as nn
