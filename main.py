@@ -103,6 +103,7 @@ def main(argv=None):
         os.environ.setdefault("HIP_VISIBLE_DEVICES", str(args.device))
 
     rank, world_size = dist_init(args)
+    torch.manual_seed(int(RUN_CONFIG.get("seed", 1234)) + rank)
     pipe, params, model, model_dir, name = build_everything(args, rank, world_size)
     if args.clear_model and rank == 0:
         clear_model(model_dir)
