@@ -263,6 +263,8 @@ def attention_qkv(qkv, mask: Optional[torch.Tensor] = None,
     if scale is None:
         scale = 1.0 / math.sqrt(D)
     keep = 1.0 - p_drop if (training and p_drop > 0) else 1.0
+    if os.environ.get("CHINESENER_NO_ATTN_DROP") == "1":
+        keep = 1.0
     if (hip_enabled(qkv) and qkv.dtype == torch.bfloat16
             and D in (32, 64) and L <= 176):
         if lens is None:

@@ -22,7 +22,7 @@ from chinesener_amd.train.trainer import Trainer
 
 def main():
     assert torch.cuda.is_available()
-    torch.manual_seed(0)
+    torch.manual_seed(int(os.environ.get("CONV_SEED", "0")))
     name = "bert_bilstm_crf"
     out_dir = os.path.join(os.path.dirname(os.path.dirname(
         os.path.abspath(__file__))), "gpurun_out")
