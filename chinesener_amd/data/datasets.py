@@ -187,8 +187,13 @@ def synthetic_corpus(spec: DatasetSpec, split: str = "train",
     key = (spec.name, split)
     if n is None and key in _SYNTH_CACHE:
         return _SYNTH_CACHE[key]
+    # Train-size cap: CPU tests keep the default small; the shipped
+    # data/ caches are built with CHINESENER_SYNTH_TRAIN=8000 so the
+    # epoch-derived LR schedule is long enough for from-scratch (random
+    # init) convergence — see profiles/convergence_r01.md.
+    cap = int(os.environ.get("CHINESENER_SYNTH_TRAIN", "2000"))
     count = n if n is not None else {
-        "train": min(spec.n_train, 2000),   # capped for CPU-side tests
+        "train": min(spec.n_train, cap),
         "valid": min(spec.n_valid, 200),
         "test": min(spec.n_test, 200),
         "predict": min(spec.n_test, 200),
