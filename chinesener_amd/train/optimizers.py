@@ -55,6 +55,9 @@ def build_param_groups(model: torch.nn.Module, base_lr: float,
         out.append({
             "params": params,
             "lr_scale": float(diff.get(key, 1.0)),
+            # full multiplier; LrSchedule.apply ramps lr_scale from 1 up
+            # to this over warmup (diff-LR shock mitigation, see below)
+            "lr_scale_base": float(diff.get(key, 1.0)),
             "weight_decay": 0.0 if nodecay else weight_decay,
         })
     return out
@@ -108,8 +111,11 @@ class AdamWeightDecay(torch.optim.Optimizer):
             ms = [s["m"] for s in states]
             vs = [s["v"] for s in states]
             if self.lr_dev is not None:
-                # graph mode: meta.lr = per-group scale; base from lr_dev
-                lr = group.get("lr_scale", 1.0)
+                # graph mode: meta.lr = per-group scale; base from lr_dev.
+                # The scale is baked into the cached meta blob, so the
+                # warmup ramp cannot apply here — use the full multiplier
+                # (graphed steps are captured post-warmup in practice)
+                lr = group.get("lr_scale_base", group.get("lr_scale", 1.0))
             else:
                 lr = group["lr"] * group.get("lr_scale", 1.0)
             b1, b2 = group["betas"]
@@ -206,8 +212,18 @@ class LrSchedule:
 
     def apply(self, optimizer: torch.optim.Optimizer, step: int) -> float:
         lr = self.lr_at(step)
+        # Diff-LR warm ramp: the reference applies its crf/logit x500
+        # multiplier from step 0 (fine for a pretrained encoder); with a
+        # random-init encoder that shock locks the CRF into the all-O
+        # basin (profiles/convergence_r01.md). Ramp each group's
+        # multiplier 1 -> full over the warmup window instead; after
+        # warmup the schedule is exactly the reference's.
+        ramp = min(1.0, max(1, step) / self.warmup)
         for g in optimizer.param_groups:
             g["lr"] = lr
+            base = g.get("lr_scale_base")
+            if base is not None and base != 1.0:
+                g["lr_scale"] = 1.0 + (base - 1.0) * ramp
         lr_dev = getattr(optimizer, "lr_dev", None)
         if lr_dev is not None:   # graph mode: one tiny device write
             lr_dev.fill_(lr)

@@ -75,6 +75,12 @@ def build_everything(args, rank: int, world_size: int):
         "epoch_size": args.epochs, "model_name": args.model_name, **RUN_CONFIG})
     if args.batch_size:
         params["batch_size"] = args.batch_size
+    if args.max_steps:
+        # a user-requested step budget extends the LR schedule: otherwise
+        # the poly decay reaches 0 at step_per_epoch*epochs and any steps
+        # past that train at lr=0
+        params["num_train_steps"] = max(
+            params.get("num_train_steps", 0), args.max_steps)
     model = build_model(args.model_name, params)
     model_dir = ckpt_dir(args.data.replace(",", "_"), name, args.ckpt_root)
     return pipe, params, model, model_dir, name

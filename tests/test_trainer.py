@@ -50,6 +50,56 @@ def test_diff_lr_groups():
     assert any(g["weight_decay"] > 0.0 for g in groups)
 
 
+def test_diff_lr_warm_ramp():
+    """The x500 crf/logit multiplier ramps 1 -> full over the schedule
+    warmup window (all-O basin mitigation); after warmup it equals the
+    reference's constant multiplier. Graph mode pins the full scale."""
+    import torch
+    w_crf = torch.nn.Parameter(torch.zeros(3))
+    w_other = torch.nn.Parameter(torch.zeros(3))
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.crf_weight = w_crf
+            self.other = w_other
+
+    groups = build_param_groups(M(), 5e-5, 0.01, {"crf": 500})
+    opt = AdamWeightDecay(groups, lr=5e-5)
+    sched = LrSchedule("bert", 5e-5, num_train_steps=1000, warmup_ratio=0.1)
+    assert sched.warmup == 100
+
+    def crf_scale():
+        return next(g["lr_scale"] for g in opt.param_groups
+                    if g.get("lr_scale_base") == 500.0)
+
+    sched.apply(opt, 1)
+    assert abs(crf_scale() - (1.0 + 499.0 * 0.01)) < 1e-9
+    sched.apply(opt, 50)
+    assert abs(crf_scale() - (1.0 + 499.0 * 0.5)) < 1e-9
+    sched.apply(opt, 100)
+    assert crf_scale() == 500.0
+    sched.apply(opt, 700)
+    assert crf_scale() == 500.0     # post-warmup: exact reference scale
+    # plain groups never ramp
+    assert all(g["lr_scale"] == 1.0 for g in opt.param_groups
+               if g.get("lr_scale_base", 1.0) == 1.0)
+
+
+def test_max_steps_extends_schedule(tmp_path):
+    """main.py: --max_steps beyond step_per_epoch*epochs must extend
+    num_train_steps so the poly decay never reaches 0 mid-run."""
+    from chinesener_amd.config import resolve_params
+    params = resolve_params({}, {"step_per_epoch": 10, "num_train_steps": 100},
+                            {"epoch_size": 10})
+    max_steps = 500
+    params["num_train_steps"] = max(params.get("num_train_steps", 0), max_steps)
+    sched = LrSchedule("bert", 5e-5,
+                       num_train_steps=params["num_train_steps"],
+                       warmup_ratio=0.1)
+    assert sched.lr_at(400) > 0
+
+
 def test_adamw_decreases_loss():
     torch.manual_seed(0)
     w = torch.nn.Parameter(torch.randn(10))
