@@ -26,6 +26,54 @@ TENSOR_KEYS_INT = ("token_ids", "label_ids", "mask", "softword_ids",
 TENSOR_KEYS_FLOAT = ("ex_softword_ids", "softlexicon_weights")
 
 
+class DevicePrefetcher:
+    """One-batch-ahead host->device staging (replaces the reference's
+    tf.data prefetch, dataset.py:43,52). Pins each host batch and issues
+    its H2D copies on a dedicated HIP stream while the previous batch
+    computes; the compute stream waits on a recorded event before the
+    batch is handed out. Pass-through on CPU."""
+
+    def __init__(self, it, device):
+        self.it = iter(it)
+        self.device = torch.device(device)
+        self.use_stream = self.device.type == "cuda"
+        if self.use_stream:
+            self.stream = torch.cuda.Stream(device=self.device)
+        self._next = None
+        self._event = None
+        self._host = None          # keep pinned source alive until copied
+        self._preload()
+
+    def _preload(self):
+        try:
+            host = next(self.it)
+        except StopIteration:
+            self._next = None
+            return
+        if not self.use_stream:
+            self._next = host
+            return
+        host = {k: v.pin_memory() for k, v in host.items()}
+        with torch.cuda.stream(self.stream):
+            self._next = {k: v.to(self.device, non_blocking=True)
+                          for k, v in host.items()}
+            self._event = torch.cuda.Event()
+            self._event.record(self.stream)
+        self._host = host
+
+    def __iter__(self):
+        return self
+
+    def __next__(self):
+        if self._next is None:
+            raise StopIteration
+        batch = self._next
+        if self.use_stream:
+            torch.cuda.current_stream(self.device).wait_event(self._event)
+        self._preload()
+        return batch
+
+
 def _to_tensors(arrays: Dict[str, np.ndarray], idx: np.ndarray) -> Dict[str, torch.Tensor]:
     out = {}
     for k, v in arrays.items():

@@ -121,6 +121,10 @@ class Trainer:
         best_eval, since_best = float("inf"), 0
         t0, last_log_step = time.time(), self.step
         losses = []
+        if str(self.device).startswith("cuda"):
+            # one-batch-ahead pinned H2D staging on a side stream
+            from ..data.loader import DevicePrefetcher
+            batches = DevicePrefetcher(batches, self.device)
         for batch in batches:
             loss = self.train_step(batch)
             losses.append(loss)

@@ -164,3 +164,15 @@ def test_text_embedding_loader(tmp_path):
                                 np.array([[9, 9, 9], [7, 7, 7]], np.float32))
     assert cv == ["你", "好", "你好"]
     assert cm.shape == (3, 3) and (cm[2] == 7).all()
+
+
+def test_device_prefetcher_cpu_passthrough():
+    from chinesener_amd.data.loader import DevicePrefetcher
+    import torch
+    src = [{"x": torch.full((2, 2), float(i))} for i in range(5)]
+    out = list(DevicePrefetcher(iter(src), "cpu"))
+    assert len(out) == 5
+    for i, b in enumerate(out):
+        assert torch.equal(b["x"], src[i]["x"])
+    # empty iterator
+    assert list(DevicePrefetcher(iter([]), "cpu")) == []
