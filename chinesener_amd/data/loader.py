@@ -53,9 +53,13 @@ class DevicePrefetcher:
         if not self.use_stream:
             self._next = host
             return
-        host = {k: v.pin_memory() for k, v in host.items()}
+        host = {k: (v.pin_memory()
+                    if isinstance(v, torch.Tensor) and v.device.type == "cpu"
+                    else v)
+                for k, v in host.items()}
         with torch.cuda.stream(self.stream):
-            self._next = {k: v.to(self.device, non_blocking=True)
+            self._next = {k: (v.to(self.device, non_blocking=True)
+                              if isinstance(v, torch.Tensor) else v)
                           for k, v in host.items()}
             self._event = torch.cuda.Event()
             self._event.record(self.stream)
