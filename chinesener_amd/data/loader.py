@@ -53,17 +53,22 @@ class DevicePrefetcher:
         if not self.use_stream:
             self._next = host
             return
-        host = {k: (v.pin_memory()
-                    if isinstance(v, torch.Tensor) and v.device.type == "cpu"
-                    else v)
-                for k, v in host.items()}
-        with torch.cuda.stream(self.stream):
-            self._next = {k: (v.to(self.device, non_blocking=True)
-                              if isinstance(v, torch.Tensor) else v)
-                          for k, v in host.items()}
-            self._event = torch.cuda.Event()
-            self._event.record(self.stream)
-        self._host = host
+        try:
+            host = {k: (v.pin_memory()
+                        if isinstance(v, torch.Tensor) and v.device.type == "cpu"
+                        else v)
+                    for k, v in host.items()}
+            with torch.cuda.stream(self.stream):
+                self._next = {k: (v.to(self.device, non_blocking=True)
+                                  if isinstance(v, torch.Tensor) else v)
+                              for k, v in host.items()}
+                self._event = torch.cuda.Event()
+                self._event.record(self.stream)
+            self._host = host
+        except Exception:
+            # degrade to synchronous hand-off (consumer moves to device)
+            self.use_stream = False
+            self._next = host
 
     def __iter__(self):
         return self
