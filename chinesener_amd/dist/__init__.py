@@ -119,12 +119,17 @@ class BucketedDataParallel:
                                          group=self.group, async_op=True)
             if b.work is not None:
                 b.work.wait()
+            b.flat.mul_(inv)     # one averaging kernel per bucket
             for i, p in enumerate(b.params):
                 if p.grad is None:
                     continue
                 off = b.offsets[i]
-                p.grad.detach().reshape(-1).copy_(b.flat[off:off + p.numel()])
-                p.grad.detach().mul_(inv)
+                # copy_ into the grad tensor directly: correct for any
+                # layout (a reshape(-1) view would silently write a temp
+                # if the grad were non-contiguous) and casts back to the
+                # grad dtype (fp32 LN affines with bf16 buckets)
+                p.grad.detach().copy_(
+                    b.flat[off:off + p.numel()].view_as(p.grad))
             b.ready = 0
             b.work = None
 
