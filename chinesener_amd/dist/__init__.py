@@ -16,12 +16,21 @@ import torch
 import torch.distributed as dist
 
 
-def init_process_group(backend: Optional[str] = None):
-    """Initialize from torchrun env vars; returns (rank, world_size)."""
+def init_process_group(backend: Optional[str] = None,
+                       timeout_minutes: float = 10.0):
+    """Initialize from torchrun env vars; returns (rank, world_size).
+
+    Rank-failure detection: async RCCL error handling is
+    forced on so a collective that a dead peer will never join raises in
+    the surviving ranks instead of hanging the job, and the collective
+    timeout is bounded so torchrun can tear the job down with a clear
+    diagnostic."""
     if dist.is_initialized():
         return dist.get_rank(), dist.get_world_size()
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
+    # 1 = TearDown: abort RCCL communicators + kill the process on error
+    os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
@@ -29,7 +38,9 @@ def init_process_group(backend: Optional[str] = None):
         torch.cuda.set_device(local_rank)
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
-    dist.init_process_group(backend=backend, rank=rank, world_size=world)
+    from datetime import timedelta
+    dist.init_process_group(backend=backend, rank=rank, world_size=world,
+                            timeout=timedelta(minutes=timeout_minutes))
     return rank, world
 
 

@@ -61,6 +61,14 @@ class Trainer:
         self.use_step_graph = (self.device.startswith("cuda")
                                and os.environ.get("CHINESENER_STEPGRAPH")
                                == "1")
+        # layer activation summaries (reference add_layer_summary,
+        # tools/utils.py:25-27): opt-in via params["verbose"] — JSONL in
+        # the checkpoint dir instead of TensorBoard events
+        self.summary = None
+        if params.get("verbose") and rank == 0:
+            from .summaries import SummaryLogger
+            self.summary = SummaryLogger(
+                model, ckpt_dir, every=params.get("summary_steps", 10))
 
     # ------------------------------------------------------------- train
     def _cast(self, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
@@ -100,8 +108,12 @@ class Trainer:
             self.dp.zero_grad()
         else:
             self.optimizer.zero_grad(set_to_none=True)
+        if self.summary is not None:
+            self.summary.maybe_arm(self.step + 1)
         with roctx_range("forward"):
             out = self._forward(batch)
+        if self.summary is not None:
+            self.summary.flush(self.step + 1)
         with roctx_range("backward"):
             out.loss.backward()
         if self.dp is not None:

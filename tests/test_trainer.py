@@ -224,3 +224,28 @@ def test_adamw_bf16_master_cpu_math():
     assert master.dtype == torch.float32
     torch.testing.assert_close(p.detach().float(),
                                master.to(torch.bfloat16).float())
+
+
+def test_summary_logger(tmp_path):
+    """verbose=True writes zero-fraction/hist JSONL at summary_steps
+    cadence (reference add_layer_summary, tools/utils.py:25-27)."""
+    import json
+    import torch
+    from chinesener_amd.train.trainer import Trainer
+    from chinesener_amd.models import build_model
+    params = make_tiny_params("bilstm_crf")
+    params.update(verbose=True, summary_steps=2, dtype="fp32")
+    model = build_model("bilstm_crf", params)
+    tr = Trainer(model, "bilstm_crf", params, str(tmp_path), device="cpu")
+    for i in range(4):
+        tr.train_step(make_tiny_batch("bilstm_crf", batch_size=2, seed=i))
+    path = tmp_path / "summaries.jsonl"
+    assert path.exists()
+    recs = [json.loads(l) for l in path.read_text().splitlines()]
+    assert [r["step"] for r in recs] == [2, 4]
+    layers = recs[0]["layers"]
+    assert layers, "no layer stats captured"
+    stat = next(iter(layers.values()))
+    assert set(stat) >= {"zero_fraction", "mean", "std", "absmax", "hist"}
+    assert len(stat["hist"]) == 16
+    assert 0.0 <= stat["zero_fraction"] <= 1.0
