@@ -15,6 +15,7 @@ Requests are padded up to the nearest captured batch size.
 from __future__ import annotations
 
 import logging
+import threading
 from typing import Dict, List, Optional, Sequence
 
 import numpy as np
@@ -66,6 +67,7 @@ class InferenceEngine:
                           else self.device.startswith("cuda"))
         self._graphs: Dict[int, _CapturedGraph] = {}
         self.n_requests = 0
+        self._lock = threading.Lock()
 
     # ---------------------------------------------------------- capture
     def _example_features(self, batch: int) -> Dict[str, torch.Tensor]:
@@ -141,7 +143,15 @@ class InferenceEngine:
     def predict(self, features: Dict[str, np.ndarray]) -> np.ndarray:
         """features: numpy arrays [B, ...] (as built by
         data.preprocess.BasicProc.build_seq_feature, stacked).
-        Returns pred_ids [B, L]."""
+        Returns pred_ids [B, L].
+
+        Serialized per engine: the gRPC server runs a thread pool and the
+        hipGraph static input/output buffers are shared state — without
+        the lock two in-flight requests would overwrite each other."""
+        with self._lock:
+            return self._predict(features)
+
+    def _predict(self, features: Dict[str, np.ndarray]) -> np.ndarray:
         self.n_requests += 1
         batch = features["token_ids"].shape[0]
         tensors = {}

@@ -78,7 +78,7 @@ def build_server(engines: Dict[str, InferenceEngine], port: int,
 
 def serve(model_names, export_root: str = EXPORT_DIR,
           port: int = rpc.DEFAULT_PORT, wait: bool = True,
-          use_graph=None, batch_sizes=(1, 4, 8)):
+          use_graph=None, batch_sizes=(1, 4, 8), max_workers: int = 4):
     engines = {}
     for name in model_names:
         eng = InferenceEngine(name, export_root, batch_sizes=batch_sizes,
@@ -87,7 +87,7 @@ def serve(model_names, export_root: str = EXPORT_DIR,
         eng.warmup(warm)
         log.info("loaded %s (%d warmup requests replayed)", name, len(warm))
         engines[name] = eng
-    server = build_server(engines, port)
+    server = build_server(engines, port, max_workers=max_workers)
     server.start()
     log.info("serving %s on :%d", sorted(engines), port)
     if wait:

@@ -173,3 +173,41 @@ def test_retry_backoff_budget():
     with pytest.raises(grpc.RpcError):
         always_down()
     assert len(calls) == 4  # 1 initial + 3 retries (UNAVAILABLE budget)
+
+
+def test_grpc_concurrent_requests(tmp_path):
+    """Thread-pooled server + per-engine lock: concurrent requests each
+    get a response consistent with their own input (the hipGraph static
+    buffers are shared state; engine.predict serializes)."""
+    import threading
+    _export_tiny(tmp_path)
+    port = _free_port()
+    from chinesener_amd.serve.server import serve
+    from chinesener_amd.serve.client import PredictionClient
+    server = serve(["bilstm_crf"], str(tmp_path), port=port, wait=False,
+                   use_graph=False, max_workers=4)
+    errs = []
+
+    def worker(seed):
+        try:
+            client = PredictionClient(port=port)
+            rng = np.random.default_rng(seed)
+            for _ in range(5):
+                b = int(rng.integers(1, 4))
+                feats = {"token_ids": rng.integers(1, 200, (b, 32)),
+                         "mask": np.ones((b, 32), dtype=np.int64)}
+                resp = client.predict("bilstm_crf", feats)
+                assert resp["outputs"]["pred_ids"].shape == (b, 32)
+            client.close()
+        except Exception as e:   # surfaced to the main thread
+            errs.append(e)
+
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(4)]
+    try:
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=120)
+    finally:
+        server.stop(0)
+    assert not errs, errs
