@@ -53,6 +53,10 @@ def main():
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
+    if args.gpus > 1 and world == 1:
+        print(f"[bench] --gpus {args.gpus} requested but WORLD_SIZE=1 — "
+              f"launch via torch.distributed.run for multi-GPU; "
+              f"measuring 1 GPU and reporting n_gpus=1", file=sys.stderr)
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     use_gpu = torch.cuda.is_available()
     device = f"cuda:{local_rank}" if use_gpu else "cpu"

@@ -1,0 +1,88 @@
+#!/usr/bin/env python3
+"""Long-run soak of the hipGraph-captured training step (docs/ROADMAP.md
+item 5): train bert_bilstm_crf for SOAK_STEPS with CHINESENER_STEPGRAPH=1
+and watch for non-finite loss, divergence from a trailing-loss envelope,
+or eval-F1 collapse. Exit code 0 = clean soak.
+
+Usage (GPU box):
+    SOAK_STEPS=10000 python scripts/soak_graph.py
+Env: SOAK_STEPS (default 5000), SOAK_EVAL_EVERY (1000), SOAK_SEED (0).
+Writes gpurun_out/soak_graph.json.
+"""
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+os.environ["CHINESENER_STEPGRAPH"] = "1"
+
+import torch
+
+from chinesener_amd.config import resolve_params
+from chinesener_amd.data.loader import NerDataset
+from chinesener_amd.eval import process_prediction
+from chinesener_amd.eval.entity_eval import entity_report
+from chinesener_amd.models import build_model, model_params
+from chinesener_amd.train.trainer import Trainer
+
+
+def main():
+    assert torch.cuda.is_available()
+    steps = int(os.environ.get("SOAK_STEPS", "5000"))
+    eval_every = int(os.environ.get("SOAK_EVAL_EVERY", "1000"))
+    torch.manual_seed(int(os.environ.get("SOAK_SEED", "0")))
+    name = "bert_bilstm_crf"
+    pipe = NerDataset("/tmp/soak_data", "msra", 64, 1, name)
+    params = resolve_params(model_params(name), pipe.params, {
+        "model_name": name, "num_train_steps": max(steps, 2000),
+        "warmup_ratio": 0.1, "lr": 5e-5})
+    trainer = Trainer(build_model(name, params), name, params, "/tmp/soak_ck")
+
+    def batches():
+        while True:
+            yield from pipe.iter_batches("train")
+
+    gen = batches()
+    t0 = time.time()
+    history, bad = [], None
+    trail = []
+    for step in range(1, steps + 1):
+        loss = trainer.train_step(next(gen))
+        if not (loss == loss and abs(loss) < 1e6):
+            bad = f"non-finite/exploded loss {loss} at step {step}"
+            break
+        trail.append(loss)
+        if len(trail) > 200:
+            trail.pop(0)
+        # after warmup, a healthy run never jumps 5x above its trailing mean
+        if step > 1000 and loss > 5 * (sum(trail) / len(trail)) + 1.0:
+            bad = f"loss spike {loss:.3f} vs trailing {sum(trail)/len(trail):.3f} at {step}"
+            break
+        if step % eval_every == 0:
+            rows = trainer.predict(pipe.iter_batches("valid", shuffle=False))
+            idx2tag = pipe.params["idx2tag"]
+            proc = [process_prediction(r, idx2tag) for r in rows]
+            f1 = entity_report([p["label_tags"] for p in proc],
+                               [p["pred_tags"] for p in proc])["micro avg"]["f1"]
+            history.append({"step": step, "loss": round(loss, 3),
+                            "f1": round(f1, 4)})
+            print(f"step {step} loss {loss:.3f} F1 {f1:.4f} "
+                  f"graphed={trainer._graph is not None}", flush=True)
+            if step >= 2 * eval_every and f1 < 0.2:
+                bad = f"F1 collapse {f1:.4f} at step {step}"
+                break
+    result = {"steps_done": step, "clean": bad is None, "failure": bad,
+              "graph_used": trainer._graph is not None,
+              "wall_s": round(time.time() - t0, 1), "history": history}
+    out = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "gpurun_out")
+    os.makedirs(out, exist_ok=True)
+    with open(os.path.join(out, "soak_graph.json"), "w") as f:
+        json.dump(result, f, indent=1)
+    print(json.dumps(result)[:500])
+    sys.exit(0 if bad is None else 1)
+
+
+if __name__ == "__main__":
+    main()
