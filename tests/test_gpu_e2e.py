@@ -257,3 +257,31 @@ def test_every_model_trains_on_gpu(model_name):
         losses.append(float(out.loss.detach()))
     assert all(torch.isfinite(torch.tensor(losses))), (model_name, losses)
     assert out.pred_ids is not None
+
+
+@pytest.mark.gpu
+def test_device_prefetcher_gpu():
+    """Pinned side-stream H2D staging delivers every batch, on-device,
+    bit-identical, in order (data/loader.py DevicePrefetcher)."""
+    _cuda()
+    from chinesener_amd.data.loader import DevicePrefetcher
+    src = [{"token_ids": torch.randint(0, 100, (4, 32)),
+            "w": torch.randn(4, 8)} for _ in range(6)]
+    out = list(DevicePrefetcher(iter(src), "cuda:0"))
+    assert len(out) == 6
+    for host, dev in zip(src, out):
+        for k in host:
+            assert dev[k].is_cuda
+            assert torch.equal(dev[k].cpu(), host[k])
+    # and through trainer.train (the wrapping call site)
+    from conftest import make_tiny_batch, make_tiny_params
+    from chinesener_amd.train.trainer import Trainer
+    from chinesener_amd.models import build_model
+    params = make_tiny_params("bilstm_crf")
+    model = build_model("bilstm_crf", params)
+    tr = Trainer(model, "bilstm_crf", params, "/tmp/prefetch_ck",
+                 device="cuda:0")
+    batches = [make_tiny_batch("bilstm_crf", batch_size=2, seed=i)
+               for i in range(6)]
+    res = tr.train(iter(batches), log_steps=100, save_steps=0)
+    assert res["step"] == 6
