@@ -72,6 +72,43 @@ MODELS: Dict[str, Tuple[Type[NerModel], Dict]] = {
 }
 
 
+# Per-model extra CLI flags (reference AddonParser pattern,
+# tools/train_utils.py:14-44 consumed in each model module's main block):
+# the train driver appends these to argparse and merges parsed values
+# into the resolved params.
+from ..train.addon_parser import Addon  # noqa: E402
+
+MODEL_ADDONS: Dict[str, list] = {
+    "bert_dice": [Addon("alpha", 0.1, float, help="dice focal down-weight"),
+                  Addon("gamma", 1.0, float, help="dice smoothing")],
+    "bert_bilstm_crf_mtl": [
+        Addon("task_weight_1", 0.5, float, help="loss weight of task 1"),
+        Addon("asymmetry", None, action="store_true",
+              help="feed task1 hidden into task2 tower")],
+    "bert_bilstm_crf_adv": [
+        Addon("task_weight_1", 0.5, float, help="loss weight of task 1"),
+        Addon("lambda_adv", 0.05, float, help="adversarial loss weight"),
+        Addon("shrink_gradient_reverse", 0.01, float,
+              help="flip-gradient scale")],
+}
+
+
+def model_addons(name: str) -> list:
+    return list(MODEL_ADDONS.get(name, []))
+
+
+def apply_addon_values(name: str, params: Dict, values: Dict) -> Dict:
+    """Merge parsed addon values into params (None = keep default)."""
+    vals = {k: v for k, v in values.items() if v is not None}
+    if "task_weight_1" in vals:
+        w1 = float(vals.pop("task_weight_1"))
+        params["task_weight"] = (w1, 1.0 - w1)
+    if "lambda_adv" in vals:
+        params["lambda"] = float(vals.pop("lambda_adv"))
+    params.update(vals)
+    return params
+
+
 def model_params(name: str) -> Dict:
     if name not in MODELS:
         raise KeyError(f"unknown model '{name}' (known: {sorted(MODELS)})")

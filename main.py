@@ -81,6 +81,9 @@ def build_everything(args, rank: int, world_size: int):
         # past that train at lr=0
         params["num_train_steps"] = max(
             params.get("num_train_steps", 0), args.max_steps)
+    from chinesener_amd.models import apply_addon_values
+    apply_addon_values(args.model_name, params,
+                       getattr(args, "_addon_values", {}))
     model = build_model(args.model_name, params)
     model_dir = ckpt_dir(args.data.replace(",", "_"), name, args.ckpt_root)
     return pipe, params, model, model_dir, name
@@ -103,7 +106,18 @@ def main(argv=None):
     ap.add_argument("--max_steps", type=int, default=None)
     ap.add_argument("--data_dir", default=DATA_DIR)
     ap.add_argument("--ckpt_root", default=CHECKPOINT_DIR)
+    # per-model extra flags (reference AddonParser: each model module
+    # declares additional hyperparameters merged into the CLI)
+    from chinesener_amd.models import model_addons
+    from chinesener_amd.train.addon_parser import AddonParser
+    pre = argparse.ArgumentParser(add_help=False)
+    pre.add_argument("--model_name", default=None)
+    known, _ = pre.parse_known_args(argv)
+    addon_parser = AddonParser(model_addons(known.model_name)
+                               if known.model_name else [])
+    addon_parser.append(ap)
     args = ap.parse_args(argv)
+    args._addon_values = addon_parser.extract(args)
 
     if args.device is not None:
         os.environ.setdefault("HIP_VISIBLE_DEVICES", str(args.device))

@@ -249,3 +249,25 @@ def test_summary_logger(tmp_path):
     assert set(stat) >= {"zero_fraction", "mean", "std", "absmax", "hist"}
     assert len(stat["hist"]) == 16
     assert 0.0 <= stat["zero_fraction"] <= 1.0
+
+
+def test_main_cli_addon_flags():
+    """Per-model AddonParser flags appear in main.py's CLI and land in
+    params (reference pattern: model modules declare extra flags)."""
+    import subprocess
+    env = dict(os.environ, PYTHONPATH=REPO)
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "main.py"),
+         "--model_name", "bert_dice", "--data", "msra", "--help"],
+        capture_output=True, text=True, env=env, timeout=120)
+    assert r.returncode == 0
+    assert "--alpha" in r.stdout and "--gamma" in r.stdout
+
+    from chinesener_amd.models import apply_addon_values, model_params
+    params = model_params("bert_bilstm_crf_adv")
+    apply_addon_values("bert_bilstm_crf_adv", params,
+                       {"task_weight_1": 0.7, "lambda_adv": 0.2,
+                        "shrink_gradient_reverse": None})
+    assert params["task_weight"] == (0.7, 0.3)
+    assert params["lambda"] == 0.2
+    assert params["shrink_gradient_reverse"] == 0.01   # None kept default
