@@ -268,6 +268,7 @@ def test_main_cli_addon_flags():
     apply_addon_values("bert_bilstm_crf_adv", params,
                        {"task_weight_1": 0.7, "lambda_adv": 0.2,
                         "shrink_gradient_reverse": None})
-    assert params["task_weight"] == (0.7, 0.3)
+    assert params["task_weight"][0] == 0.7
+    assert abs(params["task_weight"][1] - 0.3) < 1e-12
     assert params["lambda"] == 0.2
     assert params["shrink_gradient_reverse"] == 0.01   # None kept default
