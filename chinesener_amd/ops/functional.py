@@ -481,8 +481,8 @@ def bilstm(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b, lens,
         if not _LSTM_FALLBACK_WARNED:
             import logging
             logging.getLogger("chinesener_amd").warning(
-                "BiLSTM hidden=%d not kernel-eligible (needs %%32==0, <=128);"
-                " using torch recurrence", h)
+                "BiLSTM hidden=%d not kernel-eligible (kernel covers "
+                "hidden <= 256 via zero-padding); using torch recurrence", h)
             _LSTM_FALLBACK_WARNED = True
     return ref.bilstm_forward(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b,
                               lens, activation, state_dropout, cell_clip)
