@@ -1,0 +1,133 @@
+"""Property-based fuzz tests (hypothesis) for the algorithmic cores:
+CRF reference vs brute-force enumeration on random inputs, strict-span
+extraction invariants, softlexicon fuse vs a plain numpy re-derivation,
+and Viterbi optimality. Complements the fixed fixtures in
+tests/test_ops_reference.py."""
+import itertools
+
+import numpy as np
+import torch
+from hypothesis import given, settings, strategies as st
+
+from chinesener_amd.eval.entity_eval import extract_spans
+from chinesener_amd.ops import reference as ref
+
+TYPES = ["LOC", "PER", "ORG"]
+
+
+@st.composite
+def tag_sequences(draw):
+    n = draw(st.integers(1, 12))
+    tags = []
+    for _ in range(n):
+        kind = draw(st.sampled_from(["O", "B", "I"]))
+        if kind == "O":
+            tags.append("O")
+        else:
+            tags.append(f"{kind}-{draw(st.sampled_from(TYPES))}")
+    return tags
+
+
+@given(tag_sequences())
+@settings(max_examples=200, deadline=None)
+def test_extract_spans_invariants(tags):
+    spans = extract_spans(tags)
+    seen_positions = set()
+    for typ, s, e in spans:
+        # well-formed, in-range, typed
+        assert 0 <= s < e <= len(tags)
+        assert typ in TYPES
+        # spans never overlap
+        assert not (set(range(s, e)) & seen_positions)
+        seen_positions.update(range(s, e))
+        # first position of a span is B- or a type-switching I-
+        assert tags[s][2:] == typ
+    # every B- tag opens exactly one span
+    n_b = sum(1 for t in tags if t.startswith("B-"))
+    # I- after O or after a different type also opens one (seqeval default)
+    n_orphan_i = sum(
+        1 for i, t in enumerate(tags)
+        if t.startswith("I-") and (i == 0 or tags[i - 1] == "O"
+                                   or tags[i - 1][2:] != t[2:]))
+    assert len(spans) == n_b + n_orphan_i
+
+
+@given(st.integers(2, 4), st.integers(1, 5), st.randoms())
+@settings(max_examples=60, deadline=None)
+def test_crf_loglik_matches_bruteforce(T, L, rnd):
+    g = torch.Generator().manual_seed(rnd.randint(0, 2**31))
+    em = torch.randn(1, L, T, generator=g)
+    trans = torch.randn(T, T, generator=g)
+    tags = torch.randint(0, T, (1, L), generator=g)
+    mask = torch.ones(1, L, dtype=torch.long)
+    ll = ref.crf_log_likelihood(em, tags, mask, trans)[0]
+
+    def score(path):
+        s = sum(float(em[0, t, p]) for t, p in enumerate(path))
+        s += sum(float(trans[path[t], path[t + 1]]) for t in range(L - 1))
+        return s
+
+    scores = [score(p) for p in itertools.product(range(T), repeat=L)]
+    logZ = float(torch.logsumexp(torch.tensor(scores), 0))
+    expect = score([int(x) for x in tags[0]]) - logZ
+    assert abs(float(ll) - expect) < 1e-4
+
+
+@given(st.integers(2, 4), st.integers(1, 5), st.randoms())
+@settings(max_examples=60, deadline=None)
+def test_crf_viterbi_is_optimal(T, L, rnd):
+    g = torch.Generator().manual_seed(rnd.randint(0, 2**31))
+    em = torch.randn(1, L, T, generator=g)
+    trans = torch.randn(T, T, generator=g)
+    mask = torch.ones(1, L, dtype=torch.long)
+    pred = ref.crf_decode(em, mask, trans)[0, :L].tolist()
+
+    def score(path):
+        s = sum(float(em[0, t, p]) for t, p in enumerate(path))
+        s += sum(float(trans[path[t], path[t + 1]]) for t in range(L - 1))
+        return s
+
+    best = max(itertools.product(range(T), repeat=L), key=score)
+    assert abs(score(pred) - score(list(best))) < 1e-5
+
+
+@given(st.integers(1, 3), st.integers(1, 6), st.integers(2, 9),
+       st.integers(1, 8), st.randoms())
+@settings(max_examples=60, deadline=None)
+def test_softlexicon_fuse_matches_numpy(B, L, V, E, rnd):
+    g = torch.Generator().manual_seed(rnd.randint(0, 2**31))
+    table = torch.randn(V, E, generator=g)
+    ids = torch.randint(0, V, (B, L, 40), generator=g)
+    weights = torch.rand(B, L, 40, generator=g)
+    out = ref.softlexicon_fuse(table, ids, weights).numpy()
+    t, i, w = table.numpy(), ids.numpy(), weights.numpy()
+    expect = np.zeros((B, L, 4 * E), dtype=np.float32)
+    for b in range(B):
+        for l in range(L):
+            for r in range(4):
+                acc = np.zeros(E, dtype=np.float32)
+                for s in range(10):
+                    k = r * 10 + s
+                    acc += w[b, l, k] * t[i[b, l, k]]
+                expect[b, l, r * E:(r + 1) * E] = acc
+    np.testing.assert_allclose(out, expect, atol=1e-5)
+
+
+@given(st.integers(2, 5), st.integers(2, 6), st.randoms())
+@settings(max_examples=40, deadline=None)
+def test_crf_masked_suffix_ignored(T, L, rnd):
+    """Positions beyond the mask must not affect the log-likelihood."""
+    g = torch.Generator().manual_seed(rnd.randint(0, 2**31))
+    em = torch.randn(1, L, T, generator=g)
+    trans = torch.randn(T, T, generator=g)
+    tags = torch.randint(0, T, (1, L), generator=g)
+    n_valid = 1 + rnd.randint(0, L - 1)
+    mask = torch.zeros(1, L, dtype=torch.long)
+    mask[0, :n_valid] = 1
+    ll = ref.crf_log_likelihood(em, tags, mask, trans)
+    em2 = em.clone()
+    em2[0, n_valid:] = 999.0          # garbage in the padded tail
+    tags2 = tags.clone()
+    tags2[0, n_valid:] = (tags2[0, n_valid:] + 1) % T
+    ll2 = ref.crf_log_likelihood(em2, tags2, mask, trans)
+    assert abs(float(ll) - float(ll2)) < 1e-4
