@@ -131,3 +131,74 @@ def test_crf_masked_suffix_ignored(T, L, rnd):
     tags2[0, n_valid:] = (tags2[0, n_valid:] + 1) % T
     ll2 = ref.crf_log_likelihood(em2, tags2, mask, trans)
     assert abs(float(ll) - float(ll2)) < 1e-4
+
+
+@given(st.integers(1, 3), st.integers(1, 8), st.integers(1, 3),
+       st.sampled_from(["tanh", "relu"]), st.randoms())
+@settings(max_examples=30, deadline=None)
+def test_bilstm_reference_matches_cell_math(B, L, h, act, rnd):
+    """Reference BiLSTM vs explicit per-step LSTM cell recursion."""
+    g = torch.Generator().manual_seed(rnd.randint(0, 2**31))
+    E = 4
+    x = torch.randn(B, L, E, generator=g)
+    mk = lambda *sh: torch.randn(*sh, generator=g) * 0.3
+    w_ih_f, w_hh_f, b_f = mk(E, 4 * h), mk(h, 4 * h), mk(4 * h)
+    w_ih_b, w_hh_b, b_b = mk(E, 4 * h), mk(h, 4 * h), mk(4 * h)
+    lens = torch.tensor([1 + rnd.randint(0, L - 1) for _ in range(B)])
+    out = ref.bilstm_forward(x, w_ih_f, w_hh_f, b_f, w_ih_b, w_hh_b, b_b,
+                             lens, act)
+
+    actf = torch.tanh if act == "tanh" else torch.relu
+
+    def run_dir(w_ih, w_hh, b, reverse):
+        hs = torch.zeros(B, L, h)
+        hstate = torch.zeros(B, h)
+        cstate = torch.zeros(B, h)
+        steps = range(L - 1, -1, -1) if reverse else range(L)
+        for t in steps:
+            gates = x[:, t] @ w_ih + hstate @ w_hh + b
+            i, f, gg, o = gates.split(h, dim=1)
+            c_new = torch.sigmoid(f) * cstate + torch.sigmoid(i) * actf(gg)
+            h_new = torch.sigmoid(o) * actf(c_new)
+            valid = (lens > t).float()[:, None]
+            cstate = valid * c_new + (1 - valid) * cstate
+            hstate = valid * h_new + (1 - valid) * hstate
+            hs[:, t] = valid * h_new
+        return hs
+
+    expect = torch.cat([run_dir(w_ih_f, w_hh_f, b_f, False),
+                        run_dir(w_ih_b, w_hh_b, b_b, True)], dim=-1)
+    torch.testing.assert_close(out, expect, atol=2e-5, rtol=1e-4)
+
+
+@st.composite
+def cjk_sentences(draw):
+    n = draw(st.integers(1, 12))
+    return "".join(chr(0x4E00 + draw(st.integers(0, 99))) for _ in range(n))
+
+
+@given(cjk_sentences())
+@settings(max_examples=100, deadline=None)
+def test_word_enhance_shape_invariants(sentence):
+    """softword/ex_softword/softlexicon outputs always align 1:1 with the
+    characters (the reference asserts this during preprocessing,
+    data/word_enhance.py:118,225-227)."""
+    from chinesener_amd.data.word_enhance import (Lexicon, build_ex_softword,
+                                                  build_soft_lexicon,
+                                                  build_softword)
+    lex = Lexicon.synthetic([chr(0x4E00 + i) for i in range(100)],
+                            n_words=200, seed=7)
+    sw = build_softword(sentence, lex)
+    assert len(sw) == len(sentence)
+    assert all(0 <= t < 5 for t in sw)
+    ex = build_ex_softword(sentence, lex)
+    assert len(ex) == len(sentence)
+    assert all(len(r) == 5 and set(r) <= {0, 1} for r in ex)
+    ids, weights = build_soft_lexicon(sentence, lex)
+    assert ids.shape == (len(sentence), 40)
+    assert weights.shape == (len(sentence), 40)
+    assert (ids >= 0).all() and (ids < len(lex)).all()
+    # weights are a distribution over the 40 slots (or all-zero rows when
+    # only <None> entries carry zero frequency)
+    sums = weights.sum(1)
+    assert ((np.abs(sums - 1.0) < 1e-5) | (sums == 0)).all()
