@@ -202,3 +202,43 @@ def test_word_enhance_shape_invariants(sentence):
     # only <None> entries carry zero frequency)
     sums = weights.sum(1)
     assert ((np.abs(sums - 1.0) < 1e-5) | (sums == 0)).all()
+
+
+@given(st.integers(1, 4), st.integers(2, 10), st.randoms())
+@settings(max_examples=50, deadline=None)
+def test_tag_metrics_matches_sklearn(B, L, rnd):
+    """TagMetrics confusion math vs sklearn on random unmasked tokens."""
+    from sklearn.metrics import precision_recall_fscore_support
+    from chinesener_amd.train.metrics import TagMetrics
+    idx2tag = {0: "[PAD]", 1: "O", 2: "B-LOC", 3: "I-LOC"}
+    g = torch.Generator().manual_seed(rnd.randint(0, 2**31))
+    labels = torch.randint(1, 4, (B, L), generator=g)
+    preds = torch.randint(1, 4, (B, L), generator=g)
+    mask = torch.ones(B, L, dtype=torch.long)
+    m = TagMetrics(4, idx2tag)
+    m.update(preds, labels, mask)
+    out = m.compute()
+    y, p = labels.reshape(-1).numpy(), preds.reshape(-1).numpy()
+    acc = (y == p).mean()
+    assert abs(out["accuracy"] - acc) < 1e-9
+    # micro over entity tags (2, 3), sklearn labels= restricted micro
+    prec, rec, f1, _ = precision_recall_fscore_support(
+        y, p, labels=[2, 3], average="micro", zero_division=0)
+    if "micro_f1" in out:
+        assert abs(out["micro_f1"] - f1) < 1e-9
+
+
+@given(st.integers(1, 3), st.integers(1, 6), st.integers(2, 5), st.randoms())
+@settings(max_examples=50, deadline=None)
+def test_masked_ce_matches_manual(B, L, T, rnd):
+    g = torch.Generator().manual_seed(rnd.randint(0, 2**31))
+    logits = torch.randn(B, L, T, generator=g)
+    labels = torch.randint(0, T, (B, L), generator=g)
+    mask = (torch.rand(B, L, generator=g) > 0.3).long()
+    if mask.sum() == 0:
+        mask[0, 0] = 1
+    loss = ref.masked_cross_entropy(logits, labels, mask)
+    logp = torch.log_softmax(logits, -1)
+    nll = -logp.gather(-1, labels[..., None]).squeeze(-1)
+    expect = (nll * mask).sum() / mask.sum()
+    assert abs(float(loss) - float(expect)) < 1e-5
