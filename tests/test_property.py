@@ -242,3 +242,65 @@ def test_masked_ce_matches_manual(B, L, T, rnd):
     nll = -logp.gather(-1, labels[..., None]).squeeze(-1)
     expect = (nll * mask).sum() / mask.sum()
     assert abs(float(loss) - float(expect)) < 1e-5
+
+
+@given(tag_sequences())
+@settings(max_examples=100, deadline=None)
+def test_chunk_roundtrip_preserves_sentence(tags):
+    """augment.chunk_by_tag -> chunks_to_bio reproduces the text and a
+    normalized tag sequence (I- continuations after O become B- starts,
+    matching strict-span semantics)."""
+    from chinesener_amd.data.augment import chunk_by_tag, chunks_to_bio
+    sent = [chr(0x4E00 + i) for i in range(len(tags))]
+    s2, t2 = chunks_to_bio(chunk_by_tag(sent, tags))
+    assert s2 == sent
+    # spans survive the round trip exactly
+    assert extract_spans(t2) == extract_spans(tags)
+
+
+@given(st.integers(1, 3), st.integers(2, 8), st.randoms())
+@settings(max_examples=50, deadline=None)
+def test_mrc_span_labels_roundtrip(B, L, rnd):
+    """mrc.span_model.make_span_labels start/end/span tensors are mutually
+    consistent with the BIO input."""
+    from chinesener_amd.mrc.span_model import make_span_labels
+    g = torch.Generator().manual_seed(rnd.randint(0, 2**31))
+    # random BIO over {0,1,2} with I only continuing something
+    labels = torch.zeros(B, L, dtype=torch.long)
+    for b in range(B):
+        t = 0
+        while t < L:
+            if rnd.random() < 0.4:
+                length = 1 + rnd.randint(0, min(3, L - t - 1))
+                labels[b, t] = 1
+                labels[b, t + 1:t + length] = 2
+                t += length
+            else:
+                t += 1
+    start, end, span = make_span_labels(labels)
+    assert (start.sum(1) == end.sum(1)).all()        # every span closes
+    assert (span.sum((1, 2)) == start.sum(1)).all()  # one cell per span
+    for b in range(B):
+        for s, e in span[b].nonzero().tolist():
+            assert s <= e
+            assert labels[b, s] == 1
+            assert (labels[b, s + 1:e + 1] == 2).all()
+            if e + 1 < L:
+                assert labels[b, e + 1] != 2
+
+
+@given(st.integers(1, 5), st.integers(1, 7),
+       st.sampled_from(["int64", "int32", "float32", "float64"]),
+       st.randoms())
+@settings(max_examples=60, deadline=None)
+def test_rpc_msgpack_ndarray_roundtrip(a, b, dtype, rnd):
+    """serve.rpc framing round-trips arbitrary arrays bit-exactly."""
+    from chinesener_amd.serve import rpc
+    arr = (np.random.default_rng(rnd.randint(0, 2**31))
+           .standard_normal((a, b)) * 100).astype(dtype)
+    msg = {"inputs": {"x": arr}, "model_spec": {"name": "m", "version": 3}}
+    out = rpc.loads(rpc.dumps(msg))
+    assert out["model_spec"] == {"name": "m", "version": 3}
+    got = out["inputs"]["x"]
+    assert got.dtype == arr.dtype and got.shape == arr.shape
+    assert (got == arr).all()
