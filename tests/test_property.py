@@ -304,3 +304,46 @@ def test_rpc_msgpack_ndarray_roundtrip(a, b, dtype, rnd):
     got = out["inputs"]["x"]
     assert got.dtype == arr.dtype and got.shape == arr.shape
     assert (got == arr).all()
+
+
+@given(st.integers(10, 600), st.floats(0.01, 0.5), st.floats(1e-6, 1e-2))
+@settings(max_examples=25, deadline=None)
+def test_lr_schedule_bert_shape_properties(total, wr, base):
+    """bert schedule: rises monotonically through warmup, peaks at the
+    warmup boundary, decays monotonically to ~0, never negative."""
+    from chinesener_amd.train.optimizers import LrSchedule
+    s = LrSchedule("bert", base, num_train_steps=total, warmup_ratio=wr)
+    lrs = [s.lr_at(t) for t in range(1, total + 1)]
+    assert all(lr >= 0 for lr in lrs)
+    w = s.warmup
+    for a, b in zip(lrs[:w - 1], lrs[1:w]):
+        assert b >= a                       # warmup non-decreasing
+    for a, b in zip(lrs[w:], lrs[w + 1:]):
+        assert b <= a + 1e-12               # decay non-increasing
+    assert abs(max(lrs) - base) <= base * (1.0 / max(1, w)) + 1e-12
+    assert lrs[-1] <= base * 0.02 + 1e-9    # ends near zero
+
+
+@given(cjk_sentences())
+@settings(max_examples=80, deadline=None)
+def test_bert_feature_invariants(sentence):
+    """build_seq_feature: [CLS] ... [SEP] framing, mask/seq_len agree,
+    label row aligned and padded (reference base_preprocess invariants,
+    :183-201)."""
+    from chinesener_amd.data.preprocess import get_instance
+    from chinesener_amd.data.tokenizer import Vocab
+    tag2idx = {"[PAD]": 0, "O": 1, "B-LOC": 2, "I-LOC": 3,
+               "[CLS]": 4, "[SEP]": 5}
+    proc = get_instance("bert", 16, tag2idx, vocab=Vocab.synthetic())
+    tok = proc.tokenizer
+    tags = ["O"] * len(sentence)
+    feat = proc.build_seq_feature(sentence, tags)
+    ids, mask, labels = feat["token_ids"], feat["mask"], feat["label_ids"]
+    n = int(feat["seq_len"])
+    assert len(ids) == len(mask) == len(labels) == 16
+    assert mask.sum() == n
+    assert (mask[:n] == 1).all() and (mask[n:] == 0).all()
+    assert ids[0] == tok.vocab.stoi["[CLS]"]
+    assert ids[n - 1] == tok.vocab.stoi["[SEP]"]
+    assert labels[0] == tag2idx["[CLS]"] and labels[n - 1] == tag2idx["[SEP]"]
+    assert (labels[n:] == tag2idx["[PAD]"]).all()
