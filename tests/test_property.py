@@ -347,3 +347,49 @@ def test_bert_feature_invariants(sentence):
     assert ids[n - 1] == tok.vocab.stoi["[SEP]"]
     assert labels[0] == tag2idx["[CLS]"] and labels[n - 1] == tag2idx["[SEP]"]
     assert (labels[n:] == tag2idx["[PAD]"]).all()
+
+
+@given(cjk_sentences(), st.sampled_from(["LOC", "PER", "ORG"]))
+@settings(max_examples=60, deadline=None)
+def test_mrc_feature_invariants(text, tag_type):
+    """MRC [CLS]+query+[SEP]+text framing: segment/text-mask boundaries,
+    label region restricted to the text, length accounting."""
+    from chinesener_amd.data.tokenizer import Vocab, WordpieceTokenizer
+    from chinesener_amd.mrc.dataset import build_single_feature
+    tok = WordpieceTokenizer(Vocab.synthetic())
+    tags = ["O"] * len(text)
+    if text:
+        tags[0] = f"B-{tag_type}"
+        for i in range(1, min(2, len(text))):
+            tags[i] = f"I-{tag_type}"
+    f = build_single_feature(tok, "find entities", text, tags, tag_type,
+                             max_seq_len=64)
+    q, t = int(f["query_len"]), int(f["text_len"])
+    n = len(f["token_ids"])
+    assert n == q + t <= 64
+    assert (f["segment_ids"][:q] == 0).all()
+    assert (f["segment_ids"][q:] == 1).all()
+    assert (f["text_mask"][:q] == 0).all()
+    assert (f["text_mask"][q:] == 1).all()
+    # labels only inside the text region
+    assert (f["label_ids"][:q] == 0).all()
+    if text:
+        assert f["label_ids"][q] == 1      # the B we planted
+
+
+@given(cjk_sentences())
+@settings(max_examples=100, deadline=None)
+def test_tokenizer_id_roundtrip(text):
+    """convert_tokens_to_ids stays in-vocab and maps known chars back."""
+    from chinesener_amd.data.tokenizer import Vocab, WordpieceTokenizer
+    tok = WordpieceTokenizer(Vocab.synthetic())
+    toks = tok.tokenize(text)
+    ids = tok.convert_tokens_to_ids(toks)
+    assert len(ids) == len(toks)
+    itos = tok.vocab.itos
+    assert all(0 <= i < len(itos) for i in ids)
+    for t, i in zip(toks, ids):
+        if t in tok.vocab.stoi:
+            assert itos[i] == t
+        else:
+            assert itos[i] == "[UNK]"
