@@ -29,6 +29,7 @@ std::vector<at::Tensor> dropout_add_ln_fwd(const at::Tensor&,
 at::Tensor mask_scale(const at::Tensor&, const at::Tensor&, double);
 void bump_counter(const at::Tensor&);
 at::Tensor wgrad(const at::Tensor&, const at::Tensor&, long);
+at::Tensor wgrad2(const at::Tensor&, const at::Tensor&, long);
 // crf.hip
 std::vector<at::Tensor> crf_fwd(const at::Tensor&, const at::Tensor&,
                                 const at::Tensor&, const at::Tensor&);
@@ -110,6 +111,7 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("mask_scale", &mask_scale);
   m.def("bump_counter", &bump_counter);
   m.def("wgrad", &wgrad);
+  m.def("wgrad2", &wgrad2);
   m.def("crf_fwd", &crf_fwd);
   m.def("crf_viterbi", &crf_viterbi);
   m.def("softlexicon_fwd", &softlexicon_fwd);
