@@ -176,3 +176,65 @@ def test_device_prefetcher_cpu_passthrough():
         assert torch.equal(b["x"], src[i]["x"])
     # empty iterator
     assert list(DevicePrefetcher(iter([]), "cpu")) == []
+
+
+def test_real_file_loaders(tmp_path):
+    """Every dataset adapter parses its real on-disk format (reference
+    data/{msra,people_daily,cluener,msr,weibo}/preprocess.py layouts) —
+    not just the synthetic fallback."""
+    import json as _json
+    from chinesener_amd.data.datasets import load_data
+
+    # msra: {split}/sentences.txt + tags.txt, space-separated
+    msra = tmp_path / "msra" / "train"
+    msra.mkdir(parents=True)
+    (msra / "sentences.txt").write_text(
+        "\u5317 \u4eac \u5f88 \u597d\n\u6211 \u7231 \u4e2d \u56fd\n",
+        encoding="utf-8")
+    (msra / "tags.txt").write_text(
+        "B-LOC I-LOC O O\nO O B-LOC I-LOC\n", encoding="utf-8")
+    sents, tags = load_data("msra", str(tmp_path / "msra"), "train")
+    assert sents == ["\u5317\u4eac\u5f88\u597d",
+                     "\u6211\u7231\u4e2d\u56fd"]
+    assert tags[0] == ["B-LOC", "I-LOC", "O", "O"]
+
+    # people_daily: CoNLL example.{split}
+    pd = tmp_path / "people_daily"
+    pd.mkdir()
+    (pd / "example.valid").write_text(
+        "\u5317 B-LOC\n\u4eac I-LOC\n\n\u597d O\n", encoding="utf-8")
+    sents, tags = load_data("people_daily", str(pd), "valid")
+    assert sents == ["\u5317\u4eac", "\u597d"]
+    assert tags == [["B-LOC", "I-LOC"], ["O"]]
+
+    # cluener: jsonl with char-span labels
+    cl = tmp_path / "cluener"
+    cl.mkdir()
+    rec = {"text": "\u5f20\u4e09\u5728\u5317\u4eac",
+           "label": {"name": {"\u5f20\u4e09": [[0, 1]]},
+                     "address": {"\u5317\u4eac": [[3, 4]]},
+                     "game": {"x": [[2, 2]]}}}    # unmapped type ignored
+    (cl / "train.json").write_text(_json.dumps(rec, ensure_ascii=False) + "\n",
+                                   encoding="utf-8")
+    sents, tags = load_data("cluener", str(cl), "train")
+    assert tags[0] == ["B-PER", "I-PER", "O", "B-LOC", "I-LOC"]
+
+    # msr CWS: conll with BIES tags
+    msr = tmp_path / "msr"
+    msr.mkdir()
+    (msr / "train.txt").write_text(
+        "\u4e2d B\n\u56fd E\n\u597d S\n\n", encoding="utf-8")
+    sents, tags = load_data("msr", str(msr), "train")
+    assert tags == [["B", "E", "S"]]
+
+    # weibo: conll {split}.txt
+    wb = tmp_path / "weibo"
+    wb.mkdir()
+    (wb / "train.txt").write_text("\u5317 B-GPE\n\u4eac I-GPE\n\n",
+                                  encoding="utf-8")
+    sents, tags = load_data("weibo", str(wb), "train")
+    assert tags == [["B-GPE", "I-GPE"]]
+
+    # missing files -> synthetic fallback still works
+    sents, tags = load_data("weibo", str(tmp_path / "nowhere"), "train")
+    assert len(sents) > 0 and len(sents) == len(tags)
