@@ -393,3 +393,23 @@ def test_tokenizer_id_roundtrip(text):
             assert itos[i] == t
         else:
             assert itos[i] == "[UNK]"
+
+
+@given(st.lists(st.integers(1, 10000), min_size=1, max_size=12,
+                unique=True), st.integers(1, 5))
+@settings(max_examples=40, deadline=None)
+def test_checkpoint_manager_invariants(steps, keep):
+    """Random save sequences: keep_max enforced, latest() is the max
+    step, restore returns it."""
+    import tempfile
+    import torch as _t
+    from chinesener_amd.train.checkpoints import CheckpointManager
+    d = tempfile.mkdtemp(prefix="ckfuzz_")
+    cm = CheckpointManager(str(d), keep_max=keep)
+    model = _t.nn.Linear(2, 2)
+    for s_ in steps:
+        cm.save(s_, model)
+    kept = [s_ for s_, _ in cm._paths()]
+    assert kept == sorted(steps)[-keep:]
+    assert cm.latest().endswith(f"ckpt-{max(steps)}.pt")
+    assert cm.restore(_t.nn.Linear(2, 2)) == max(steps)
