@@ -78,7 +78,14 @@ class DevicePrefetcher:
             raise StopIteration
         batch = self._next
         if self.use_stream:
-            torch.cuda.current_stream(self.device).wait_event(self._event)
+            cur = torch.cuda.current_stream(self.device)
+            cur.wait_event(self._event)
+            # the caching allocator attributes these tensors to the side
+            # stream; without record_stream it may hand their memory to the
+            # next prefetch while compute kernels still read them
+            for v in batch.values():
+                if isinstance(v, torch.Tensor) and v.is_cuda:
+                    v.record_stream(cur)
         self._preload()
         return batch
 
@@ -200,14 +207,20 @@ class MultiDataset:
         epochs = epochs if epochs is not None else (self.epochs if split == "train" else 1)
 
         def sample_stream(pipe, task_id):
+            passes = 0
             while True:   # repeat like reference .repeat()
-                for batch in pipe.iter_batches(split, shuffle, epochs=1, seed=seed):
+                # reshuffle each pass over the corpus (the single-task path
+                # uses seed+ep; without this every repeat replays the same
+                # order for the shorter task)
+                for batch in pipe.iter_batches(split, shuffle, epochs=1,
+                                               seed=seed + passes):
                     bsz = batch["token_ids"].shape[0]
                     L = batch["token_ids"].shape[1]
                     batch = dict(batch)
                     batch["task_ids"] = torch.full((bsz, L), task_id, dtype=torch.int64)
                     for b in range(bsz):
                         yield {k: v[b] for k, v in batch.items()}
+                passes += 1
 
         streams = [sample_stream(p, i) for i, p in enumerate(self.pipes)]
         n_steps = max(p.params["step_per_epoch"] for p in self.pipes) * 2 * epochs

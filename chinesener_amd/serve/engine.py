@@ -30,6 +30,15 @@ log = logging.getLogger("chinesener_amd.serve")
 _INT_KEYS = ("token_ids", "segment_ids", "mask", "label_ids", "task_ids",
              "softword_ids", "softlexicon_ids", "bichar_ids")
 
+# Captured-graph inputs a request may legitimately omit: an all-zeros
+# value is their semantic default (segment 0 = single-sentence BERT
+# input; padding-id word-enhance features). token_ids/mask (and task_ids
+# for mtl/adv routing) must always be supplied.
+_DEFAULTABLE_KEYS = frozenset({
+    "segment_ids", "softword_ids", "ex_softword_ids",
+    "softlexicon_ids", "softlexicon_weights", "bichar_ids",
+})
+
 
 class _CapturedGraph:
     def __init__(self, graph, static_in: Dict[str, torch.Tensor],
@@ -66,6 +75,7 @@ class InferenceEngine:
         self.use_graph = (use_graph if use_graph is not None
                           else self.device.startswith("cuda"))
         self._graphs: Dict[int, _CapturedGraph] = {}
+        self._DEFAULTABLE = _DEFAULTABLE_KEYS
         self.n_requests = 0
         self._lock = threading.Lock()
 
@@ -169,15 +179,23 @@ class InferenceEngine:
         bucket = self._bucket(batch) if self.use_graph else None
         if bucket is not None and bucket in self._graphs:
             g = self._graphs[bucket]
-            for k, t in tensors.items():
-                if k not in g.static_in:
-                    continue
-                buf = g.static_in[k]
+            # Zero EVERY captured buffer first: the graph consumes all of
+            # them, and a key absent from this request must not silently
+            # retain the previous request's data (cross-request leakage).
+            missing = [k for k in g.static_in
+                       if k not in tensors and k not in self._DEFAULTABLE]
+            if missing:
+                raise KeyError(
+                    f"request for model '{self.name}' is missing required "
+                    f"feature(s) {missing}; captured graph consumes "
+                    f"{sorted(g.static_in)}")
+            for k, buf in g.static_in.items():
                 buf.zero_()
+                t = tensors.get(k)
+                if t is None:
+                    continue  # defaultable key: zeros are its default
                 sl = [slice(0, s) for s in t.shape]
                 buf[tuple(sl)].copy_(t.to(self.device), non_blocking=True)
-            if "mask" not in tensors:
-                g.static_in["mask"].zero_()
             g.graph.replay()
             torch.cuda.synchronize()
             return g.static_out[:batch].cpu().numpy()

@@ -60,7 +60,8 @@ class GraphedTrainStep:
         return tuple(sorted((k, tuple(v.shape), str(v.dtype))
                             for k, v in batch.items()))
 
-    def try_capture(self, batch: Dict[str, torch.Tensor]) -> bool:
+    def try_capture(self, batch: Dict[str, torch.Tensor],
+                    step: int = 1) -> bool:
         if self.failed:
             return False
         try:
@@ -68,8 +69,20 @@ class GraphedTrainStep:
                 raise RuntimeError("optimizer has no device-lr support")
             device = next(self.model.parameters()).device
             self.opt.enable_graph_lr(device)
-            self.schedule.apply(self.opt, 1)
+            self.schedule.apply(self.opt, max(1, step))
             self.static = {k: v.clone() for k, v in batch.items()}
+            # The two warmup passes below perform REAL optimizer updates
+            # from this one batch; snapshot params so the training
+            # trajectory is unchanged by capture (the first replay then
+            # applies the genuine step for this batch).
+            params = [p for p in self.model.parameters() if p.requires_grad]
+            snap = [p.detach().clone() for p in params]
+            state_snap = {}
+            for p in params:
+                st = self.opt.state.get(p)
+                if st:  # mid-training capture: preserve EMA state
+                    state_snap[p] = {k: v.clone() for k, v in st.items()
+                                     if isinstance(v, torch.Tensor)}
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
@@ -79,6 +92,24 @@ class GraphedTrainStep:
             self.graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self.graph):
                 self.static_loss = self._body()
+            # restore params + reset m/v/master so warmup updates vanish
+            # (state tensor ADDRESSES are baked into the graph's meta
+            # blobs — reset in place, never reallocate)
+            with torch.no_grad():
+                for p, s in zip(params, snap):
+                    p.copy_(s)
+                    st = self.opt.state.get(p)
+                    if not st:
+                        continue
+                    prev = state_snap.get(p)
+                    if prev:
+                        for k, v in prev.items():
+                            st[k].copy_(v)
+                    else:   # state born during warmup: pristine init
+                        st["m"].zero_()
+                        st["v"].zero_()
+                        if "master" in st:
+                            st["master"].copy_(p.detach().float())
             self.sig = self._signature(batch)
             log.info("training step captured in hipGraph")
             return True

@@ -50,17 +50,21 @@ class Trainer:
         self.use_bf16 = False  # autocast replaced by pure-bf16 weights
         self.step = self.ckpt.restore(model, self.optimizer,
                                       map_location=self.device)
-        # hipGraph-captured step (opt-in: CHINESENER_STEPGRAPH=1).
-        # Replays are verified bit-exact vs eager steps, but interleaved
-        # evaluation plus chaotic high-LR regimes (diff-lr x500 groups)
-        # showed instability in long runs, so real training defaults to
-        # eager; bench.py (short, no evals, warmup-phase LRs) uses graphs
-        # by default.
+        # hipGraph-captured step (default ON on GPU; CHINESENER_STEPGRAPH=0
+        # disables). Replays are verified bit-exact vs eager steps. Capture
+        # is DEFERRED until the diff-LR warm ramp finishes (schedule.warmup)
+        # when x500-style multiplier groups exist: the fused kernel bakes
+        # each group's full lr_scale_base into its cached meta blob, so a
+        # pre-warmup capture would bypass the ramp that prevents all-O-basin
+        # divergence (optimizers.py LrSchedule.apply).
         self._graph: GraphedTrainStep | None = None
         self._last_sig = None
         self.use_step_graph = (self.device.startswith("cuda")
                                and os.environ.get("CHINESENER_STEPGRAPH")
-                               == "1")
+                               != "0")
+        self._has_diff_lr = any(
+            g.get("lr_scale_base", 1.0) != 1.0
+            for g in self.optimizer.param_groups)
         # layer activation summaries (reference add_layer_summary,
         # tools/utils.py:25-27): opt-in via params["verbose"] — JSONL in
         # the checkpoint dir instead of TensorBoard events
@@ -86,7 +90,9 @@ class Trainer:
     def train_step(self, batch: Dict[str, torch.Tensor]) -> float:
         batch = {k: v.to(self.device, non_blocking=True)
                  for k, v in batch.items()}
-        if self.use_step_graph:
+        graph_ready = (not self._has_diff_lr
+                       or self.step >= self.schedule.warmup)
+        if self.use_step_graph and graph_ready:
             if self._graph is not None and self._graph.matches(batch):
                 self.step += 1
                 loss = self._graph.replay(batch, self.step)
@@ -97,7 +103,7 @@ class Trainer:
                     self.model, self.optimizer, self.schedule,
                     lambda m: clip_gradients(m, self.family),
                     dp=self.dp, cast=self._cast)
-                if g.try_capture(batch):
+                if g.try_capture(batch, step=self.step + 1):
                     self._graph = g
                     self.step += 1
                     return float(g.replay(batch, self.step).detach())
