@@ -24,11 +24,18 @@ import time
 # every run tunes fresh — ~1 min of warmup cost, ~10% steady-state gain.
 # CHINESENER_NO_TUNABLE=1 opts out.
 if os.environ.get("CHINESENER_NO_TUNABLE") != "1":
+    _rank_tag = os.environ.get("RANK", "0")
     os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
     os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
-    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", "100")
+    # multi-rank: bound per-solution tuning time harder (ranks tune the
+    # same shapes in lockstep between collectives; the slowest gates all)
+    _dur = "100" if os.environ.get("WORLD_SIZE", "1") == "1" else "30"
+    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", _dur)
+    # per-rank filename: concurrent ranks must not clobber one CSV
     os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
-                          "/tmp/chinesener_tunableop_.csv")
+                          f"/tmp/chinesener_tunableop_r{_rank_tag}_.csv")
+# a dead peer must raise in surviving ranks, not hang the driver window
+os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
 
 import torch
 
@@ -59,15 +66,21 @@ def main():
               f"measuring 1 GPU and reporting n_gpus=1", file=sys.stderr)
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     use_gpu = torch.cuda.is_available()
-    device = f"cuda:{local_rank}" if use_gpu else "cpu"
+    # modulo lets a 2-rank RCCL smoke run on a 1-GPU box (oversubscribed)
+    dev_idx = local_rank % max(1, torch.cuda.device_count()) if use_gpu else 0
+    device = f"cuda:{dev_idx}" if use_gpu else "cpu"
     dist = None
     if world > 1:
+        from datetime import timedelta
         import torch.distributed as dist_mod
         dist = dist_mod
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         if use_gpu:
-            torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl" if use_gpu else "gloo")
+            torch.cuda.set_device(dev_idx)
+        # bounded collective timeout: a desynced rank fails the job fast
+        # with a diagnostic instead of eating the driver's SCALE window
+        dist.init_process_group("nccl" if use_gpu else "gloo",
+                                timeout=timedelta(minutes=5))
 
     torch.manual_seed(1234 + rank)
     # MSRA-shaped config: vocab 21128, 10 BIO labels, BERT-base encoder

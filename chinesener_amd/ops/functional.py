@@ -124,6 +124,13 @@ def add_layernorm(x, residual, weight, bias, eps: float = 1e-12):
 
 
 # ---------------------------------------------------------- fused linear
+def _use_wgrad2() -> bool:
+    """Opt-in switch for the in-tree MFMA wgrad kernel (wgrad2) on the
+    dW GEMMs. Off by default until the hardware microbenchmark shows it
+    at parity with hipBLASLt (scripts/bench_wgrad2.py)."""
+    return os.environ.get("CHINESENER_WGRAD2") == "1"
+
+
 class _LinearFn(torch.autograd.Function):
     """nn.Linear math with a custom column-sum bias grad (torch's generic
     reduce is ~4.5x off memory-bound for [tokens, features] dbias)."""
@@ -138,10 +145,16 @@ class _LinearFn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         K = x.shape[-1]
         N = w.shape[0]
-        dyf = dy.reshape(-1, N)
+        dyf = dy.reshape(-1, N).contiguous()
         dx = (dyf @ w.to(dy.dtype)).reshape(x.shape)
-        dw = (dyf.T @ x.reshape(-1, K)).to(w.dtype)
-        db = get_ext().colsum(dyf.contiguous())
+        x2 = x.reshape(-1, K)
+        if (_use_wgrad2() and dy.dtype == torch.bfloat16
+                and x.dtype == torch.bfloat16 and N % 8 == 0 and K % 8 == 0):
+            # in-tree MFMA split-K wgrad (csrc/wgrad.hip wgrad2)
+            dw = get_ext().wgrad2(dyf, x2.contiguous(), 0).to(w.dtype)
+        else:
+            dw = (dyf.T @ x2).to(w.dtype)
+        db = get_ext().colsum(dyf)
         return dx, dw, db
 
 
