@@ -78,9 +78,12 @@ def main():
         if use_gpu:
             torch.cuda.set_device(dev_idx)
         # bounded collective timeout: a desynced rank fails the job fast
-        # with a diagnostic instead of eating the driver's SCALE window
-        dist.init_process_group("nccl" if use_gpu else "gloo",
-                                timeout=timedelta(minutes=5))
+        # with a diagnostic instead of eating the driver's SCALE window.
+        # CHINESENER_DP_BACKEND=gloo lets a multi-rank smoke share one
+        # GPU (RCCL refuses two ranks on one device).
+        backend = os.environ.get("CHINESENER_DP_BACKEND") or (
+            "nccl" if use_gpu else "gloo")
+        dist.init_process_group(backend, timeout=timedelta(minutes=5))
 
     torch.manual_seed(1234 + rank)
     # MSRA-shaped config: vocab 21128, 10 BIO labels, BERT-base encoder

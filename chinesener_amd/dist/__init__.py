@@ -28,14 +28,16 @@ def init_process_group(backend: Optional[str] = None,
     if dist.is_initialized():
         return dist.get_rank(), dist.get_world_size()
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = os.environ.get("CHINESENER_DP_BACKEND") or (
+            "nccl" if torch.cuda.is_available() else "gloo")
     # 1 = TearDown: abort RCCL communicators + kill the process on error
     os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     if torch.cuda.is_available():
-        torch.cuda.set_device(local_rank)
+        # modulo: a gloo-backend multi-rank smoke may share one GPU
+        torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
     from datetime import timedelta

@@ -9,6 +9,14 @@ import torch.distributed as dist
 
 
 def main():
+    # standalone world=1 mode: RCCL refuses >1 rank per GPU (verified:
+    # "Duplicate GPU detected" on a 1-GPU box), so single-rank RCCL is
+    # the deepest RCCL validation a 1-GPU lease allows — it still runs
+    # real ncclAllReduce kernels and their hipGraph capture.
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("WORLD_SIZE", "1")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29617")
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local = int(os.environ.get("LOCAL_RANK", rank))
