@@ -149,9 +149,14 @@ def main():
                   f"tunable enabled={_tun.is_enabled()} "
                   f"results={len(_tun.get_results())}", file=sys.stderr)
 
-    # hipGraph-capture the whole training step (see train/graph_step.py)
+    # hipGraph-capture the whole training step (see train/graph_step.py).
+    # Gated on an RCCL backend when DP is active: gloo collectives
+    # invalidate the capture AND leave the stream wedged in capture state
+    # (verified on hardware) — RCCL supports capture.
     graphed = None
-    if use_gpu and os.environ.get("CHINESENER_NO_STEPGRAPH") != "1":
+    capture_safe = dist is None or dist.get_backend() == "nccl"
+    if (use_gpu and capture_safe
+            and os.environ.get("CHINESENER_NO_STEPGRAPH") != "1"):
         import logging
         logging.basicConfig(level=logging.INFO, stream=sys.stderr)
         from chinesener_amd.train.graph_step import GraphedTrainStep

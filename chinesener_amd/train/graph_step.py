@@ -118,6 +118,16 @@ class GraphedTrainStep:
             self.opt.lr_dev = None
             self.graph = None
             self.failed = True
+            # A failed capture can leave the stream wedged in capture
+            # state (any later CUDA op then fails). Probe and fail loudly
+            # rather than letting every subsequent step error obscurely.
+            try:
+                torch.cuda.synchronize()
+            except Exception as e2:
+                raise RuntimeError(
+                    "stream left in invalid capture state after failed "
+                    f"hipGraph capture: {e2}; set CHINESENER_NO_STEPGRAPH=1 "
+                    "(or CHINESENER_STEPGRAPH=0 for Trainer)") from e
             return False
 
     def matches(self, batch) -> bool:

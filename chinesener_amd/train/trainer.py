@@ -92,6 +92,12 @@ class Trainer:
                  for k, v in batch.items()}
         graph_ready = (not self._has_diff_lr
                        or self.step >= self.schedule.warmup)
+        if self.dp is not None:
+            # capture is only safe over RCCL (gloo collectives wedge the
+            # stream's capture state — see bench.py)
+            import torch.distributed as _dist
+            if _dist.is_initialized() and _dist.get_backend() != "nccl":
+                graph_ready = False
         if self.use_step_graph and graph_ready:
             if self._graph is not None and self._graph.matches(batch):
                 self.step += 1
