@@ -122,7 +122,8 @@ class GraphedTrainStep:
             # state (any later CUDA op then fails). Probe and fail loudly
             # rather than letting every subsequent step error obscurely.
             try:
-                torch.cuda.synchronize()
+                if torch.cuda.is_available():
+                    torch.cuda.synchronize()
             except Exception as e2:
                 raise RuntimeError(
                     "stream left in invalid capture state after failed "
