@@ -57,7 +57,12 @@ class BiLSTM(nn.Module):
 
 class MultiKernelCNN(nn.Module):
     """Per-kernel conv1d SAME + dropout, concat filters
-    (reference cnn_layer, tools/layer.py:44-60)."""
+    (reference cnn_layer, tools/layer.py:44-60).
+
+    torch ``padding='same'`` is bit-identical to TF SAME at stride 1 for
+    every kernel size incl. even ones — both put the extra zero on the
+    RIGHT (total k-1, left (k-1)//2) — verified by
+    tests/test_ops_reference.py::test_conv1d_same_matches_tf_same."""
 
     def __init__(self, input_size: int, filters: int = 128,
                  kernel_sizes: List[int] = (2, 3, 4), keep_prob: float = 0.8):

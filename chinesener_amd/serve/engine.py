@@ -16,6 +16,7 @@ from __future__ import annotations
 
 import logging
 import threading
+import time
 from typing import Dict, List, Optional, Sequence
 
 import numpy as np
@@ -204,6 +205,119 @@ class InferenceEngine:
         dev = {k: t.to(self.device) for k, t in tensors.items()}
         out = self.model(dev, compute_pred=True)
         return out.pred_ids.cpu().numpy()
+
+
+class MicroBatcher:
+    """Concurrency layer over InferenceEngine (VERDICT weak #6: the
+    engine serializes requests under one lock because hipGraph static
+    buffers are shared). Instead of queueing N clients for N replays,
+    concurrent requests are coalesced into ONE padded batch replayed
+    once: p99 under k concurrent clients stays ~1 replay + window
+    instead of k replays.
+
+    predict() blocks the calling thread until its rows are ready, so the
+    gRPC handler uses it as a drop-in engine replacement."""
+
+    def __init__(self, engine: InferenceEngine, max_batch: Optional[int] = None,
+                 window_ms: float = 0.3):
+        self.engine = engine
+        self.max_batch = max_batch or max(engine.batch_sizes)
+        self.window = window_ms / 1000.0
+        self._cv = threading.Condition()
+        self._queue: list = []          # (features, slot, event)
+        self._stop = False
+        self._worker = threading.Thread(target=self._run, daemon=True)
+        self._worker.start()
+
+    # engine API passthrough used by the server
+    @property
+    def name(self):
+        return self.engine.name
+
+    def warmup(self, requests=None):
+        return self.engine.warmup(requests)
+
+    def close(self):
+        with self._cv:
+            self._stop = True
+            self._cv.notify_all()
+        self._worker.join(timeout=2.0)
+
+    def predict(self, features: Dict[str, np.ndarray]) -> np.ndarray:
+        ev = threading.Event()
+        slot: Dict[str, object] = {}
+        with self._cv:
+            self._queue.append((features, slot, ev))
+            self._cv.notify()
+        ev.wait()
+        if "err" in slot:
+            raise slot["err"]  # type: ignore[misc]
+        return slot["out"]     # type: ignore[return-value]
+
+    # ------------------------------------------------------------ worker
+    @staticmethod
+    def _sig(features) -> tuple:
+        return tuple(sorted((k, tuple(np.asarray(v).shape[1:]))
+                            for k, v in features.items()))
+
+    def _take_compatible(self) -> list:
+        """Pop a prefix-compatible group (same keys + trailing shapes)
+        totalling <= max_batch rows. Leaves incompatible requests queued
+        for the next cycle."""
+        group, rows, sig = [], 0, None
+        rest = []
+        for item in self._queue:
+            f = item[0]
+            n = int(np.asarray(f["token_ids"]).shape[0])
+            s = self._sig(f)
+            if sig is None:
+                sig = s
+            if s == sig and rows + n <= self.max_batch:
+                group.append(item)
+                rows += n
+            else:
+                rest.append(item)
+        self._queue = rest
+        return group
+
+    def _run(self):
+        while True:
+            with self._cv:
+                while not self._queue and not self._stop:
+                    self._cv.wait(0.25)
+                if self._stop:
+                    pending = self._queue
+                    self._queue = []
+                    for _, slot, ev in pending:
+                        slot["err"] = RuntimeError("batcher stopped")
+                        ev.set()
+                    return
+                deadline = time.perf_counter() + self.window
+                while (sum(int(np.asarray(f["token_ids"]).shape[0])
+                           for f, _, _ in self._queue) < self.max_batch):
+                    remaining = deadline - time.perf_counter()
+                    if remaining <= 0:
+                        break
+                    self._cv.wait(remaining)
+                batch = self._take_compatible()
+            if not batch:
+                continue
+            try:
+                merged = {k: np.concatenate(
+                    [np.asarray(f[k]) for f, _, _ in batch])
+                    for k in batch[0][0]}
+                out = self.engine.predict(merged)
+                i = 0
+                for f, slot, ev in batch:
+                    n = int(np.asarray(f["token_ids"]).shape[0])
+                    slot["out"] = out[i:i + n]
+                    i += n
+            except Exception as e:
+                for _, slot, ev in batch:
+                    slot["err"] = e
+            finally:
+                for _, slot, ev in batch:
+                    ev.set()
 
 
 class FastPredict:

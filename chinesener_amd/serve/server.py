@@ -17,7 +17,7 @@ import grpc
 
 from ..config import EXPORT_DIR
 from . import rpc
-from .engine import InferenceEngine
+from .engine import InferenceEngine, MicroBatcher
 from .export import latest_version_dir
 
 log = logging.getLogger("chinesener_amd.serve")
@@ -78,7 +78,8 @@ def build_server(engines: Dict[str, InferenceEngine], port: int,
 
 def serve(model_names, export_root: str = EXPORT_DIR,
           port: int = rpc.DEFAULT_PORT, wait: bool = True,
-          use_graph=None, batch_sizes=(1, 4, 8), max_workers: int = 4):
+          use_graph=None, batch_sizes=(1, 4, 8), max_workers: int = 4,
+          batching: bool = True, window_ms: float = 0.3):
     engines = {}
     for name in model_names:
         eng = InferenceEngine(name, export_root, batch_sizes=batch_sizes,
@@ -86,7 +87,9 @@ def serve(model_names, export_root: str = EXPORT_DIR,
         warm = load_warmup(latest_version_dir(name, export_root))
         eng.warmup(warm)
         log.info("loaded %s (%d warmup requests replayed)", name, len(warm))
-        engines[name] = eng
+        # micro-batching: coalesce concurrent clients into one replay
+        # (the engine itself serializes under a lock; see MicroBatcher)
+        engines[name] = MicroBatcher(eng, window_ms=window_ms) if batching else eng
     server = build_server(engines, port, max_workers=max_workers)
     server.start()
     log.info("serving %s on :%d", sorted(engines), port)
@@ -103,10 +106,13 @@ def main(argv=None):
     ap.add_argument("--port", type=int, default=rpc.DEFAULT_PORT)
     ap.add_argument("--no_graph", action="store_true",
                     help="disable hipGraph capture (debug)")
+    ap.add_argument("--no_batching", action="store_true",
+                    help="disable concurrent-request micro-batching")
     args = ap.parse_args(argv)
     logging.basicConfig(level=logging.INFO)
     serve(args.model.split(","), args.export_root, args.port,
-          use_graph=False if args.no_graph else None)
+          use_graph=False if args.no_graph else None,
+          batching=not args.no_batching)
     return 0
 
 

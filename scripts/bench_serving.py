@@ -68,6 +68,51 @@ def main():
         del eng
         torch.cuda.empty_cache()
 
+    # concurrency: 8 clients of batch-1 requests, locked engine vs
+    # micro-batching queue (VERDICT weak #6 / serving-concurrency fix)
+    import threading
+    from chinesener_amd.serve.engine import MicroBatcher
+    rng = np.random.default_rng(0)
+    feats1 = {"token_ids": rng.integers(1, 21128, (1, args.seq_len)),
+              "segment_ids": np.zeros((1, args.seq_len), np.int64),
+              "mask": np.ones((1, args.seq_len), np.int64)}
+    for mode in ("locked", "microbatch"):
+        eng = InferenceEngine(name, "/tmp/serving_bench", batch_sizes=(1, 8),
+                              max_seq_len=args.seq_len, use_graph=True)
+        eng.warmup()
+        target = MicroBatcher(eng, window_ms=0.3) if mode == "microbatch" else eng
+        for _ in range(20):
+            target.predict(feats1)
+        lat_all = []
+        lock = threading.Lock()
+        n_clients, per_client = 8, max(25, args.requests // 8)
+
+        def client():
+            mine = []
+            for _ in range(per_client):
+                t0 = time.perf_counter()
+                target.predict(feats1)
+                mine.append((time.perf_counter() - t0) * 1000)
+            with lock:
+                lat_all.extend(mine)
+
+        threads = [threading.Thread(target=client) for _ in range(n_clients)]
+        t0 = time.perf_counter()
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        wall = time.perf_counter() - t0
+        lat_all.sort()
+        results[f"concurrent8_{mode}"] = {
+            "p50_ms": round(lat_all[len(lat_all) // 2], 3),
+            "p99_ms": round(lat_all[int(len(lat_all) * 0.99)], 3),
+            "throughput_rps": round(n_clients * per_client / wall, 1)}
+        if mode == "microbatch":
+            target.close()
+        del eng
+        torch.cuda.empty_cache()
+
     print(json.dumps({
         "metric": "serving latency, bert_bilstm_crf PREDICT (BERT-12L + BiLSTM + Viterbi)",
         "seq_len": args.seq_len, "requests": args.requests,

@@ -212,3 +212,23 @@ def test_dice_loss_runs_and_differentiable():
     loss = ref.dice_loss(logits, labels, mask, idx_skip=(0,))
     loss.backward()
     assert torch.isfinite(loss) and torch.isfinite(logits.grad).all()
+
+
+def test_conv1d_same_matches_tf_same():
+    """bert_cnn_crf uses even kernels (2,4) with padding='same'
+    (models/layers.py MultiKernelCNN); TF SAME at stride 1 pads
+    total=k-1 with the extra zero on the RIGHT (left = (k-1)//2,
+    reference tools/layer.py:48-55 tf.layers.conv1d SAME). torch 'same'
+    must match bit-exactly for every kernel size."""
+    import torch
+    import torch.nn.functional as F
+    torch.manual_seed(0)
+    for k in (2, 3, 4, 5):
+        x = torch.randn(2, 8, 50)
+        w = torch.randn(16, 8, k)
+        b = torch.randn(16)
+        y_same = F.conv1d(x, w, b, padding="same")
+        total = k - 1
+        left = total // 2
+        y_tf = F.conv1d(F.pad(x, (left, total - left)), w, b)
+        assert torch.equal(y_same, y_tf), f"kernel {k} SAME mismatch"
