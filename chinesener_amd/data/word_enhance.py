@@ -3,10 +3,19 @@
 Re-implements the four feature families of the reference's
 data/word_enhance.py (Soft2Idx :18-24, build_softword :235-259,
 build_ex_softword :262-299, build_soft_lexicon :302-337,
-postproc_soft_lexicon :163-205) without the jieba/gensim dependencies:
-segmentation is trie-based forward maximum matching over the same
-lexicon, and lexicon vectors come from an ``npz`` file or a deterministic
-synthetic builder (no network in this environment).
+postproc_soft_lexicon :163-205) without the jieba/gensim dependencies.
+
+Segmentation for softword: the reference segments with ``jieba.cut``
+(word_enhance.py:244). jieba's core algorithm is a prefix-dictionary DAG
+with a max-log-probability route DP over word frequencies, plus an HMM
+pass over leftover single-char runs. ``Lexicon.max_prob_segment``
+re-implements the DAG + max-probability route exactly (= jieba
+``cut(HMM=False)`` semantics) over this lexicon's frequencies; the HMM
+OOV pass is intentionally absent (it requires jieba's trained emission
+tables, unavailable offline — the delta is measured in
+profiles/segmentation_delta_r02.md). Forward maximum matching remains
+available as ``segmenter="maxmatch"``. Lexicon vectors come from an
+``npz`` file or a deterministic synthetic builder (no network).
 """
 from __future__ import annotations
 
@@ -45,6 +54,42 @@ class Lexicon:
     def __post_init__(self):
         self.word2idx = {w: i for i, w in enumerate(self.words)}
         self.trie = Trie(w for w in self.words if len(w) >= 2)
+        self._logtotal = float(np.log(max(float(self.freq.sum()), 1.0)))
+
+    def _word_logp(self, w: str) -> float:
+        """log p(word) with jieba's OOV convention: unseen words count
+        frequency 1 (jieba calc(): log(FREQ.get(word) or 1) - logtotal)."""
+        i = self.word2idx.get(w)
+        f = float(self.freq[i]) if i is not None else 1.0
+        return float(np.log(max(f, 1.0))) - self._logtotal
+
+    def max_prob_segment(self, text: str,
+                         max_len: int = MAX_WORD_LEN
+                         ) -> List[Tuple[int, int]]:
+        """jieba ``cut(HMM=False)`` core: prefix-dict DAG + right-to-left
+        max-log-probability route DP. Ties prefer the longer word (jieba
+        max() over (score, end) tuples). Returns [(start, end)) spans."""
+        n = len(text)
+        route = [0.0] * (n + 1)
+        nxt = [0] * n
+        for i in range(n - 1, -1, -1):
+            best_score, best_j = None, i + 1
+            cands = self.trie.prefixes(text, i, max_len)
+            cands.append(text[i])          # single char (in-dict or OOV)
+            for w in cands:
+                j = i + len(w)
+                score = self._word_logp(w) + route[j]
+                if (best_score is None or score > best_score
+                        or (score == best_score and j > best_j)):
+                    best_score, best_j = score, j
+            route[i] = best_score
+            nxt[i] = best_j
+        spans: List[Tuple[int, int]] = []
+        i = 0
+        while i < n:
+            spans.append((i, nxt[i]))
+            i = nxt[i]
+        return spans
 
     @property
     def pad_id(self) -> int:
@@ -87,13 +132,21 @@ class Lexicon:
         return cls(words, freq, emb)
 
 
-def build_softword(sentence: str, lexicon: Lexicon) -> List[int]:
+def build_softword(sentence: str, lexicon: Lexicon,
+                   segmenter: str = "maxprob") -> List[int]:
     """Per-char BMES segmentation ids (reference build_softword :235-259).
 
-    Uses trie max-matching in place of jieba; single chars -> S.
+    segmenter="maxprob" (default) replicates jieba's DAG +
+    max-probability route (the reference's jieba.cut semantics minus the
+    HMM OOV pass); "maxmatch" keeps forward maximum matching. Single
+    chars -> S.
     """
+    if segmenter == "maxprob":
+        spans = lexicon.max_prob_segment(sentence, MAX_WORD_LEN)
+    else:
+        spans = lexicon.trie.max_match_segment(sentence, MAX_WORD_LEN)
     ids: List[int] = [0] * len(sentence)
-    for start, end in lexicon.trie.max_match_segment(sentence, MAX_WORD_LEN):
+    for start, end in spans:
         if end - start == 1:
             ids[start] = SOFT2IDX["S"]
         else:

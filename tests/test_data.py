@@ -238,3 +238,34 @@ def test_real_file_loaders(tmp_path):
     # missing files -> synthetic fallback still works
     sents, tags = load_data("weibo", str(tmp_path / "nowhere"), "train")
     assert len(sents) > 0 and len(sents) == len(tags)
+
+
+def test_max_prob_segment_vs_max_match():
+    """The maxprob segmenter follows jieba's DAG + max-log-prob route:
+    where forward max-match greedily grabs the longest prefix, maxprob
+    picks the higher-frequency path (reference jieba.cut semantics,
+    word_enhance.py:244)."""
+    import numpy as np
+    from chinesener_amd.data.word_enhance import Lexicon, build_softword
+
+    words = ["<PAD>", "<None>", "<eos>", "ab", "abc", "cd"]
+    freq = np.array([0, 0, 0, 1000.0, 2.0, 1000.0])
+    emb = np.ones((len(words), 4), dtype=np.float32)
+    lex = Lexicon(words, freq, emb)
+
+    # FMM grabs 'abc' then 'd'; maxprob prefers ab|cd (2 common words)
+    assert lex.trie.max_match_segment("abcd") == [(0, 3), (3, 4)]
+    assert lex.max_prob_segment("abcd") == [(0, 2), (2, 4)]
+
+    # softword ids follow the chosen segmenter
+    assert build_softword("abcd", lex, segmenter="maxmatch") == [1, 2, 3, 4]
+    assert build_softword("abcd", lex, segmenter="maxprob") == [1, 3, 1, 3]
+
+    # tie on score prefers the longer word (jieba max((score, end)))
+    words2 = ["<PAD>", "<None>", "<eos>", "xy", "xyz"]
+    freq2 = np.array([0, 0, 0, 5.0, 5.0])
+    lex2 = Lexicon(words2, freq2, np.ones((5, 4), np.float32))
+    assert lex2.max_prob_segment("xyz") == [(0, 3)]
+
+    # OOV text degrades to single chars (freq-1 convention)
+    assert lex.max_prob_segment("zz") == [(0, 1), (1, 2)]
