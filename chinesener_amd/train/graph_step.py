@@ -20,6 +20,7 @@ captured ones.
 from __future__ import annotations
 
 import logging
+import os
 from typing import Callable, Dict, Optional
 
 import torch
@@ -75,14 +76,16 @@ class GraphedTrainStep:
             # from this one batch; snapshot params so the training
             # trajectory is unchanged by capture (the first replay then
             # applies the genuine step for this batch).
+            do_restore = os.environ.get("CHINESENER_CAPTURE_NO_RESTORE") != "1"
             params = [p for p in self.model.parameters() if p.requires_grad]
-            snap = [p.detach().clone() for p in params]
+            snap = [p.detach().clone() for p in params] if do_restore else []
             state_snap = {}
-            for p in params:
-                st = self.opt.state.get(p)
-                if st:  # mid-training capture: preserve EMA state
-                    state_snap[p] = {k: v.clone() for k, v in st.items()
-                                     if isinstance(v, torch.Tensor)}
+            if do_restore:
+                for p in params:
+                    st = self.opt.state.get(p)
+                    if st:  # mid-training capture: preserve EMA state
+                        state_snap[p] = {k: v.clone() for k, v in st.items()
+                                         if isinstance(v, torch.Tensor)}
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
@@ -95,21 +98,22 @@ class GraphedTrainStep:
             # restore params + reset m/v/master so warmup updates vanish
             # (state tensor ADDRESSES are baked into the graph's meta
             # blobs — reset in place, never reallocate)
-            with torch.no_grad():
-                for p, s in zip(params, snap):
-                    p.copy_(s)
-                    st = self.opt.state.get(p)
-                    if not st:
-                        continue
-                    prev = state_snap.get(p)
-                    if prev:
-                        for k, v in prev.items():
-                            st[k].copy_(v)
-                    else:   # state born during warmup: pristine init
-                        st["m"].zero_()
-                        st["v"].zero_()
-                        if "master" in st:
-                            st["master"].copy_(p.detach().float())
+            if do_restore:
+                with torch.no_grad():
+                    for p, s in zip(params, snap):
+                        p.copy_(s)
+                        st = self.opt.state.get(p)
+                        if not st:
+                            continue
+                        prev = state_snap.get(p)
+                        if prev:
+                            for k, v in prev.items():
+                                st[k].copy_(v)
+                        else:   # state born during warmup: pristine init
+                            st["m"].zero_()
+                            st["v"].zero_()
+                            if "master" in st:
+                                st["master"].copy_(p.detach().float())
             self.sig = self._signature(batch)
             log.info("training step captured in hipGraph")
             return True
