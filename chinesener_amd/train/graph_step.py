@@ -135,6 +135,28 @@ class GraphedTrainStep:
                     "(or CHINESENER_STEPGRAPH=0 for Trainer)") from e
             return False
 
+    def release(self):
+        """Drop the captured graph and restore eager-optimizer state.
+
+        Replays are only safe while the allocator sees no FOREIGN
+        activity: differently-shaped allocations between replays (an
+        eval pass, even plain torch.randn churn) corrupt subsequent
+        replays on this stack within a few steps (NaN loss or
+        HSA_STATUS_ERROR_MEMORY_APERTURE_VIOLATION — reproduced and
+        bisected on MI355X, scripts/debug_trainer_graph.py arms
+        alloc/emptyalloc/rngonly). The Trainer therefore releases the
+        graph at every eval/predict/checkpoint boundary and re-captures
+        on the next repeated-shape step (~1 s per boundary)."""
+        self.graph = None
+        self.static = {}
+        self.static_loss = None
+        self.sig = None
+        self.opt.lr_dev = None
+        if hasattr(self.opt, "_meta_cache"):
+            self.opt._meta_cache = {}
+        if hasattr(self.model, "_clip_meta"):
+            self.model._clip_meta = None
+
     def matches(self, batch) -> bool:
         return self.graph is not None and self.sig == self._signature(batch)
 

@@ -74,6 +74,16 @@ class Trainer:
             self.summary = SummaryLogger(
                 model, ckpt_dir, every=params.get("summary_steps", 10))
 
+    def _drop_graph(self):
+        """Release the captured step graph before any foreign GPU work
+        (eval, predict, checkpoint): allocator activity between replays
+        corrupts them on this stack (see GraphedTrainStep.release).
+        Re-capture happens automatically on the next repeated-shape
+        train step."""
+        if self._graph is not None:
+            self._graph.release()
+            self._graph = None
+
     # ------------------------------------------------------------- train
     def _cast(self, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
         if not self.pure_bf16:
@@ -160,6 +170,10 @@ class Trainer:
                          self.step, sum(losses) / len(losses),
                          self.schedule.lr_at(self.step), sps)
                 losses, t0, last_log_step = [], time.time(), self.step
+            if save_steps and self.step % save_steps == 0:
+                # ALL ranks drop together: re-capture's warmup passes
+                # issue DP collectives, which must stay rank-aligned
+                self._drop_graph()
             if save_steps and self.step % save_steps == 0 and self.rank == 0:
                 self.ckpt.save(self.step, self.model, self.optimizer)
                 if eval_fn is not None:
@@ -183,6 +197,7 @@ class Trainer:
     def evaluate(self, batches: Iterable[Dict[str, torch.Tensor]],
                  idx2tag: Optional[Dict[int, str]] = None,
                  label_size: Optional[int] = None) -> Dict[str, float]:
+        self._drop_graph()
         self.model.eval()
         metrics = TagMetrics(label_size or self.params.get("label_size", 10),
                              idx2tag or self.params.get("idx2tag"))
@@ -208,6 +223,7 @@ class Trainer:
     def predict(self, batches: Iterable[Dict[str, torch.Tensor]]):
         """Returns list of per-sample dicts (pred_ids, label_ids, mask) —
         the reference pickles the Estimator.predict list (main.py:52-55)."""
+        self._drop_graph()
         self.model.eval()
         out_rows = []
         for batch in batches:
