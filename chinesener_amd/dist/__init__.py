@@ -146,6 +146,35 @@ class BucketedDataParallel:
             b.ready = 0
             b.work = None
 
+    def reduce_in_graph(self):
+        """Bucket fill + all-reduce + averaged writeback in one explicit
+        pass — used by the hipGraph-captured step body, whose
+        autograd.grad gradient path fires no post-accumulate hooks
+        (train/graph_step.py). All reduces are launched async first so
+        RCCL can pipeline buckets; writeback follows in launch order."""
+        for b in self.buckets:
+            for i, p in enumerate(b.params):
+                off = b.offsets[i]
+                n = p.numel()
+                if p.grad is not None:
+                    b.flat[off:off + n].copy_(p.grad.detach().reshape(-1))
+                else:
+                    b.flat[off:off + n].zero_()
+            b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                     group=self.group, async_op=True)
+        inv = 1.0 / self.world_size
+        for b in self.buckets:
+            b.work.wait()
+            b.flat.mul_(inv)
+            for i, p in enumerate(b.params):
+                if p.grad is None:
+                    continue
+                off = b.offsets[i]
+                p.grad.detach().copy_(
+                    b.flat[off:off + p.numel()].view_as(p.grad))
+            b.ready = 0
+            b.work = None
+
     def zero_grad(self):
         for b in self.buckets:
             b.flat.zero_()

@@ -42,17 +42,36 @@ class GraphedTrainStep:
         self.static_loss = None
         self.sig = None
         self.failed = False
+        self._params = [p for p in model.parameters() if p.requires_grad]
 
     def _body(self):
-        self.opt.zero_grad(set_to_none=False)
-        if self.dp is not None:
-            for b in self.dp.buckets:
-                b.flat.zero_()
-                b.ready = 0
+        """One whole training step, capture-safe.
+
+        Gradients come from ``torch.autograd.grad`` — NOT ``backward()``:
+        a captured ``backward()`` goes through AccumulateGrad nodes and
+        the resulting graph corrupts within a few replays once ANY
+        foreign allocator/stream activity happens between replays (NaN
+        loss or HSA aperture faults; reproduced and bisected on MI355X,
+        scripts/debug_capture_bisect.py: backward-capture arms corrupt,
+        autograd.grad arms stay clean — the same reason upstream
+        make_graphed_callables captures via autograd.grad). Grads are
+        copied into persistent ``p.grad`` buffers so the cached
+        clip/optimizer meta blobs keep stable pointers."""
         out = self.model(self.cast(self.static))
-        out.loss.backward()
+        grads = torch.autograd.grad(out.loss, self._params,
+                                    allow_unused=True)
+        with torch.no_grad():
+            for p, g in zip(self._params, grads):
+                if p.grad is None:
+                    p.grad = torch.zeros_like(p)  # persistent buffer
+                if g is None:
+                    p.grad.zero_()
+                else:
+                    p.grad.copy_(g)
         if self.dp is not None:
-            self.dp.finalize_backward()
+            # autograd.grad fires no post-accumulate hooks: run the
+            # bucketed all-reduce explicitly (recorded into the graph)
+            self.dp.reduce_in_graph()
         self.clip_fn(self.model)
         self.opt.step()
         return out.loss

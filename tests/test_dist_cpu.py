@@ -168,3 +168,62 @@ def test_nerdataset_rank_sharding(tmp_path):
     # shards must not overlap (synthetic rows are distinct w.h.p.)
     assert not (seen[0] & seen[1])
     assert len(seen[0]) + len(seen[1]) >= 100
+
+
+def test_reduce_in_graph_matches_finalize_backward():
+    """reduce_in_graph (the captured-step DP path, no hooks) must
+    produce the same averaged grads as the hook + finalize path."""
+    import torch.distributed as dist
+    from chinesener_amd.dist import BucketedDataParallel, init_process_group
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29534")
+    os.environ["RANK"] = "0"
+    os.environ["WORLD_SIZE"] = "1"
+    created = not dist.is_initialized()
+    if created:
+        init_process_group("gloo")
+    try:
+        class M(torch.nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.a = torch.nn.Linear(32, 32)
+                self.b = torch.nn.Linear(32, 32)
+                self.unused = torch.nn.Linear(4, 4)
+
+            def forward(self, x):
+                return self.b(self.a(x)).sum()
+
+        torch.manual_seed(1)
+        m = M()
+        dp = BucketedDataParallel(m, bucket_cap_mb=0.005)
+        x = torch.randn(4, 32)
+
+        # path 1: hooks + finalize
+        dp.zero_grad()
+        m(x).backward()
+        dp.finalize_backward()
+        ref = {n: p.grad.clone() for n, p in m.named_parameters()
+               if p.grad is not None}
+
+        # path 2: autograd.grad into p.grad + reduce_in_graph
+        params = [p for p in m.parameters() if p.requires_grad]
+        loss = m(x)
+        grads = torch.autograd.grad(loss, params, allow_unused=True)
+        with torch.no_grad():
+            for p, g in zip(params, grads):
+                if p.grad is None and g is None:
+                    continue
+                if p.grad is None:
+                    p.grad = torch.zeros_like(p)
+                if g is None:
+                    p.grad.zero_()
+                else:
+                    p.grad.copy_(g)
+        dp.reduce_in_graph()
+        for n, p in m.named_parameters():
+            if n in ref:
+                torch.testing.assert_close(p.grad, ref[n],
+                                           atol=1e-6, rtol=1e-5)
+    finally:
+        if created:
+            dist.destroy_process_group()
