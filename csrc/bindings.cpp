@@ -30,6 +30,8 @@ at::Tensor mask_scale(const at::Tensor&, const at::Tensor&, double);
 void bump_counter(const at::Tensor&);
 at::Tensor wgrad(const at::Tensor&, const at::Tensor&, long);
 at::Tensor wgrad2(const at::Tensor&, const at::Tensor&, long);
+at::Tensor gemm_nt(const at::Tensor&, const at::Tensor&,
+                   c10::optional<at::Tensor>, bool);
 // crf.hip
 std::vector<at::Tensor> crf_fwd(const at::Tensor&, const at::Tensor&,
                                 const at::Tensor&, const at::Tensor&);
@@ -112,6 +114,8 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("bump_counter", &bump_counter);
   m.def("wgrad", &wgrad);
   m.def("wgrad2", &wgrad2);
+  m.def("gemm_nt", &gemm_nt, py::arg("a"), py::arg("b"),
+        py::arg("bias") = c10::nullopt, py::arg("fp32_out") = false);
   m.def("crf_fwd", &crf_fwd);
   m.def("crf_viterbi", &crf_viterbi);
   m.def("softlexicon_fwd", &softlexicon_fwd);
