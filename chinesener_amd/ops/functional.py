@@ -131,6 +131,19 @@ def _use_wgrad2() -> bool:
     return os.environ.get("CHINESENER_WGRAD2") == "1"
 
 
+def _use_gemm_nt() -> bool:
+    """Route eligible forward linears through the in-tree NT MFMA GEMM
+    (csrc/gemm_nt.hip) instead of hipBLASLt. Opt-in until the hardware
+    microbenchmark (scripts/bench_gemm_nt.py) shows parity."""
+    return os.environ.get("CHINESENER_GEMM_NT") == "1"
+
+
+def _gemm_nt_ok(x2, w):
+    return (x2.shape[0] % 128 == 0 and w.shape[0] % 128 == 0
+            and w.shape[1] % 64 == 0 and x2.dtype == torch.bfloat16
+            and w.dtype == torch.bfloat16)
+
+
 class _LinearFn(torch.autograd.Function):
     """nn.Linear math with a custom column-sum bias grad (torch's generic
     reduce is ~4.5x off memory-bound for [tokens, features] dbias)."""
@@ -138,6 +151,13 @@ class _LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b):
         ctx.save_for_backward(x, w)
+        if _use_gemm_nt():
+            x2 = x.reshape(-1, x.shape[-1])
+            if _gemm_nt_ok(x2, w):
+                y = get_ext().gemm_nt(x2.contiguous(), w.contiguous(),
+                                      b.float() if b is not None else None,
+                                      False)
+                return y.reshape(*x.shape[:-1], w.shape[0])
         return F.linear(x, w, b)
 
     @staticmethod
