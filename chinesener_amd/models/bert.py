@@ -87,7 +87,7 @@ class BertLayer(nn.Module):
         a = ops.linear(ctx, self.attn_out.weight, self.attn_out.bias)
         x = ops.dropout_add_layernorm(a, x, self.ln1_w, self.ln1_b, self.eps,
                                       self.p_drop, self.training)
-        f = ops.bias_gelu(self.ffn_in(x), self.ffn_in_bias)
+        f = ops.bias_gelu(ops.linear(x, self.ffn_in.weight), self.ffn_in_bias)
         f = ops.linear(f, self.ffn_out.weight, self.ffn_out.bias)
         return ops.dropout_add_layernorm(f, x, self.ln2_w, self.ln2_b,
                                          self.eps, self.p_drop, self.training)

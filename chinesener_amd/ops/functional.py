@@ -131,17 +131,49 @@ def _use_wgrad2() -> bool:
     return os.environ.get("CHINESENER_WGRAD2") == "1"
 
 
-def _use_gemm_nt() -> bool:
-    """Route eligible forward linears through the in-tree NT MFMA GEMM
-    (csrc/gemm_nt.hip) instead of hipBLASLt. Opt-in until the hardware
-    microbenchmark (scripts/bench_gemm_nt.py) shows parity."""
-    return os.environ.get("CHINESENER_GEMM_NT") == "1"
+_GEMM_NT_CHOICE: dict = {}   # (M, N, K, has_bias) -> bool (use in-tree)
+
+
+def _gemm_nt_mode() -> str:
+    """'auto' (default): measure in-tree NT MFMA GEMM vs hipBLASLt once
+    per shape and keep the winner; 'force': always in-tree; 'off'."""
+    return os.environ.get("CHINESENER_GEMM_NT", "auto")
 
 
 def _gemm_nt_ok(x2, w):
     return (x2.shape[0] % 128 == 0 and w.shape[0] % 128 == 0
             and w.shape[1] % 64 == 0 and x2.dtype == torch.bfloat16
             and w.dtype == torch.bfloat16)
+
+
+def _pick_gemm_nt(x2, w, bf) -> bool:
+    """Measured dispatch: first time a (M,N,K,bias) shape shows up, time
+    both paths (median of 5 after warmup) and cache the winner. The
+    hand-written kernel carries every shape it wins on; hipBLASLt keeps
+    the rest (scripts/bench_gemm_nt.py has the per-shape table)."""
+    key = (x2.shape[0], w.shape[0], w.shape[1], bf is not None)
+    hit = _GEMM_NT_CHOICE.get(key)
+    if hit is not None:
+        return hit
+    ext = get_ext()
+    import time as _time
+
+    def _med(fn):
+        fn(); fn()
+        torch.cuda.synchronize()
+        ts = []
+        for _ in range(5):
+            t0 = _time.perf_counter()
+            fn()
+            torch.cuda.synchronize()
+            ts.append(_time.perf_counter() - t0)
+        return sorted(ts)[2]
+
+    bb = bf.to(torch.bfloat16) if bf is not None else None
+    t_ours = _med(lambda: ext.gemm_nt(x2, w, bf, False))
+    t_lib = _med(lambda: F.linear(x2, w, bb))
+    _GEMM_NT_CHOICE[key] = bool(t_ours <= t_lib)
+    return _GEMM_NT_CHOICE[key]
 
 
 class _LinearFn(torch.autograd.Function):
@@ -151,13 +183,17 @@ class _LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b):
         ctx.save_for_backward(x, w)
-        if _use_gemm_nt():
+        ctx.has_bias = b is not None
+        mode = _gemm_nt_mode()
+        if mode != "off":
             x2 = x.reshape(-1, x.shape[-1])
             if _gemm_nt_ok(x2, w):
-                y = get_ext().gemm_nt(x2.contiguous(), w.contiguous(),
-                                      b.float() if b is not None else None,
-                                      False)
-                return y.reshape(*x.shape[:-1], w.shape[0])
+                x2 = x2.contiguous()
+                wc = w.contiguous()
+                bf = b.float() if b is not None else None
+                if mode == "force" or _pick_gemm_nt(x2, wc, bf):
+                    y = get_ext().gemm_nt(x2, wc, bf, False)
+                    return y.reshape(*x.shape[:-1], w.shape[0])
         return F.linear(x, w, b)
 
     @staticmethod
@@ -174,18 +210,20 @@ class _LinearFn(torch.autograd.Function):
             dw = get_ext().wgrad2(dyf, x2.contiguous(), 0).to(w.dtype)
         else:
             dw = (dyf.T @ x2).to(w.dtype)
-        db = get_ext().colsum(dyf)
+        db = get_ext().colsum(dyf) if ctx.has_bias else None
         return dx, dw, db
 
 
 def linear(x, weight, bias=None):
-    """Linear with fused-epilogue forward (GemmAndBias) and custom dbias.
+    """Linear with measured GEMM dispatch (in-tree NT MFMA kernel vs
+    hipBLASLt, see _pick_gemm_nt) and custom dbias.
 
     Used on the uniform-dtype pure-bf16 path; under autocast (mixed
     param/activation dtypes) or off-GPU it falls back to F.linear."""
-    if (bias is not None and hip_enabled(x) and x.dtype == weight.dtype
+    if (hip_enabled(x) and x.dtype == weight.dtype
             and not torch.is_autocast_enabled()):
-        return _LinearFn.apply(x, weight, bias.to(x.dtype))
+        return _LinearFn.apply(x, weight,
+                               bias.to(x.dtype) if bias is not None else None)
     return F.linear(x, weight, bias)
 
 
