@@ -57,6 +57,30 @@ for shape in [(8192, 2304, 768), (8192, 768, 768), (8192, 3072, 768),
 check(8192, 3072, 768, bias=False)
 check(4096, 4096, 4096, bias=False)   # guide ladder reference shape
 
+
+def check_bwd(M, N, K):
+    """dgrad/wgrad candidates incl. the transpose cost."""
+    dy = torch.randn(M, N, device="cuda", dtype=torch.bfloat16)
+    x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+    fl_dx = 2.0 * M * N * K
+    t1 = bench(lambda: ext.gemm_nt(dy, w.t().contiguous(), None, False))
+    t2 = bench(lambda: dy @ w)
+    t3 = bench(lambda: ext.gemm_nt(dy.t().contiguous(), x.t().contiguous(),
+                                   None, False))
+    t4 = bench(lambda: dy.T @ x)
+    line = (f"bwd M={M:5d} N={N:4d} K={K:4d}: "
+            f"dx ours {t1:6.1f}us ({fl_dx/t1/1e6:4.0f}) lib {t2:6.1f}us "
+            f"({fl_dx/t2/1e6:4.0f}) | dw ours {t3:6.1f}us "
+            f"({fl_dx/t3/1e6:4.0f}) lib {t4:6.1f}us ({fl_dx/t4/1e6:4.0f})")
+    print(line, flush=True)
+    lines.append(line)
+
+
+for shape in [(8192, 3072, 768), (8192, 768, 3072), (8192, 2304, 768),
+              (8192, 768, 768), (9600, 3072, 768)]:
+    check_bwd(*shape)
+
 os.makedirs("gpurun_out", exist_ok=True)
 with open("gpurun_out/gemm_nt_bench.txt", "w") as f:
     f.write("\n".join(lines) + "\n")
