@@ -132,3 +132,75 @@ def test_tf_bert_name_mapping_loader():
                                v_bias.double())
     emb = torch.as_tensor(tf["bert/embeddings/word_embeddings"])
     torch.testing.assert_close(model.embeddings.word.weight.double(), emb.double())
+
+
+def test_tf_checkpoint_bundle_roundtrip_and_bert_load(tmp_path):
+    """TF v2 tensor-bundle file format: write a bert-base-google-shaped
+    checkpoint (the FULL TF variable inventory incl. pooler/cls heads,
+    as in /root/reference pretrain_model/ch_google) with the in-tree
+    writer, read it back with the from-scratch parser, and load it into
+    BertModel via the name mapping with zero unexpected keys."""
+    import numpy as np
+    import torch
+    from chinesener_amd.models.bert import BertConfig, BertModel
+    from chinesener_amd.models.bert_loader import load_tf_bert
+    from chinesener_amd.models.tf_checkpoint import (read_tf_checkpoint,
+                                                     write_tf_checkpoint)
+
+    H, I, V, LYR = 24, 48, 120, 2
+    rng = np.random.default_rng(9)
+    tensors = {
+        "bert/embeddings/word_embeddings": rng.normal(size=(V, H)),
+        "bert/embeddings/position_embeddings": rng.normal(size=(40, H)),
+        "bert/embeddings/token_type_embeddings": rng.normal(size=(2, H)),
+        "bert/embeddings/LayerNorm/gamma": rng.normal(size=(H,)),
+        "bert/embeddings/LayerNorm/beta": rng.normal(size=(H,)),
+        # heads a real bert_model.ckpt carries (ignored by the mapping)
+        "bert/pooler/dense/kernel": rng.normal(size=(H, H)),
+        "bert/pooler/dense/bias": rng.normal(size=(H,)),
+        "cls/predictions/output_bias": rng.normal(size=(V,)),
+        "cls/predictions/transform/dense/kernel": rng.normal(size=(H, H)),
+        "cls/predictions/transform/dense/bias": rng.normal(size=(H,)),
+        "cls/predictions/transform/LayerNorm/gamma": rng.normal(size=(H,)),
+        "cls/predictions/transform/LayerNorm/beta": rng.normal(size=(H,)),
+        "cls/seq_relationship/output_weights": rng.normal(size=(2, H)),
+        "cls/seq_relationship/output_bias": rng.normal(size=(2,)),
+        "global_step": np.array(123, dtype=np.int64),
+    }
+    for i in range(LYR):
+        p = f"bert/encoder/layer_{i}"
+        for nm, shape in [("attention/self/query", (H, H)),
+                          ("attention/self/key", (H, H)),
+                          ("attention/self/value", (H, H)),
+                          ("attention/output/dense", (H, H)),
+                          ("intermediate/dense", (H, I)),
+                          ("output/dense", (I, H))]:
+            tensors[f"{p}/{nm}/kernel"] = rng.normal(size=shape)
+            tensors[f"{p}/{nm}/bias"] = rng.normal(size=(shape[1],))
+        for ln in ("attention/output/LayerNorm", "output/LayerNorm"):
+            tensors[f"{p}/{ln}/gamma"] = rng.normal(size=(H,))
+            tensors[f"{p}/{ln}/beta"] = rng.normal(size=(H,))
+    tensors = {k: (v.astype(np.float32) if v.dtype == np.float64 else v)
+               for k, v in tensors.items()}
+
+    prefix = str(tmp_path / "bert_model.ckpt")
+    write_tf_checkpoint(prefix, tensors)
+    back = read_tf_checkpoint(prefix)
+    assert set(back) == set(tensors), (
+        set(tensors) - set(back), set(back) - set(tensors))
+    for k, v in tensors.items():
+        got = back[k]
+        assert got.shape == tuple(np.shape(v)), k
+        np.testing.assert_array_equal(np.asarray(got), np.asarray(v), err_msg=k)
+
+    cfg = BertConfig(vocab_size=V, hidden_size=H, num_hidden_layers=LYR,
+                     num_attention_heads=2, intermediate_size=I,
+                     max_position_embeddings=40)
+    model = BertModel(cfg)
+    missing, unexpected = load_tf_bert(model, back)
+    assert not unexpected, unexpected
+    assert not missing, missing      # every model tensor received a value
+    q = torch.as_tensor(
+        tensors["bert/encoder/layer_1/attention/self/query/kernel"]).T
+    torch.testing.assert_close(model.layers[1].qkv.weight[:H].double(),
+                               q.double())
