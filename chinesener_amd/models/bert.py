@@ -47,12 +47,11 @@ class BertEmbeddings(nn.Module):
 
     def forward(self, token_ids, token_type_ids=None):
         B, L = token_ids.shape
-        pos = torch.arange(L, device=token_ids.device).unsqueeze(0)
-        x = self.word(token_ids) + self.position(pos)
-        if token_type_ids is not None:
-            x = x + self.token_type(token_type_ids)
-        else:
-            x = x + self.token_type.weight[0]
+        if token_type_ids is None:
+            token_type_ids = torch.zeros_like(token_ids)
+        # fused 3-table gather-sum (K1); torch fallback off-GPU
+        x = ops.embed3(self.word.weight, self.position.weight,
+                       self.token_type.weight, token_ids, token_type_ids)
         x = ops.layernorm(x, self.ln_weight, self.ln_bias, self.eps)
         return self.dropout(x)
 

@@ -55,6 +55,7 @@ def get_ext():
 
 from .functional import (  # noqa: E402,F401
     attention, attention_qkv, bias_gelu, add_layernorm, dropout_add_layernorm,
+    embed3,
     layernorm, linear, crf_nll,
     crf_viterbi, bilstm, softlexicon_fuse, masked_cross_entropy, dice_loss,
     tener_attention,

@@ -272,6 +272,47 @@ def linear(x, weight, bias=None):
     return F.linear(x, weight, bias)
 
 
+# ---------------------------------------------------- embedding gather
+class _Embed3Fn(torch.autograd.Function):
+    """Fused word+position+token_type gather-sum: one
+    kernel instead of three gathers + two adds. Backward keeps the
+    standard scatter/sum torch ops (same work nn.Embedding does)."""
+
+    @staticmethod
+    def forward(ctx, word_w, pos_w, tok_w, token_ids, segment_ids):
+        ctx.save_for_backward(token_ids, segment_ids)
+        ctx.shapes = (word_w.shape, pos_w.shape, tok_w.shape)
+        return get_ext().embed3_fwd(word_w, pos_w, tok_w,
+                                    token_ids.contiguous(),
+                                    segment_ids.contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):
+        token_ids, segment_ids = ctx.saved_tensors
+        (vs, ps, ts) = ctx.shapes
+        B, L, H = dy.shape
+        dyf = dy.reshape(-1, H)
+        dw = torch.zeros(vs, dtype=dy.dtype, device=dy.device)
+        dw.index_add_(0, token_ids.reshape(-1), dyf)
+        dw[0].zero_()          # padding_idx=0 semantics (nn.Embedding)
+        dp = torch.zeros(ps, dtype=dy.dtype, device=dy.device)
+        dp[:L] = dy.sum(0)
+        dt = torch.zeros(ts, dtype=dy.dtype, device=dy.device)
+        dt.index_add_(0, segment_ids.reshape(-1), dyf)
+        return dw, dp, dt, None, None
+
+
+def embed3(word_w, pos_w, tok_w, token_ids, segment_ids):
+    """out[b,l] = word_w[token_ids[b,l]] + pos_w[l] + tok_w[seg[b,l]]."""
+    if (hip_enabled(word_w) and word_w.dtype == torch.bfloat16
+            and word_w.shape[1] % 8 == 0
+            and token_ids.shape[1] <= pos_w.shape[0]):
+        return _Embed3Fn.apply(word_w, pos_w, tok_w, token_ids, segment_ids)
+    pos = torch.arange(token_ids.shape[1], device=token_ids.device)
+    return (F.embedding(token_ids, word_w, padding_idx=0) + pos_w[pos]
+            + F.embedding(segment_ids, tok_w))
+
+
 # ------------------------------------------------------------ bias gelu
 class _BiasGeluFn(torch.autograd.Function):
     @staticmethod

@@ -32,6 +32,9 @@ at::Tensor wgrad(const at::Tensor&, const at::Tensor&, long);
 at::Tensor wgrad2(const at::Tensor&, const at::Tensor&, long);
 at::Tensor gemm_nt(const at::Tensor&, const at::Tensor&,
                    c10::optional<at::Tensor>, bool);
+at::Tensor embed3_fwd(const at::Tensor&, const at::Tensor&,
+                      const at::Tensor&, const at::Tensor&,
+                      const at::Tensor&);
 // crf.hip
 std::vector<at::Tensor> crf_fwd(const at::Tensor&, const at::Tensor&,
                                 const at::Tensor&, const at::Tensor&);
@@ -116,6 +119,7 @@ PYBIND11_MODULE(_hip_ops, m) {
   m.def("wgrad2", &wgrad2);
   m.def("gemm_nt", &gemm_nt, py::arg("a"), py::arg("b"),
         py::arg("bias") = c10::nullopt, py::arg("fp32_out") = false);
+  m.def("embed3_fwd", &embed3_fwd);
   m.def("crf_fwd", &crf_fwd);
   m.def("crf_viterbi", &crf_viterbi);
   m.def("softlexicon_fwd", &softlexicon_fwd);

@@ -18,7 +18,7 @@ BUILD = os.path.join(REPO, "build")
 
 HIP_SOURCES = ["elementwise.hip", "crf.hip", "softlexicon.hip", "adam.hip",
                "attention.hip", "tener.hip", "lstm.hip", "probe.hip",
-               "wgrad.hip", "gemm_nt.hip"]
+               "wgrad.hip", "gemm_nt.hip", "embed.hip"]
 CPP_SOURCES = ["bindings.cpp"]
 
 

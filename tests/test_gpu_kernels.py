@@ -692,3 +692,40 @@ def test_gemm_nt_kernel():
     b = torch.randn(128, 64, device="cuda", dtype=torch.bfloat16)
     with pytest.raises(RuntimeError):
         ext.gemm_nt(a, b, None, False)
+
+
+def test_embed3_fused_gather():
+    """Fused word+pos+token_type gather-sum (csrc/embed.hip) vs the
+    torch three-gather reference, fwd + bwd (incl. padding_idx=0 grad
+    zeroing)."""
+    _cuda()
+    torch.manual_seed(91)
+    from chinesener_amd.ops import functional as fn
+    V, P, S, H, B, L = 500, 64, 2, 96, 4, 33
+    w = torch.randn(V, H, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    p = torch.randn(P, H, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    t = torch.randn(S, H, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    ids = torch.randint(0, V, (B, L), device="cuda")
+    segs = torch.randint(0, S, (B, L), device="cuda")
+    out = fn.embed3(w, p, t, ids, segs)
+    w2 = w.detach().clone().requires_grad_()
+    p2 = p.detach().clone().requires_grad_()
+    t2 = t.detach().clone().requires_grad_()
+    import torch.nn.functional as F
+    ref = (F.embedding(ids, w2, padding_idx=0)
+           + p2[torch.arange(L, device="cuda")]
+           + F.embedding(segs, t2))
+    torch.testing.assert_close(out.float(), ref.float(), atol=2e-2,
+                               rtol=1e-2)
+    g = torch.randn_like(ref)
+    out.backward(g)
+    ref.backward(g)
+    torch.testing.assert_close(w.grad.float(), w2.grad.float(),
+                               atol=5e-2, rtol=2e-2)
+    torch.testing.assert_close(p.grad.float(), p2.grad.float(),
+                               atol=5e-2, rtol=2e-2)
+    torch.testing.assert_close(t.grad.float(), t2.grad.float(),
+                               atol=5e-2, rtol=2e-2)
