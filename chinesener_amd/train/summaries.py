@@ -91,6 +91,27 @@ class SummaryLogger:
         self._records = {}
         return self.path
 
+    def flush_artifacts(self, model: torch.nn.Module, step: int) -> str:
+        """Dump the visual-summary artifacts the reference renders as
+        TensorBoard images (CRF transition matrix image+histogram,
+        tools/layer.py:129-130; attention image summaries,
+        tools/transformer/modules.py:128): the raw matrices land in
+        ``{model_dir}/artifacts/step_{N}.npz`` for offline plotting —
+        no TensorBoard in this stack, the npz IS the artifact."""
+        art_dir = os.path.join(os.path.dirname(self.path), "artifacts")
+        os.makedirs(art_dir, exist_ok=True)
+        import numpy as np
+        blobs = {}
+        for name, p in model.named_parameters():
+            low = name.lower()
+            if "transitions" in low:          # CRF transition matrices
+                blobs[name] = p.detach().float().cpu().numpy()
+            elif low.endswith(("u", "v")) and p.dim() == 2 and p.shape[0] <= 16:
+                blobs[name] = p.detach().float().cpu().numpy()  # TENER u/v
+        path = os.path.join(art_dir, f"step_{step}.npz")
+        np.savez(path, **blobs) if blobs else None
+        return path
+
     def close(self):
         for h in self._handles:
             h.remove()

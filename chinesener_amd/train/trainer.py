@@ -176,6 +176,8 @@ class Trainer:
                 self._drop_graph()
             if save_steps and self.step % save_steps == 0 and self.rank == 0:
                 self.ckpt.save(self.step, self.model, self.optimizer)
+                if self.summary is not None:
+                    self.summary.flush_artifacts(self.model, self.step)
                 if eval_fn is not None:
                     ev = eval_fn()
                     if ev < best_eval:

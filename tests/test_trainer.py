@@ -272,3 +272,27 @@ def test_main_cli_addon_flags():
     assert abs(params["task_weight"][1] - 0.3) < 1e-12
     assert params["lambda"] == 0.2
     assert params["shrink_gradient_reverse"] == 0.01   # None kept default
+
+
+def test_summary_artifacts_dump(tmp_path):
+    """flush_artifacts writes the CRF transition matrix npz (the
+    reference's TB transition image/histogram parity,
+    tools/layer.py:129-130)."""
+    import numpy as np
+    from chinesener_amd.config import resolve_params
+    from chinesener_amd.models import build_model, model_params
+    from chinesener_amd.train.summaries import SummaryLogger
+
+    params = resolve_params(model_params("bilstm_crf"), {
+        "vocab_size": 50, "label_size": 7, "embedding_dim": 8,
+        "model_name": "bilstm_crf",
+        "rnn_params": {"hidden_units_list": [8], "cell_activation": "tanh",
+                       "keep_prob_list": [1.0]}})
+    model = build_model("bilstm_crf", params)
+    sl = SummaryLogger(model, str(tmp_path), every=10)
+    path = sl.flush_artifacts(model, 10)
+    blob = np.load(path)
+    trans_keys = [k for k in blob.files if "transitions" in k]
+    assert trans_keys, blob.files
+    assert blob[trans_keys[0]].shape == (7, 7)
+    sl.close()
