@@ -42,7 +42,28 @@ class DevicePrefetcher:
         self._next = None
         self._event = None
         self._host = None          # keep pinned source alive until copied
+        # 2-deep ring of PERSISTENT device buffers per batch signature:
+        # the training loop must stay allocation-free between hipGraph
+        # replays (fresh per-batch device tensors are foreign allocator
+        # activity — see Trainer._stage), and reuse also spares the
+        # caching-allocator cross-stream bookkeeping
+        self._ring: list = [None, None]
+        self._ring_sig: list = [None, None]
+        self._slot = 0
         self._preload()
+
+    def _ring_buffers(self, host):
+        sig = tuple(sorted((k, tuple(v.shape), str(v.dtype))
+                           for k, v in host.items()
+                           if isinstance(v, torch.Tensor)))
+        s = self._slot
+        self._slot = 1 - s
+        if self._ring_sig[s] != sig:
+            self._ring[s] = {k: torch.empty_like(v, device=self.device)
+                             for k, v in host.items()
+                             if isinstance(v, torch.Tensor)}
+            self._ring_sig[s] = sig
+        return self._ring[s]
 
     def _preload(self):
         try:
@@ -58,10 +79,19 @@ class DevicePrefetcher:
                         if isinstance(v, torch.Tensor) and v.device.type == "cpu"
                         else v)
                     for k, v in host.items()}
+            bufs = self._ring_buffers(host)
+            # _preload(t+2) is issued after the consumer ENQUEUED all
+            # reads of batch t (the for-loop calls __next__ after
+            # train_step returns), so an event on the compute stream here
+            # orders the slot overwrite after those reads
+            done = torch.cuda.Event()
+            done.record(torch.cuda.current_stream(self.device))
+            self.stream.wait_event(done)
             with torch.cuda.stream(self.stream):
-                self._next = {k: (v.to(self.device, non_blocking=True)
-                                  if isinstance(v, torch.Tensor) else v)
-                              for k, v in host.items()}
+                for k, v in host.items():
+                    if isinstance(v, torch.Tensor):
+                        bufs[k].copy_(v, non_blocking=True)
+                self._next = dict(bufs)
                 self._event = torch.cuda.Event()
                 self._event.record(self.stream)
             self._host = host
@@ -78,14 +108,9 @@ class DevicePrefetcher:
             raise StopIteration
         batch = self._next
         if self.use_stream:
-            cur = torch.cuda.current_stream(self.device)
-            cur.wait_event(self._event)
-            # the caching allocator attributes these tensors to the side
-            # stream; without record_stream it may hand their memory to the
-            # next prefetch while compute kernels still read them
-            for v in batch.values():
-                if isinstance(v, torch.Tensor) and v.is_cuda:
-                    v.record_stream(cur)
+            # ring buffers are persistent (no allocator hand-off), so a
+            # stream wait on the copy event is all that's needed
+            torch.cuda.current_stream(self.device).wait_event(self._event)
         self._preload()
         return batch
 

@@ -158,7 +158,14 @@ def _pick2(key, fn_ours, fn_lib) -> bool:
             ts.append(_time.perf_counter() - t0)
         return sorted(ts)[2]
 
-    _GEMM_NT_CHOICE[key] = bool(_med(fn_ours) <= _med(fn_lib))
+    t_ours, t_lib = _med(fn_ours), _med(fn_lib)
+    _GEMM_NT_CHOICE[key] = bool(t_ours <= t_lib)
+    if os.environ.get("CHINESENER_GEMM_DEBUG") == "1":
+        import sys
+        print(f"[gemm-dispatch] {key}: ours {t_ours*1e6:.1f}us "
+              f"lib {t_lib*1e6:.1f}us -> "
+              f"{'ours' if _GEMM_NT_CHOICE[key] else 'lib'}",
+              file=sys.stderr, flush=True)
     return _GEMM_NT_CHOICE[key]
 
 
@@ -189,6 +196,12 @@ def _pick_gemm_nt(x2, w, bf) -> bool:
     t_ours = _med(lambda: ext.gemm_nt(x2, w, bf, False))
     t_lib = _med(lambda: F.linear(x2, w, bb))
     _GEMM_NT_CHOICE[key] = bool(t_ours <= t_lib)
+    if os.environ.get("CHINESENER_GEMM_DEBUG") == "1":
+        import sys
+        print(f"[gemm-dispatch] fwd {key}: ours {t_ours*1e6:.1f}us "
+              f"lib {t_lib*1e6:.1f}us -> "
+              f"{'ours' if _GEMM_NT_CHOICE[key] else 'lib'}",
+              file=sys.stderr, flush=True)
     return _GEMM_NT_CHOICE[key]
 
 

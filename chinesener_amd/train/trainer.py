@@ -59,6 +59,8 @@ class Trainer:
         # divergence (optimizers.py LrSchedule.apply).
         self._graph: GraphedTrainStep | None = None
         self._last_sig = None
+        self._staging: Dict[str, torch.Tensor] = {}
+        self._staging_sig = None
         self.use_step_graph = (self.device.startswith("cuda")
                                and os.environ.get("CHINESENER_STEPGRAPH")
                                != "0")
@@ -97,9 +99,26 @@ class Trainer:
                 return self.model(batch)
         return self.model(batch)
 
+    def _stage(self, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+        """Copy a batch into PERSISTENT device staging buffers (same
+        shapes -> same storage every step). The replay loop must be
+        allocation-free: fresh per-step `.to(device)` tensors are
+        foreign allocator activity between graph replays, which this
+        stack tolerates only statistically (soak runs crashed with HSA
+        aperture faults after ~1-2k replays; allocation-free loops are
+        stable — scripts/debug_capture_bisect.py premade arms)."""
+        sig = tuple(sorted((k, tuple(v.shape), str(v.dtype))
+                           for k, v in batch.items()))
+        if self._staging_sig != sig:
+            self._staging = {k: torch.empty_like(v, device=self.device)
+                             for k, v in batch.items()}
+            self._staging_sig = sig
+        for k, v in batch.items():
+            self._staging[k].copy_(v, non_blocking=True)
+        return self._staging
+
     def train_step(self, batch: Dict[str, torch.Tensor]) -> float:
-        batch = {k: v.to(self.device, non_blocking=True)
-                 for k, v in batch.items()}
+        batch = self._stage(batch)
         graph_ready = (not self._has_diff_lr
                        or self.step >= self.schedule.warmup)
         if self.dp is not None:
