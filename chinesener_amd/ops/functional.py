@@ -139,12 +139,26 @@ def _gemm_nt_ok(x2, w):
             and w.dtype == torch.bfloat16)
 
 
+def _tuning_active() -> bool:
+    """hipBLASLt TunableOp tuning in progress: library timings are 2-3x
+    inflated, so dispatch decisions must wait (bench.py freezes tuning
+    after its warmup steps)."""
+    try:
+        import torch.cuda.tunable as tun
+        return tun.is_enabled() and tun.tuning_is_enabled()
+    except Exception:
+        return False
+
+
 def _pick2(key, fn_ours, fn_lib) -> bool:
-    """Measured dispatch between two GEMM paths: first time `key` shows
-    up, time both (median of 5 after warmup) and cache the winner."""
+    """Measured dispatch between two GEMM paths: once the library's
+    TunableOp tuning is done, time both (median of 5) and cache the
+    winner."""
     hit = _GEMM_NT_CHOICE.get(key)
     if hit is not None:
         return hit
+    if torch.cuda.is_current_stream_capturing() or _tuning_active():
+        return False               # never measure now; decide later
     import time as _time
 
     def _med(fn):
@@ -178,6 +192,8 @@ def _pick_gemm_nt(x2, w, bf) -> bool:
     hit = _GEMM_NT_CHOICE.get(key)
     if hit is not None:
         return hit
+    if torch.cuda.is_current_stream_capturing() or _tuning_active():
+        return False
     ext = get_ext()
     import time as _time
 
