@@ -48,10 +48,10 @@ def load_tuned_gemm_table(path: str | None = None, tune: bool = False) -> bool:
 
 
 def freeze(dump_path: str | None = None) -> None:
-    """Stop tuning (call after warmup). This torch has no write_file;
-    results are flushed to PYTORCH_TUNABLEOP_FILENAME at process exit."""
-    if "PYTORCH_TUNABLEOP_ENABLED" in os.environ:
-        return  # env-configured: leave the env behaviour alone
+    """Stop tuning (call after warmup) — ALWAYS, env-configured or not:
+    callers rely on it (the measured GEMM dispatch defers its decisions
+    until tuning is off, functional._tuning_active). Results are flushed
+    to PYTORCH_TUNABLEOP_FILENAME at process exit."""
     try:
         import torch.cuda.tunable as tunable
         tunable.tuning_enable(False)
