@@ -162,15 +162,18 @@ def _pick2(key, fn_ours, fn_lib) -> bool:
     import time as _time
 
     def _med(fn):
+        # batch 8 launches per sync: a per-call synchronize adds a
+        # constant ~10-30us that swamps (and skews) 20-80us kernels
         fn(); fn()
         torch.cuda.synchronize()
         ts = []
-        for _ in range(5):
+        for _ in range(3):
             t0 = _time.perf_counter()
-            fn()
+            for _i in range(8):
+                fn()
             torch.cuda.synchronize()
-            ts.append(_time.perf_counter() - t0)
-        return sorted(ts)[2]
+            ts.append((_time.perf_counter() - t0) / 8)
+        return sorted(ts)[1]
 
     t_ours, t_lib = _med(fn_ours), _med(fn_lib)
     _GEMM_NT_CHOICE[key] = bool(t_ours <= t_lib)
@@ -198,15 +201,18 @@ def _pick_gemm_nt(x2, w, bf) -> bool:
     import time as _time
 
     def _med(fn):
+        # batch 8 launches per sync: a per-call synchronize adds a
+        # constant ~10-30us that swamps (and skews) 20-80us kernels
         fn(); fn()
         torch.cuda.synchronize()
         ts = []
-        for _ in range(5):
+        for _ in range(3):
             t0 = _time.perf_counter()
-            fn()
+            for _i in range(8):
+                fn()
             torch.cuda.synchronize()
-            ts.append(_time.perf_counter() - t0)
-        return sorted(ts)[2]
+            ts.append((_time.perf_counter() - t0) / 8)
+        return sorted(ts)[1]
 
     bb = bf.to(torch.bfloat16) if bf is not None else None
     t_ours = _med(lambda: ext.gemm_nt(x2, w, bf, False))
