@@ -17,16 +17,10 @@ import os
 import sys
 import time
 
-# hipBLASLt GEMM algorithm tuning (TunableOp) must be configured via env
-# BEFORE torch initializes. Tuning runs lazily inside the untimed warmup
-# steps, so the timed region sees only tuned algos. Saved tables do NOT
-# replay across processes (hipBLASLt algo indices are process-local), so
-# every run tunes fresh — ~1 min of warmup cost, ~10% steady-state gain.
-# CHINESENER_NO_TUNABLE=1 opts out.
-# TunableOp GEMM tuning is now a measured net LOSS on this stack
-# (17.55 ms/step without vs 17.82 with, plus ~45 s tuning per process
-# that multi-rank SCALE runs cannot afford) - opt-in via
-# CHINESENER_TUNABLE=1.
+# TunableOp GEMM tuning (env must be set BEFORE torch init) is a
+# measured net LOSS on this stack now (17.55 ms/step without vs 17.82
+# with, plus ~45 s tuning per process that multi-rank SCALE runs cannot
+# afford) — opt-in via CHINESENER_TUNABLE=1.
 if os.environ.get("CHINESENER_TUNABLE") == "1":
     _rank_tag = os.environ.get("RANK", "0")
     os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
@@ -97,7 +91,8 @@ def main():
     })
     if use_gpu:
         from chinesener_amd.ops.tunable import freeze, load_tuned_gemm_table
-        load_tuned_gemm_table(tune=True)
+        if os.environ.get("CHINESENER_TUNABLE") == "1":
+            load_tuned_gemm_table(tune=True)
     model = build_model(args.model, params).to(device)
     use_bf16 = use_gpu and params.get("dtype", "bf16") == "bf16"
     if use_bf16:
