@@ -17,8 +17,8 @@ def run(B, L, E, h):
     x = torch.randn(B, L, E, device="cuda", dtype=torch.bfloat16)
     args = []
     for _ in range(2):  # fw, bw
-        args += [torch.randn(4 * h, E, device="cuda", dtype=torch.bfloat16) * 0.05,
-                 torch.randn(4 * h, h, device="cuda", dtype=torch.bfloat16) * 0.05,
+        args += [torch.randn(E, 4 * h, device="cuda", dtype=torch.bfloat16) * 0.05,
+                 torch.randn(h, 4 * h, device="cuda", dtype=torch.bfloat16) * 0.05,
                  torch.zeros(4 * h, device="cuda", dtype=torch.bfloat16)]
     lens = torch.full((B,), L, dtype=torch.long, device="cuda")
     x.requires_grad_(True)
