@@ -27,8 +27,6 @@
 // in the wrapper.
 #include "common.h"
 
-#include <type_traits>
-
 #define H_MAX 256
 #define NQ_MAX 4  // column frags per wave: h/16/4 <= 4 for h <= 256
 #define NC_MAX 8  // gates_x chunks per thread: 8h/256 <= 8
@@ -267,58 +265,16 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
     mylen[r] = (b < B) ? lens[b] : 0;
   }
 
-  // Software pipeline (the fwd kernel's pattern, mirrored): step t's
-  // HBM state (gates f32x4, c_t, c_prev, upstream dh) is prefetched
-  // into registers during step t+1's MFMA panel, so the per-step
-  // latency chain is LDS+MFMA, not an HBM round trip. The c chain
-  // reuses c_prev(t) as c_t of the next (earlier-time) step.
-  float g4r[4][NQ_MAX][4], ct_r[4][NQ_MAX], cp_r[4][NQ_MAX],
-      dup_r[4][NQ_MAX];
-
-  auto issue_loads = [&](int t, bool first) {
+  for (int step = L - 1; step >= 0; --step) {
+    const int t = reverse ? (L - 1 - step) : step;  // reverse of fwd order
     const int t_prev = reverse ? (t + 1) : (t - 1);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int b = b0 + lrow + r;
       const bool inb = b < B;
       const bool valid = inb && (t < mylen[r]);
-      const long gbase = valid ? ((dbase + (long)b * L + t) * 4 * h) : 0;
-#pragma unroll
-      for (int q = 0; q < NQ_MAX; ++q) {
-        const int f = wid + 4 * q;
-        if (q >= NQ || f >= N16) continue;
-        const int j = f * 16 + (lane & 15);
-        const f32x4 g4 = valid ? *reinterpret_cast<const f32x4*>(
-                                     gates + gbase + (long)j * 4)
-                               : f32x4{0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-        for (int e = 0; e < 4; ++e) g4r[r][q][e] = g4[e];
-        if (first)
-          ct_r[r][q] = valid ? cs[(dbase + (long)b * L + t) * h + j] : 0.f;
-        else
-          ct_r[r][q] = cp_r[r][q];    // c chain: c_t(next) = c_prev(cur)
-        // NOTE: cp validity depends on t_prev alone (not on t): the
-        // chain must carry the true cs[t_prev] across the padded-region
-        // boundary so the first VALID step sees its real c_t
-        cp_r[r][q] = (inb && t_prev >= 0 && t_prev < L
-                      && t_prev < mylen[r])
-                         ? cs[(dbase + (long)b * L + t_prev) * h + j]
-                         : 0.f;
-        dup_r[r][q] = valid ? to_f32(
-                          dhs[((long)b * L + t) * hss + hs_off + j])
-                            : 0.f;
-      }
-    }
-  };
-
-  issue_loads(reverse ? 0 : (L - 1), /*first=*/true);
-
-  for (int step = L - 1; step >= 0; --step) {
-    const int t = reverse ? (L - 1 - step) : step;  // reverse of fwd order
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int b = b0 + lrow + r;
-      const bool valid = (b < B) && (t < mylen[r]);
+      const long gbase = inb ? ((dbase + (long)b * L + t) * 4 * h) : 0;
+      const long xbase = inb ? (((long)b * L + t) * gxs + gx_off) : 0;
 #pragma unroll
       for (int q = 0; q < NQ_MAX; ++q) {
         const int f = wid + 4 * q;
@@ -326,12 +282,19 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
         const int j = f * 16 + (lane & 15);
         float dgi = 0.f, dgf = 0.f, dgg = 0.f, dgo = 0.f;
         if (valid) {
-          const float gi = g4r[r][q][0], gf = g4r[r][q][1],
-                      gg = g4r[r][q][2], go = g4r[r][q][3];
-          const float c_t = ct_r[r][q];
-          const float c_prev = cp_r[r][q];
+          const long obase = (dbase + (long)b * L + t) * h + j;
+          // interleaved layout: one f32x4 load
+          const f32x4 g4 =
+              *reinterpret_cast<const f32x4*>(gates + gbase + (long)j * 4);
+          const float gi = g4[0], gf = g4[1], gg = g4[2], go = g4[3];
+          const float c_t = cs[obase];
+          const float c_prev =
+              (t_prev >= 0 && t_prev < L && t_prev < mylen[r])
+                  ? cs[(dbase + (long)b * L + t_prev) * h + j]
+                  : 0.f;
           const float ac = act_f(c_t, relu);
-          const float dh = dh_reg[r][q] + dup_r[r][q];
+          const float dh =
+              dh_reg[r][q] + to_f32(dhs[((long)b * L + t) * hss + hs_off + j]);
           float dc = dc_reg[r][q] + dh * go * dact_from_out(ac, relu);
           dgo = dh * ac * go * (1.f - go);
           // clamp boundary: no grad through a clipped cell state
@@ -341,40 +304,21 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
           dgg = dc * gi * dact_from_out(gg, relu);
           dc_reg[r][q] = dc * gf;  // carry to previous step
         }
-        const int lr = lrow + r;
-        dg_s[lr * g_ld + 0 * h + j] = __float2bfloat16(dgi);
-        dg_s[lr * g_ld + 1 * h + j] = __float2bfloat16(dgf);
-        dg_s[lr * g_ld + 2 * h + j] = __float2bfloat16(dgg);
-        dg_s[lr * g_ld + 3 * h + j] = __float2bfloat16(dgo);
-        // fp32 output path keeps full precision via direct stores
-        if (!std::is_same<T, bf16>::value && b < B) {
-          const long xbase = ((long)b * L + t) * gxs + gx_off;
+        // write pre-activation gate grads (global + LDS for the MFMA)
+        if (inb) {
           from_f32(dgi, &dgates_x[xbase + 0 * h + j]);
           from_f32(dgf, &dgates_x[xbase + 1 * h + j]);
           from_f32(dgg, &dgates_x[xbase + 2 * h + j]);
           from_f32(dgo, &dgates_x[xbase + 3 * h + j]);
         }
+        const int lr = lrow + r;
+        dg_s[lr * g_ld + 0 * h + j] = __float2bfloat16(dgi);
+        dg_s[lr * g_ld + 1 * h + j] = __float2bfloat16(dgf);
+        dg_s[lr * g_ld + 2 * h + j] = __float2bfloat16(dgg);
+        dg_s[lr * g_ld + 3 * h + j] = __float2bfloat16(dgo);
       }
     }
     __syncthreads();
-    // prefetch the NEXT (earlier-time) step's HBM state now — its
-    // latency hides behind the MFMA panel + coop store below
-    if (step > 0) issue_loads(reverse ? (L - step) : (step - 1), false);
-    // bf16 path: vectorized cooperative store of the whole [16, 4h]
-    // dgates tile (replaces 4 strided scalar stores per lane per row)
-    if (std::is_same<T, bf16>::value) {
-      const int vec_per_row = 4 * h / 8;
-      for (int i = threadIdx.x; i < 16 * vec_per_row; i += blockDim.x) {
-        const int r = i / vec_per_row;
-        const int c = (i % vec_per_row) * 8;
-        const int b = b0 + r;
-        if (b >= B) continue;
-        const long xbase = ((long)b * L + t) * gxs + gx_off;
-        *reinterpret_cast<s16x8*>(
-            reinterpret_cast<bf16*>(dgates_x) + xbase + c) =
-            *reinterpret_cast<const s16x8*>(dg_s + (long)r * g_ld + c);
-      }
-    }
     // dh_prev = dgates @ W_hh^T : [16,4h] @ [4h,h]; col frag f = wid+4q
     cfrag acc[NQ_MAX];
 #pragma unroll
