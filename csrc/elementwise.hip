@@ -607,8 +607,11 @@ at::Tensor colsum(const at::Tensor& dy) {
   auto ws = at::zeros({N}, dy.options().dtype(at::kFloat));
   auto out = at::empty({N}, dy.options());
   const int block = 256;
-  const int rows_per_chunk = 64;
   const int nvec = N / 8;
+  // size chunks for >=64k resident threads (vectorization cut the
+  // thread count 8x; without this the kernel goes latency-bound)
+  long rpc = (M * (long)nvec) / 65536;
+  const int rows_per_chunk = (int)std::max<long>(8, std::min<long>(64, rpc));
   const dim3 grid((nvec + block - 1) / block,
                   (unsigned)((M + rows_per_chunk - 1) / rows_per_chunk));
   auto stream = cur_stream(dy);
