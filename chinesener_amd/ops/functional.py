@@ -326,15 +326,18 @@ class _Embed3Fn(torch.autograd.Function):
         token_ids, segment_ids = ctx.saved_tensors
         (vs, ps, ts) = ctx.shapes
         B, L, H = dy.shape
-        dyf = dy.reshape(-1, H)
-        dw = torch.zeros(vs, dtype=dy.dtype, device=dy.device)
+        # fp32 scatter accumulation: thousands of bf16 adds per heavy row
+        # (PAD, segment 0) would otherwise lose mass to rounding
+        dyf = dy.reshape(-1, H).float()
+        dw = torch.zeros(vs, dtype=torch.float32, device=dy.device)
         dw.index_add_(0, token_ids.reshape(-1), dyf)
         dw[0].zero_()          # padding_idx=0 semantics (nn.Embedding)
-        dp = torch.zeros(ps, dtype=dy.dtype, device=dy.device)
-        dp[:L] = dy.sum(0)
-        dt = torch.zeros(ts, dtype=dy.dtype, device=dy.device)
+        dp = torch.zeros(ps, dtype=torch.float32, device=dy.device)
+        dp[:L] = dy.float().sum(0)
+        dt = torch.zeros(ts, dtype=torch.float32, device=dy.device)
         dt.index_add_(0, segment_ids.reshape(-1), dyf)
-        return dw, dp, dt, None, None
+        return (dw.to(dy.dtype), dp.to(dy.dtype), dt.to(dy.dtype),
+                None, None)
 
 
 def embed3(word_w, pos_w, tok_w, token_ids, segment_ids):
