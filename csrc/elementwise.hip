@@ -559,13 +559,11 @@ at::Tensor masked_ce_bwd(const at::Tensor& dloss, const at::Tensor& probs,
 // torch's generic reduce runs ~4.5x off memory bound for this shape.
 // ---------------------------------------------------------------------
 template <typename T>
-__global__ void colsum_kernel(const T* __restrict__ dy,
-                              float* __restrict__ partial,  // [chunks, N]
+__global__ void colsum_kernel(const T* __restrict__ dy, float* __restrict__ ws,
                               long M, int N, int rows_per_chunk) {
-  // stage 1, no atomics: each thread owns 8 consecutive columns of one
-  // row chunk (b128 loads — scalar bf16 loads ran ~8x off the memory
-  // roofline, guide §5 common mistake 2) and writes its partial sums.
-  // Chunk count is free (no contention), so occupancy comes from rows.
+  // vectorized: each thread owns 8 consecutive columns (one b128 load
+  // per row on the bf16 path — scalar bf16 loads ran ~8x off the
+  // memory roofline, guide §5 common mistake 2)
   const int v = blockIdx.x * blockDim.x + threadIdx.x;  // 8-col vector id
   const int c0 = v * 8;
   if (c0 >= N) return;
@@ -590,20 +588,8 @@ __global__ void colsum_kernel(const T* __restrict__ dy,
       }
     }
   }
-  float* out = partial + (long)blockIdx.y * N + c0;
 #pragma unroll
-  for (int e = 0; e < 4; ++e)
-    *reinterpret_cast<f32x2*>(out + e * 2) = f32x2{acc[e * 2], acc[e * 2 + 1]};
-}
-
-template <typename T>
-__global__ void colsum_reduce_kernel(const float* __restrict__ partial,
-                                     T* __restrict__ out, int N, int chunks) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= N) return;
-  float s = 0.f;
-  for (int k = 0; k < chunks; ++k) s += partial[(long)k * N + c];
-  from_f32(s, &out[c]);
+  for (int e = 0; e < 8; ++e) atomicAdd(&ws[c0 + e], acc[e]);
 }
 
 template <typename T>
@@ -618,33 +604,28 @@ at::Tensor colsum(const at::Tensor& dy) {
   const long M = dy.size(0);
   const int N = dy.size(1);
   TORCH_CHECK(N % 8 == 0, "colsum: N % 8 (got ", N, ")");
+  auto ws = at::zeros({N}, dy.options().dtype(at::kFloat));
   auto out = at::empty({N}, dy.options());
   const int block = 256;
+  const int rows_per_chunk = 64;
   const int nvec = N / 8;
-  // chunk rows so stage 1 has >= 64k resident threads (8-col vectors
-  // cut the thread count 8x); chunks are free — no atomics
-  const long want = (65536 + nvec - 1) / nvec;
-  const int rows_per_chunk =
-      (int)std::max<long>(4, std::min<long>(M, (M + want - 1) / want));
-  const int chunks = (int)((M + rows_per_chunk - 1) / rows_per_chunk);
-  auto partial = at::empty({(long)chunks, (long)N},
-                           dy.options().dtype(at::kFloat));
-  const dim3 grid((nvec + block - 1) / block, (unsigned)chunks);
+  const dim3 grid((nvec + block - 1) / block,
+                  (unsigned)((M + rows_per_chunk - 1) / rows_per_chunk));
   auto stream = cur_stream(dy);
   if (dy.scalar_type() == at::kBFloat16) {
     hipLaunchKernelGGL(colsum_kernel<bf16>, grid, dim3(block), 0, stream,
-                       (const bf16*)dy.data_ptr(), partial.data_ptr<float>(),
-                       M, N, rows_per_chunk);
-    hipLaunchKernelGGL(colsum_reduce_kernel<bf16>, dim3((N + 255) / 256),
-                       dim3(256), 0, stream, partial.data_ptr<float>(),
-                       (bf16*)out.data_ptr(), N, chunks);
+                       (const bf16*)dy.data_ptr(), ws.data_ptr<float>(), M, N,
+                       rows_per_chunk);
+    hipLaunchKernelGGL(colsum_cast_kernel<bf16>, dim3((N + 255) / 256),
+                       dim3(256), 0, stream, ws.data_ptr<float>(),
+                       (bf16*)out.data_ptr(), N);
   } else {
     hipLaunchKernelGGL(colsum_kernel<float>, grid, dim3(block), 0, stream,
-                       dy.data_ptr<float>(), partial.data_ptr<float>(), M, N,
+                       dy.data_ptr<float>(), ws.data_ptr<float>(), M, N,
                        rows_per_chunk);
-    hipLaunchKernelGGL(colsum_reduce_kernel<float>, dim3((N + 255) / 256),
-                       dim3(256), 0, stream, partial.data_ptr<float>(),
-                       out.data_ptr<float>(), N, chunks);
+    hipLaunchKernelGGL(colsum_cast_kernel<float>, dim3((N + 255) / 256),
+                       dim3(256), 0, stream, ws.data_ptr<float>(),
+                       out.data_ptr<float>(), N);
   }
   HIP_CHECK_LAST();
   return out;
