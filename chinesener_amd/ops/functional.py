@@ -342,7 +342,8 @@ class _Embed3Fn(torch.autograd.Function):
 
 def embed3(word_w, pos_w, tok_w, token_ids, segment_ids):
     """out[b,l] = word_w[token_ids[b,l]] + pos_w[l] + tok_w[seg[b,l]]."""
-    if (hip_enabled(word_w) and word_w.dtype == torch.bfloat16
+    if (os.environ.get("CHINESENER_NO_EMBED3") != "1"
+            and hip_enabled(word_w) and word_w.dtype == torch.bfloat16
             and word_w.shape[1] % 8 == 0
             and token_ids.shape[1] <= pos_w.shape[0]):
         return _Embed3Fn.apply(word_w, pos_w, tok_w, token_ids, segment_ids)

@@ -561,35 +561,13 @@ at::Tensor masked_ce_bwd(const at::Tensor& dloss, const at::Tensor& probs,
 template <typename T>
 __global__ void colsum_kernel(const T* __restrict__ dy, float* __restrict__ ws,
                               long M, int N, int rows_per_chunk) {
-  // vectorized: each thread owns 8 consecutive columns (one b128 load
-  // per row on the bf16 path — scalar bf16 loads ran ~8x off the
-  // memory roofline, guide §5 common mistake 2)
-  const int v = blockIdx.x * blockDim.x + threadIdx.x;  // 8-col vector id
-  const int c0 = v * 8;
-  if (c0 >= N) return;
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= N) return;
   const long r0 = (long)blockIdx.y * rows_per_chunk;
   const long r1 = min(M, r0 + rows_per_chunk);
-  float acc[8] = {};
-  for (long r = r0; r < r1; ++r) {
-    if (sizeof(T) == 2) {
-      const bf16x8 x = *reinterpret_cast<const bf16x8*>(
-          reinterpret_cast<const bf16*>(dy) + r * N + c0);
-#pragma unroll
-      for (int e = 0; e < 8; ++e) acc[e] += (float)x[e];
-    } else {
-      const f32x4 lo = *reinterpret_cast<const f32x4*>(
-          reinterpret_cast<const float*>(dy) + r * N + c0);
-      const f32x4 hi = *reinterpret_cast<const f32x4*>(
-          reinterpret_cast<const float*>(dy) + r * N + c0 + 4);
-#pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        acc[e] += lo[e];
-        acc[e + 4] += hi[e];
-      }
-    }
-  }
-#pragma unroll
-  for (int e = 0; e < 8; ++e) atomicAdd(&ws[c0 + e], acc[e]);
+  float acc = 0.f;
+  for (long r = r0; r < r1; ++r) acc += to_f32(dy[r * N + c]);
+  atomicAdd(&ws[c], acc);
 }
 
 template <typename T>
@@ -603,13 +581,11 @@ at::Tensor colsum(const at::Tensor& dy) {
   CHECK_CUDA_CONTIG(dy);
   const long M = dy.size(0);
   const int N = dy.size(1);
-  TORCH_CHECK(N % 8 == 0, "colsum: N % 8 (got ", N, ")");
   auto ws = at::zeros({N}, dy.options().dtype(at::kFloat));
   auto out = at::empty({N}, dy.options());
   const int block = 256;
   const int rows_per_chunk = 64;
-  const int nvec = N / 8;
-  const dim3 grid((nvec + block - 1) / block,
+  const dim3 grid((N + block - 1) / block,
                   (unsigned)((M + rows_per_chunk - 1) / rows_per_chunk));
   auto stream = cur_stream(dy);
   if (dy.scalar_type() == at::kBFloat16) {
