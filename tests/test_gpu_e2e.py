@@ -262,17 +262,24 @@ def test_every_model_trains_on_gpu(model_name):
 @pytest.mark.gpu
 def test_device_prefetcher_gpu():
     """Pinned side-stream H2D staging delivers every batch, on-device,
-    bit-identical, in order (data/loader.py DevicePrefetcher)."""
+    bit-identical, in order. Ring-buffer contract (allocation-free
+    replay loops): a handed-out batch is valid until the NEXT two
+    batches are drawn — consume as you iterate, like the train loop."""
     _cuda()
     from chinesener_amd.data.loader import DevicePrefetcher
     src = [{"token_ids": torch.randint(0, 100, (4, 32)),
             "w": torch.randn(4, 8)} for _ in range(6)]
-    out = list(DevicePrefetcher(iter(src), "cuda:0"))
-    assert len(out) == 6
-    for host, dev in zip(src, out):
+    n = 0
+    ptrs = set()
+    for host, dev in zip(src, DevicePrefetcher(iter(src), "cuda:0")):
+        n += 1
         for k in host:
             assert dev[k].is_cuda
             assert torch.equal(dev[k].cpu(), host[k])
+            ptrs.add(dev[k].data_ptr())
+    assert n == 6
+    # persistent 2-deep ring: 2 buffers per key, not 6
+    assert len(ptrs) == 4, len(ptrs)
     # and through trainer.train (the wrapping call site)
     from conftest import make_tiny_batch, make_tiny_params
     from chinesener_amd.train.trainer import Trainer
