@@ -52,7 +52,7 @@ class GraphedTrainStep:
         the resulting graph corrupts within a few replays once ANY
         foreign allocator/stream activity happens between replays (NaN
         loss or HSA aperture faults; reproduced and bisected on MI355X,
-        scripts/debug_capture_bisect.py: backward-capture arms corrupt,
+        scripts/debug/debug_capture_bisect.py: backward-capture arms corrupt,
         autograd.grad arms stay clean — the same reason upstream
         make_graphed_callables captures via autograd.grad). Grads are
         copied into persistent ``p.grad`` buffers so the cached
@@ -162,7 +162,7 @@ class GraphedTrainStep:
         eval pass, even plain torch.randn churn) corrupt subsequent
         replays on this stack within a few steps (NaN loss or
         HSA_STATUS_ERROR_MEMORY_APERTURE_VIOLATION — reproduced and
-        bisected on MI355X, scripts/debug_trainer_graph.py arms
+        bisected on MI355X, scripts/debug/debug_trainer_graph.py arms
         alloc/emptyalloc/rngonly). The Trainer therefore releases the
         graph at every eval/predict/checkpoint boundary and re-captures
         on the next repeated-shape step (~1 s per boundary)."""

@@ -106,7 +106,7 @@ class Trainer:
         foreign allocator activity between graph replays, which this
         stack tolerates only statistically (soak runs crashed with HSA
         aperture faults after ~1-2k replays; allocation-free loops are
-        stable — scripts/debug_capture_bisect.py premade arms)."""
+        stable — scripts/debug/debug_capture_bisect.py premade arms)."""
         sig = tuple(sorted((k, tuple(v.shape), str(v.dtype))
                            for k, v in batch.items()))
         if self._staging_sig != sig:
