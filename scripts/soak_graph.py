@@ -48,8 +48,10 @@ def main():
     history, bad = [], None
     trail = []
     spikes, spike_run = [], 0
+    ever_graphed = False
     for step in range(1, steps + 1):
         loss = trainer.train_step(next(gen))
+        ever_graphed = ever_graphed or trainer._graph is not None
         if not (loss == loss and abs(loss) < 1e6):
             bad = f"non-finite/exploded loss {loss} at step {step}"
             break
@@ -85,7 +87,9 @@ def main():
                 bad = f"F1 collapse {f1:.4f} at step {step}"
                 break
     result = {"steps_done": step, "clean": bad is None, "failure": bad,
-              "graph_used": trainer._graph is not None,
+              # _graph is None right after an eval boundary releases it;
+              # report whether graphs were EVER captured this run
+              "graph_used": trainer._graph is not None or ever_graphed,
               "spike_warnings": spikes[:20],
               "wall_s": round(time.time() - t0, 1), "history": history}
     out = os.path.join(os.path.dirname(os.path.dirname(
