@@ -256,15 +256,18 @@ def dice_loss(logits: torch.Tensor, labels: torch.Tensor, mask: torch.Tensor,
     with focal-style (1-p)^alpha down-weight and gamma smoothing."""
     T = logits.shape[-1]
     probs = torch.softmax(logits.float(), dim=-1)
-    m = mask.to(probs.dtype).reshape(-1)
+    m = mask.to(probs.dtype).reshape(-1, 1)
     probs = probs.reshape(-1, T)
     y = F.one_hot(labels.reshape(-1), T).to(probs.dtype)
-    loss = logits.new_zeros(())
-    for t in range(T):
-        if t in idx_skip:
-            continue
-        p = probs[:, t] * (1 - probs[:, t]) ** alpha * m
-        g = y[:, t] * m
-        dsc = 1 - (2 * (p * g).sum() + gamma) / (p.sum() + g.sum() + gamma)
-        loss = loss + dsc
-    return loss
+    # vectorized over tags (the per-tag python loop launched ~50 tiny
+    # kernels per call and cost 2.6 ms fwd+bwd on GPU; this form is a
+    # handful of [R,T] passes, same math)
+    p = probs * (1 - probs) ** alpha * m
+    g = y * m
+    num = 2 * (p * g).sum(0) + gamma
+    den = p.sum(0) + g.sum(0) + gamma
+    dsc = 1 - num / den                     # [T]
+    keep = torch.ones(T, dtype=torch.bool, device=logits.device)
+    for t in idx_skip:
+        keep[t] = False
+    return dsc[keep].sum()
