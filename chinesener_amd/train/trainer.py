@@ -132,8 +132,19 @@ class Trainer:
                 self.step += 1
                 loss = self._graph.replay(batch, self.step)
                 return float(loss.detach())
+            if self._graph is not None:
+                # shape changed (e.g. MRC's ragged padded batches): the
+                # eager step below is foreign allocator activity that
+                # corrupts a LIVE graph's later replays — drop it first
+                self._drop_graph()
+                self._shape_misses = getattr(self, "_shape_misses", 0) + 1
+                if self._shape_misses >= 3:
+                    # shape-unstable input stream: graphs thrash; go eager
+                    log.info("batch shapes unstable; disabling step graph")
+                    self.use_step_graph = False
             sig = tuple(sorted((k, tuple(v.shape)) for k, v in batch.items()))
-            if self._graph is None and sig == self._last_sig:
+            if (self.use_step_graph and self._graph is None
+                    and sig == self._last_sig):
                 g = GraphedTrainStep(
                     self.model, self.optimizer, self.schedule,
                     lambda m: clip_gradients(m, self.family),

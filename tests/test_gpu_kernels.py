@@ -73,7 +73,7 @@ def test_bias_gelu_fwd_bwd():
 
 
 # ------------------------------------------------------------ attention
-@pytest.mark.parametrize("L,D", [(128, 64), (150, 64), (64, 32)])
+@pytest.mark.parametrize("L,D", [(128, 64), (150, 64), (64, 32), (170, 64), (176, 64)])
 def test_attention_fwd_bwd(L, D):
     _cuda()
     torch.manual_seed(3)
