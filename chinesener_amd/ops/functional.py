@@ -396,13 +396,15 @@ def attention(q, k, v, mask: Optional[torch.Tensor] = None,
               scale: Optional[float] = None, lens: Optional[torch.Tensor] = None):
     """Fused attention. q,k,v [B,H,L,D]; mask [B,L] prefix mask or lens [B].
 
-    HIP path: bf16, head dim padded to 32/64, L <= 176 (covers every
-    reference config; falls back to the torch path otherwise)."""
+    HIP path: bf16, head dim padded to 32/64, padded L <= 176 i.e.
+    L <= 160 (the kernel pads L to a multiple of 32; covers every
+    reference NER config — L=150). Longer sequences (MRC's ragged
+    batches can pad past 160) fall back to the torch path."""
     D = q.shape[-1]
     if scale is None:
         scale = 1.0 / math.sqrt(D)
     if (hip_enabled(q) and q.dtype == torch.bfloat16 and D <= 64
-            and q.shape[2] <= 176):
+            and ((q.shape[2] + 31) // 32) * 32 <= 176):
         Dp = 32 if D <= 32 else 64
         if lens is None:
             lens = (mask.long().sum(1) if mask is not None
@@ -462,7 +464,7 @@ def attention_qkv(qkv, mask: Optional[torch.Tensor] = None,
     if os.environ.get("CHINESENER_NO_ATTN_DROP") == "1":
         keep = 1.0
     if (hip_enabled(qkv) and qkv.dtype == torch.bfloat16
-            and D in (32, 64) and L <= 176):
+            and D in (32, 64) and ((L + 31) // 32) * 32 <= 176):
         if lens is None:
             lens = (mask.long().sum(1) if mask is not None
                     else torch.full((B,), L, dtype=torch.long,
