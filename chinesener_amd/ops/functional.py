@@ -125,6 +125,11 @@ def add_layernorm(x, residual, weight, bias, eps: float = 1e-12):
 
 # ---------------------------------------------------------- fused linear
 _GEMM_NT_CHOICE: dict = {}   # shape key -> bool (use in-tree kernel)
+_GEMM_NT_CALLS: dict = {}    # shape key -> times seen
+_PICK_MIN_RECUR = 6          # only time shapes that actually recur:
+                             # ragged inputs (MRC's padded batches) make
+                             # every step a fresh shape, and a timing
+                             # race per step is slower than any GEMM
 
 
 def _gemm_nt_mode() -> str:
@@ -159,6 +164,10 @@ def _pick2(key, fn_ours, fn_lib) -> bool:
         return hit
     if torch.cuda.is_current_stream_capturing() or _tuning_active():
         return False               # never measure now; decide later
+    n = _GEMM_NT_CALLS.get(key, 0) + 1
+    _GEMM_NT_CALLS[key] = n
+    if n < _PICK_MIN_RECUR:
+        return False               # shape hasn't proven it recurs
     import time as _time
 
     def _med(fn):
@@ -196,6 +205,10 @@ def _pick_gemm_nt(x2, w, bf) -> bool:
     if hit is not None:
         return hit
     if torch.cuda.is_current_stream_capturing() or _tuning_active():
+        return False
+    n = _GEMM_NT_CALLS.get(key, 0) + 1
+    _GEMM_NT_CALLS[key] = n
+    if n < _PICK_MIN_RECUR:
         return False
     ext = get_ext()
     import time as _time
