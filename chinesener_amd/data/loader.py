@@ -75,10 +75,19 @@ class DevicePrefetcher:
             self._next = host
             return
         try:
-            host = {k: (v.pin_memory()
-                        if isinstance(v, torch.Tensor) and v.device.type == "cpu"
-                        else v)
-                    for k, v in host.items()}
+            sig = tuple(sorted((k, tuple(v.shape), str(v.dtype))
+                               for k, v in host.items()
+                               if isinstance(v, torch.Tensor)))
+            stable = sig == getattr(self, "_last_host_sig", None)
+            self._last_host_sig = sig
+            if stable:
+                # pin only recurring shapes: torch's pinned-block cache
+                # misses on every ragged batch (MRC padded batching) and
+                # a fresh cudaHostAlloc costs 10-1000 ms
+                host = {k: (v.pin_memory()
+                            if isinstance(v, torch.Tensor)
+                            and v.device.type == "cpu" else v)
+                        for k, v in host.items()}
             bufs = self._ring_buffers(host)
             # _preload(t+2) is issued after the consumer ENQUEUED all
             # reads of batch t (the for-loop calls __next__ after
