@@ -185,7 +185,8 @@ class Trainer:
         best_eval, since_best = float("inf"), 0
         t0, last_log_step = time.time(), self.step
         losses = []
-        if str(self.device).startswith("cuda"):
+        if (str(self.device).startswith("cuda")
+                and os.environ.get("CHINESENER_NO_PREFETCH") != "1"):
             # one-batch-ahead pinned H2D staging on a side stream
             from ..data.loader import DevicePrefetcher
             batches = DevicePrefetcher(batches, self.device)
