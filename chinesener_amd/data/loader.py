@@ -80,14 +80,20 @@ class DevicePrefetcher:
                                if isinstance(v, torch.Tensor)))
             stable = sig == getattr(self, "_last_host_sig", None)
             self._last_host_sig = sig
-            if stable:
-                # pin only recurring shapes: torch's pinned-block cache
-                # misses on every ragged batch (MRC padded batching) and
-                # a fresh cudaHostAlloc costs 10-1000 ms
-                host = {k: (v.pin_memory()
-                            if isinstance(v, torch.Tensor)
-                            and v.device.type == "cpu" else v)
-                        for k, v in host.items()}
+            if not stable:
+                # ragged stream (MRC padded batching): hand the host
+                # batch through untouched — fresh pinned allocations
+                # (cudaHostAlloc misses the block cache every time) and
+                # side-stream pageable copies both cost 10-100x the
+                # model step; the consumer's _stage does the H2D
+                self._next = host
+                self._event = None
+                self._host = host
+                return
+            host = {k: (v.pin_memory()
+                        if isinstance(v, torch.Tensor)
+                        and v.device.type == "cpu" else v)
+                    for k, v in host.items()}
             bufs = self._ring_buffers(host)
             # _preload(t+2) is issued after the consumer ENQUEUED all
             # reads of batch t (the for-loop calls __next__ after
@@ -116,7 +122,7 @@ class DevicePrefetcher:
         if self._next is None:
             raise StopIteration
         batch = self._next
-        if self.use_stream:
+        if self.use_stream and self._event is not None:
             # ring buffers are persistent (no allocator hand-off), so a
             # stream wait on the copy event is all that's needed
             torch.cuda.current_stream(self.device).wait_event(self._event)

@@ -128,6 +128,19 @@ class Trainer:
             if _dist.is_initialized() and _dist.get_backend() != "nccl":
                 graph_ready = False
         if self.use_step_graph and graph_ready:
+            # ragged input streams (MRC padded batching) never settle on
+            # one shape: every capture would be dropped a step later.
+            # Give up on graphs as soon as shape diversity shows up.
+            ssig = self._staging_sig
+            seen = getattr(self, "_sigs_seen", None)
+            if seen is None:
+                seen = self._sigs_seen = set()
+            seen.add(ssig)
+            if len(seen) >= 4 and self._graph is None:
+                log.info("ragged batch shapes (%d distinct); disabling "
+                         "step graph", len(seen))
+                self.use_step_graph = False
+        if self.use_step_graph and graph_ready:
             if self._graph is not None and self._graph.matches(batch):
                 self.step += 1
                 loss = self._graph.replay(batch, self.step)
