@@ -78,8 +78,15 @@ class DevicePrefetcher:
             sig = tuple(sorted((k, tuple(v.shape), str(v.dtype))
                                for k, v in host.items()
                                if isinstance(v, torch.Tensor)))
-            stable = sig == getattr(self, "_last_host_sig", None)
+            if sig == getattr(self, "_last_host_sig", None):
+                self._stable_run = getattr(self, "_stable_run", 0) + 1
+            else:
+                self._stable_run = 0
             self._last_host_sig = sig
+            # require a SUSTAINED stable run: ragged streams repeat a
+            # shape by chance, and a single chance repeat must not pay
+            # a fresh cudaHostAlloc (10-1000 ms)
+            stable = self._stable_run >= 8
             if not stable:
                 # ragged stream (MRC padded batching): hand the host
                 # batch through untouched — fresh pinned allocations
