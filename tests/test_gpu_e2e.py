@@ -268,17 +268,23 @@ def test_device_prefetcher_gpu():
     _cuda()
     from chinesener_amd.data.loader import DevicePrefetcher
     src = [{"token_ids": torch.randint(0, 100, (4, 32)),
-            "w": torch.randn(4, 8)} for _ in range(6)]
+            "w": torch.randn(4, 8)} for _ in range(16)]
     n = 0
     ptrs = set()
+    staged = 0
     for host, dev in zip(src, DevicePrefetcher(iter(src), "cuda:0")):
         n += 1
         for k in host:
-            assert dev[k].is_cuda
+            # staging engages only after a sustained stable-shape run
+            # (ragged streams must not pay pinned allocations); early
+            # batches pass through as host tensors
+            if dev[k].is_cuda:
+                staged += 1
+                ptrs.add(dev[k].data_ptr())
             assert torch.equal(dev[k].cpu(), host[k])
-            ptrs.add(dev[k].data_ptr())
-    assert n == 6
-    # persistent 2-deep ring: 2 buffers per key, not 6
+    assert n == 16
+    assert staged >= 10   # steady state is device-staged
+    # persistent 2-deep ring: 2 buffers per key, not one per batch
     assert len(ptrs) == 4, len(ptrs)
     # and through trainer.train (the wrapping call site)
     from conftest import make_tiny_batch, make_tiny_params
