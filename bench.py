@@ -173,6 +173,33 @@ def main():
                     g.graph = None
                     opt.lr_dev = None
                 ok = False
+        if ok and dist is not None and world > 1:
+            # The captured DP body end-launches its collectives
+            # (autograd.grad fires no hooks), losing backward/comm
+            # overlap; eager keeps the bucket-hook overlap. Measure both
+            # and keep the faster — every rank sees the same reduced
+            # timings, so the choice stays collective-consistent.
+            def _timed(fn, n=5):
+                dist.barrier()
+                torch.cuda.synchronize()
+                t0 = time.perf_counter()
+                for i in range(n):
+                    fn(batches[i % len(batches)])
+                dist.barrier()
+                torch.cuda.synchronize()
+                t = torch.tensor([time.perf_counter() - t0], device=device)
+                dist.all_reduce(t, op=dist.ReduceOp.MAX)
+                return float(t.item())
+
+            t_graph = _timed(lambda b: g.replay(b, 1))
+            t_eager = _timed(train_step)
+            if t_eager < t_graph:
+                g.release()
+                ok = False
+                if rank == 0:
+                    print(f"[bench] eager DP ({t_eager:.3f}s/5) beats "
+                          f"graphed ({t_graph:.3f}s/5); using eager",
+                          file=sys.stderr)
         if ok:
             graphed = g
 
