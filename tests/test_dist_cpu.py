@@ -227,3 +227,75 @@ def test_reduce_in_graph_matches_finalize_backward():
     finally:
         if created:
             dist.destroy_process_group()
+
+
+WORKER_GRAPHPATH = """
+import os, sys, pickle
+sys.path.insert(0, os.environ["REPO"])
+import torch
+import torch.distributed as dist
+from chinesener_amd.dist import BucketedDataParallel, init_process_group
+from chinesener_amd.models import build_model
+from conftest import make_tiny_batch, make_tiny_params
+
+def main():
+    rank, world = init_process_group("gloo")
+    torch.manual_seed(123)
+    name = os.environ["MODEL"]
+    params = make_tiny_params(name)
+    model = build_model(name, params)
+    dp = BucketedDataParallel(model, bucket_cap_mb=0.05)
+    batch = make_tiny_batch(name, batch_size=2, seed=rank)
+    params_list = [p for p in model.parameters() if p.requires_grad]
+
+    # hook path (eager DP)
+    dp.zero_grad()
+    model(batch).loss.backward()
+    dp.finalize_backward()
+    ref = {n: p.grad.clone() for n, p in model.named_parameters()
+           if p.grad is not None}
+
+    # graphed-body path: autograd.grad into p.grad + reduce_in_graph
+    # (what GraphedTrainStep records; world-2 equivalence)
+    loss = model(batch).loss
+    grads = torch.autograd.grad(loss, params_list, allow_unused=True)
+    with torch.no_grad():
+        for p, g in zip(params_list, grads):
+            if p.grad is None and g is None:
+                continue
+            if p.grad is None:
+                p.grad = torch.zeros_like(p)
+            if g is None:
+                p.grad.zero_()
+            else:
+                p.grad.copy_(g)
+    dp.reduce_in_graph()
+    for n, p in model.named_parameters():
+        if n in ref:
+            assert torch.allclose(p.grad, ref[n], atol=1e-6), n
+    dist.destroy_process_group()
+
+main()
+"""
+
+
+def test_reduce_in_graph_world2_matches_hooks(tmp_path):
+    """World-size-2 gloo: the captured-step DP reduction
+    (autograd.grad + reduce_in_graph) produces the same averaged grads
+    as the hook + finalize path on every rank."""
+    with tempfile.NamedTemporaryFile("w", suffix=".py", delete=False) as f:
+        f.write(WORKER_GRAPHPATH)
+        script = f.name
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ, REPO=REPO, MODEL="bilstm_crf",
+                   RANK=str(rank), WORLD_SIZE="2",
+                   MASTER_ADDR="127.0.0.1", MASTER_PORT="29512",
+                   LOCAL_RANK=str(rank))
+        procs.append(subprocess.Popen([sys.executable, script],
+                                      env=env, stdout=subprocess.PIPE,
+                                      stderr=subprocess.PIPE, text=True))
+    for p in procs:
+        out, err = p.communicate(timeout=600)
+        assert p.returncode == 0, err[-3000:]
+    os.unlink(script)
