@@ -232,6 +232,7 @@ def test_reduce_in_graph_matches_finalize_backward():
 WORKER_GRAPHPATH = """
 import os, sys, pickle
 sys.path.insert(0, os.environ["REPO"])
+sys.path.insert(0, os.path.join(os.environ["REPO"], "tests"))
 import torch
 import torch.distributed as dist
 from chinesener_amd.dist import BucketedDataParallel, init_process_group
