@@ -10,6 +10,7 @@ from typing import List, Optional
 
 import torch
 import torch.nn as nn
+import torch.nn.functional as F
 
 from .. import ops
 
@@ -74,6 +75,21 @@ class MultiKernelCNN(nn.Module):
         self.output_size = filters * len(kernel_sizes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:   # [B,L,E]
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            # SAME conv1d as pad + unfold + linear: the unfolded GEMM is
+            # K-major NT, so it rides the in-tree MFMA GEMM / hipBLASLt
+            # measured dispatch like every other linear
+            outs = []
+            for c in self.convs:
+                k = c.kernel_size[0]
+                left = (k - 1) // 2
+                xp = F.pad(x, (0, 0, left, (k - 1) - left))     # [B,L+k-1,E]
+                u = xp.unfold(1, k, 1)                          # [B,L,E,k]
+                u = u.transpose(2, 3).reshape(x.shape[0], x.shape[1], -1)
+                w = c.weight.permute(0, 2, 1).reshape(c.weight.shape[0], -1)
+                y = ops.linear(u.contiguous(), w.contiguous(), c.bias)
+                outs.append(self.dropout(torch.relu(y)))
+            return torch.cat(outs, dim=-1)
         xt = x.transpose(1, 2)
         outs = [self.dropout(torch.relu(c(xt))) for c in self.convs]
         return torch.cat(outs, dim=1).transpose(1, 2)

@@ -732,3 +732,22 @@ def test_embed3_fused_gather():
                                atol=5e-2, rtol=2e-2)
     torch.testing.assert_close(t.grad.float(), t2.grad.float(),
                                atol=5e-2, rtol=2e-2)
+
+
+def test_cnn_unfold_gemm_matches_torch_conv():
+    """MultiKernelCNN's GPU path (pad + unfold + in-tree GEMM dispatch,
+    SURVEY K12) vs the torch conv1d path in fp32, incl. even kernels."""
+    _cuda()
+    torch.manual_seed(55)
+    from chinesener_amd.models.layers import MultiKernelCNN
+    m = MultiKernelCNN(96, filters=64, kernel_sizes=(2, 3, 4),
+                       keep_prob=1.0).cuda()
+    x = torch.randn(4, 32, 96, device="cuda")
+    ref = m(x)                              # fp32 -> torch conv branch
+    m16 = m.to(torch.bfloat16)
+    out = m16(x.to(torch.bfloat16))         # bf16 -> unfold+GEMM branch
+    torch.testing.assert_close(out.float(), ref, atol=0.08, rtol=0.05)
+    # backward runs through the custom path
+    x2 = x.to(torch.bfloat16).requires_grad_()
+    m16(x2).float().sum().backward()
+    assert torch.isfinite(x2.grad.float()).all()
