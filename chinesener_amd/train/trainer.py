@@ -164,6 +164,11 @@ class Trainer:
                     dp=self.dp, cast=self._cast)
                 if g.try_capture(batch, step=self.step + 1):
                     self._graph = g
+                    # a successful re-capture proves the stream is only
+                    # periodically ragged (epoch-end partial batches):
+                    # reset the instability counter so per-epoch drops
+                    # never permanently disable graphs
+                    self._shape_misses = 0
                     self.step += 1
                     return float(g.replay(batch, self.step).detach())
                 self.use_step_graph = False
