@@ -1,6 +1,11 @@
 #!/usr/bin/env python3
 """Offline hipBLASLt GEMM tuning for the training step's shapes.
 
+NOTE (round 2): TunableOp measured as a net loss on this stack and is
+now opt-in (CHINESENER_TUNABLE=1, see bench.py header and
+profiles/gemm_nt_r02.md). This script remains for re-evaluating it on
+future ROCm stacks.
+
 Runs the bench training loop with TunableOp tuning enabled, then writes
 the winners to gpurun_out/tunableop_gfx950.csv (merge back + commit to
 profiles/). Also times 10 post-tuning steps so the expected gain is
