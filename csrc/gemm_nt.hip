@@ -120,11 +120,14 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_kernel(
       for (int i = 0; i < 4; ++i) af[i] = frag(at, wr + i * 16, kk * 32);
 #pragma unroll
       for (int j = 0; j < 4; ++j) bfr[j] = frag(bt, wc + j * 16, kk * 32);
+      // favor this wave while its MFMA cluster runs (guide T5)
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < 4; ++i)
 #pragma unroll
         for (int j = 0; j < 4; ++j)
           acc[i][j] = mfma16(af[i], bfr[j], acc[i][j]);
+      __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();                         // joins + drains prefetch
     buf ^= 1;
