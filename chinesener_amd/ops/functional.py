@@ -155,6 +155,22 @@ def _tuning_active() -> bool:
         return False
 
 
+def _med_time(fn):
+    """Median kernel time: 3 batches of 8 launches per synchronize (a
+    per-call sync adds a constant ~10-30us that swamps 20-80us kernels)."""
+    import time as _time
+    fn(); fn()
+    torch.cuda.synchronize()
+    ts = []
+    for _ in range(3):
+        t0 = _time.perf_counter()
+        for _i in range(8):
+            fn()
+        torch.cuda.synchronize()
+        ts.append((_time.perf_counter() - t0) / 8)
+    return sorted(ts)[1]
+
+
 def _pick2(key, fn_ours, fn_lib) -> bool:
     """Measured dispatch between two GEMM paths: once the library's
     TunableOp tuning is done, time both (median of 5) and cache the
@@ -168,23 +184,8 @@ def _pick2(key, fn_ours, fn_lib) -> bool:
     _GEMM_NT_CALLS[key] = n
     if n < _PICK_MIN_RECUR:
         return False               # shape hasn't proven it recurs
-    import time as _time
 
-    def _med(fn):
-        # batch 8 launches per sync: a per-call synchronize adds a
-        # constant ~10-30us that swamps (and skews) 20-80us kernels
-        fn(); fn()
-        torch.cuda.synchronize()
-        ts = []
-        for _ in range(3):
-            t0 = _time.perf_counter()
-            for _i in range(8):
-                fn()
-            torch.cuda.synchronize()
-            ts.append((_time.perf_counter() - t0) / 8)
-        return sorted(ts)[1]
-
-    t_ours, t_lib = _med(fn_ours), _med(fn_lib)
+    t_ours, t_lib = _med_time(fn_ours), _med_time(fn_lib)
     _GEMM_NT_CHOICE[key] = bool(t_ours <= t_lib)
     if os.environ.get("CHINESENER_GEMM_DEBUG") == "1":
         import sys
@@ -211,25 +212,10 @@ def _pick_gemm_nt(x2, w, bf) -> bool:
     if n < _PICK_MIN_RECUR:
         return False
     ext = get_ext()
-    import time as _time
-
-    def _med(fn):
-        # batch 8 launches per sync: a per-call synchronize adds a
-        # constant ~10-30us that swamps (and skews) 20-80us kernels
-        fn(); fn()
-        torch.cuda.synchronize()
-        ts = []
-        for _ in range(3):
-            t0 = _time.perf_counter()
-            for _i in range(8):
-                fn()
-            torch.cuda.synchronize()
-            ts.append((_time.perf_counter() - t0) / 8)
-        return sorted(ts)[1]
 
     bb = bf.to(torch.bfloat16) if bf is not None else None
-    t_ours = _med(lambda: ext.gemm_nt(x2, w, bf, False))
-    t_lib = _med(lambda: F.linear(x2, w, bb))
+    t_ours = _med_time(lambda: ext.gemm_nt(x2, w, bf, False))
+    t_lib = _med_time(lambda: F.linear(x2, w, bb))
     _GEMM_NT_CHOICE[key] = bool(t_ours <= t_lib)
     if os.environ.get("CHINESENER_GEMM_DEBUG") == "1":
         import sys
