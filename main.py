@@ -63,6 +63,14 @@ def build_everything(args, rank: int, world_size: int):
     datas = args.data.split(",")
     name = args.rename or args.model_name
     mparams = model_params(args.model_name)
+    if len(datas) > 1 and not ("_mtl" in args.model_name
+                               or "_adv" in args.model_name):
+        raise SystemExit(
+            f"--data {args.data}: comma-separated datasets select "
+            f"multi-task training, which needs a shared-BERT mtl/adv "
+            f"model (bert_bilstm_crf_mtl or bert_bilstm_crf_adv) — "
+            f"'{args.model_name}' is single-task (reference main.py "
+            f"multitask_train has the same contract)")
     if len(datas) == 1:
         pipe = NerDataset(args.data_dir, datas[0], args.batch_size or
                           mparams.get("batch_size", 32), args.epochs,
