@@ -413,3 +413,44 @@ def test_checkpoint_manager_invariants(steps, keep):
     assert kept == sorted(steps)[-keep:]
     assert cm.latest().endswith(f"ckpt-{max(steps)}.pt")
     assert cm.restore(_t.nn.Linear(2, 2)) == max(steps)
+
+
+@given(
+    st.lists(
+        st.tuples(
+            st.text(alphabet="abcdefgh/_0123456789", min_size=1, max_size=24),
+            st.lists(st.integers(1, 5), min_size=0, max_size=3),
+            st.sampled_from(["float32", "int64", "int32", "float16"]),
+        ),
+        min_size=1, max_size=8, unique_by=lambda t: t[0]),
+    st.integers(0, 2**31 - 1),
+)
+@settings(max_examples=30, deadline=None)
+def test_tf_bundle_roundtrip_fuzz(specs, seed):
+    """TF v2 tensor-bundle writer->reader round-trip over random
+    variable names, shapes (incl. scalars) and dtypes."""
+    import tempfile
+
+    import numpy as np
+
+    from chinesener_amd.models.tf_checkpoint import (read_tf_checkpoint,
+                                                     write_tf_checkpoint)
+    rng = np.random.default_rng(seed)
+    tensors = {}
+    for name, shape, dt in specs:
+        if dt.startswith("int"):
+            arr = rng.integers(-100, 100, size=shape).astype(dt)
+        else:
+            arr = rng.standard_normal(size=shape).astype(dt)
+        tensors[name] = arr
+    with tempfile.TemporaryDirectory() as d:
+        import os as _os
+        prefix = _os.path.join(d, "m.ckpt")
+        write_tf_checkpoint(prefix, tensors)
+        back = read_tf_checkpoint(prefix)
+    assert set(back) == set(tensors)
+    for k, v in tensors.items():
+        got = back[k]
+        assert got.dtype == v.dtype, (k, got.dtype, v.dtype)
+        assert got.shape == v.shape, (k, got.shape, v.shape)
+        np.testing.assert_array_equal(np.asarray(got), v, err_msg=k)
