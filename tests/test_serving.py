@@ -266,3 +266,42 @@ def test_micro_batcher_mixed_keysets(tmp_path):
     assert ra.shape == (2, 32) and rb.shape == (1, 32)
     assert (ra == eng.predict(a)).all() and (rb == eng.predict(b)).all()
     batcher.close()
+
+
+def test_micro_batcher_ragged_stress(tmp_path):
+    """Many concurrent clients with MIXED batch sizes: every request
+    gets its own rows back bit-exactly; errors in one group don't leak
+    into others."""
+    import threading
+    _export_tiny(tmp_path)
+    from chinesener_amd.serve.engine import InferenceEngine, MicroBatcher
+    eng = InferenceEngine("bilstm_crf", str(tmp_path), use_graph=False,
+                          max_seq_len=32, batch_sizes=(1, 4, 8))
+    batcher = MicroBatcher(eng, window_ms=3.0)
+    rng = np.random.default_rng(11)
+    reqs = [{"token_ids": rng.integers(1, 200, (int(rng.integers(1, 5)), 32)),
+             "mask": np.ones((1, 32), dtype=np.int64).repeat(1, 0)}
+            for _ in range(24)]
+    for r in reqs:
+        r["mask"] = np.ones((r["token_ids"].shape[0], 32), dtype=np.int64)
+    expected = [eng.predict(r) for r in reqs]
+    results = [None] * len(reqs)
+    errs = []
+
+    def worker(i):
+        try:
+            results[i] = batcher.predict(reqs[i])
+        except Exception as e:
+            errs.append((i, e))
+
+    threads = [threading.Thread(target=worker, args=(i,))
+               for i in range(len(reqs))]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    assert not errs, errs
+    for got, want in zip(results, expected):
+        assert got is not None and got.shape == want.shape
+        assert (got == want).all()
+    batcher.close()
