@@ -296,3 +296,25 @@ def test_summary_artifacts_dump(tmp_path):
     assert trans_keys, blob.files
     assert blob[trans_keys[0]].shape == (7, 7)
     sl.close()
+
+
+def test_cpu_training_deterministic(tmp_path):
+    """Same seed -> bitwise-identical CPU training losses (SURVEY §5.2:
+    the determinism story; GPU runs are statistical — see
+    docs/NOTES.md)."""
+    import torch
+    from chinesener_amd.config import resolve_params
+    from chinesener_amd.models import build_model, model_params
+    from chinesener_amd.train.trainer import Trainer
+    from conftest import make_tiny_batch, make_tiny_params
+
+    def run(tag):
+        torch.manual_seed(7)
+        params = make_tiny_params("bilstm_crf")
+        model = build_model("bilstm_crf", params)
+        tr = Trainer(model, "bilstm_crf", params, str(tmp_path / tag))
+        return [tr.train_step(make_tiny_batch("bilstm_crf", batch_size=2,
+                                              seed=i))
+                for i in range(4)]
+
+    assert run("a") == run("b")
