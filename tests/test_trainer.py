@@ -318,3 +318,18 @@ def test_cpu_training_deterministic(tmp_path):
                 for i in range(4)]
 
     assert run("a") == run("b")
+
+
+def test_profiling_and_tunable_helpers():
+    """roctx ranges are safe no-ops without the marker library; the
+    TunableOp helpers are inert unless explicitly opted in."""
+    from chinesener_amd.utils.profiling import roctx_range, range_push, \
+        range_pop
+    with roctx_range("unit-test"):
+        range_push("inner")
+        range_pop()
+    import os
+    from chinesener_amd.ops import tunable
+    assert os.environ.get("CHINESENER_TUNABLE") != "1"
+    assert tunable.load_tuned_gemm_table() is False   # opt-in since r2
+    tunable.freeze()                                  # never raises
