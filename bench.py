@@ -170,9 +170,22 @@ def main():
             dist.all_reduce(flag, op=dist.ReduceOp.MIN)
             if not bool(flag.item()):
                 if ok:
-                    g.graph = None
+                    g.release()          # drops graph + graph-mode caches
+                else:
                     opt.lr_dev = None
                 ok = False
+                # a rank whose capture FAILED skipped the post-capture
+                # param restore (its warmup passes applied real updates)
+                # — re-sync params from rank 0 and restart optimizer
+                # state so all ranks continue identically
+                with torch.no_grad():
+                    for p in model.parameters():
+                        dist.broadcast(p.data, src=0)
+                opt.state.clear()
+                if hasattr(opt, "_meta_cache"):
+                    opt._meta_cache = {}
+                if hasattr(model, "_clip_meta"):
+                    model._clip_meta = None
         if ok and dist is not None and world > 1:
             # The captured DP body end-launches its collectives
             # (autograd.grad fires no hooks), losing backward/comm
