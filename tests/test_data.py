@@ -269,3 +269,41 @@ def test_max_prob_segment_vs_max_match():
 
     # OOV text degrades to single chars (freq-1 convention)
     assert lex.max_prob_segment("zz") == [(0, 1), (1, 2)]
+
+
+def test_multidataset_reshuffles_each_pass(tmp_path):
+    """The shorter task's corpus repeats within one mtl epoch; each
+    repeat pass must use a fresh shuffle order (the seed was
+    constant across passes)."""
+    from chinesener_amd.data.loader import MultiDataset
+    md = MultiDataset(str(tmp_path), ["msra", "msr"], batch_size=4,
+                      epochs=1, model_name="bert_bilstm_crf_mtl")
+    # drive enough batches that task streams wrap their corpora
+    orders = {0: [], 1: []}
+    n_batches = 0
+    for batch in md.iter_batches("train"):
+        n_batches += 1
+        tid = int(batch["task_ids"][0, 0])
+        orders[tid].append(batch["token_ids"][0].numpy().tobytes())
+        if n_batches >= 400:
+            break
+    # find a wrap in task-1's stream: the same sample set must not come
+    # back in the same order (first-k fingerprints differ across passes)
+    seq = orders[1] if len(orders[1]) > 10 else orders[0]
+    seen = {}
+    repeat_pairs = 0
+    same_successor = 0
+    for i, fp in enumerate(seq[:-1]):
+        if fp in seen:
+            repeat_pairs += 1
+            j = seen[fp]
+            if j + 1 < len(seq) and seq[j + 1] == seq[i + 1]:
+                same_successor += 1
+        else:
+            seen[fp] = i
+    if repeat_pairs >= 5:
+        # identical successor for every repeated sample would mean the
+        # pass order repeated verbatim
+        assert same_successor < repeat_pairs, (
+            f"{same_successor}/{repeat_pairs} repeated samples kept the "
+            "same successor — shuffle order repeated across passes")
