@@ -333,3 +333,33 @@ def test_profiling_and_tunable_helpers():
     assert os.environ.get("CHINESENER_TUNABLE") != "1"
     assert tunable.load_tuned_gemm_table() is False   # opt-in since r2
     tunable.freeze()                                  # never raises
+
+
+def test_bench_json_driver_contract():
+    """bench.py's single JSON line must satisfy the driver contract
+    (fields, types, aggregate semantics) — guards future edits."""
+    import json
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch_size", "2", "--seq_len", "32"],
+        capture_output=True, text=True, timeout=600,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    r = json.loads(line)
+    for k, t in [("metric", str), ("value", (int, float)), ("unit", str),
+                 ("n_gpus", int), ("steps", int), ("warmup", int),
+                 ("ms_per_step", (int, float)), ("higher_is_better", bool),
+                 ("scaling", str), ("dtype", str), ("data", str),
+                 ("config", dict)]:
+        assert k in r and isinstance(r[k], t), (k, r.get(k))
+    assert r["vs_baseline"] is None          # no published baseline
+    assert r["scaling"] == "weak"
+    assert r["data"] == "synthetic"
+    assert r["n_gpus"] == 1 and r["steps"] == 2 and r["warmup"] == 1
+    assert r["config"]["global_batch"] == 2
+    assert r["config"]["parallelism"] == "dp1"
+    # value is the whole-job aggregate: steps*batch/elapsed
+    assert abs(r["value"] - 2 * 2 / (r["ms_per_step"] * 2 / 1000)) < 1.0
