@@ -277,6 +277,12 @@ class MicroBatcher:
                 rows += n
             else:
                 rest.append(item)
+        if not group and self._queue:
+            # an oversize request (batch > max_batch) can never fill a
+            # group: dispatch it alone (the engine's eager path handles
+            # batches beyond the largest captured bucket)
+            group = [self._queue[0]]
+            rest = self._queue[1:]
         self._queue = rest
         return group
 

@@ -305,3 +305,20 @@ def test_micro_batcher_ragged_stress(tmp_path):
         assert got is not None and got.shape == want.shape
         assert (got == want).all()
     batcher.close()
+
+
+def test_micro_batcher_oversize_request(tmp_path):
+    """A request larger than max_batch must dispatch alone via the
+    engine's eager path, not starve in the queue."""
+    _export_tiny(tmp_path)
+    from chinesener_amd.serve.engine import InferenceEngine, MicroBatcher
+    eng = InferenceEngine("bilstm_crf", str(tmp_path), use_graph=False,
+                          max_seq_len=32, batch_sizes=(1, 4))
+    batcher = MicroBatcher(eng, window_ms=2.0)
+    rng = np.random.default_rng(5)
+    big = {"token_ids": rng.integers(1, 200, (9, 32)),
+           "mask": np.ones((9, 32), dtype=np.int64)}
+    out = batcher.predict(big)             # must not hang
+    assert out.shape == (9, 32)
+    assert (out == eng.predict(big)).all()
+    batcher.close()
