@@ -513,13 +513,11 @@ def test_wgrad_kernel_exact():
         torch.testing.assert_close(got, ref, atol=2.0, rtol=2e-2)
 
 
-@pytest.mark.xfail(strict=False,
-                   reason="wgrad2 (LDS-transposed B operand) is new this "
-                          "round and has not run on hardware yet")
 def test_wgrad2_kernel_exact():
     """v3 wgrad (vectorized B-fragment reads via two-pass LDS transpose,
-    csrc/wgrad.hip wgrad2): exact vs fp32 matmul. Non-strict xfail until
-    first hardware validation — see docs/ROADMAP.md item 1."""
+    csrc/wgrad.hip wgrad2): exact vs fp32 matmul. Hardware-validated in
+    round 2 (exact but SLOWER than v2/hipBLASLt — kept as a measured
+    baseline, profiles/gemm_nt_r02.md)."""
     _cuda()
     torch.manual_seed(34)
     ext = ops.get_ext()
